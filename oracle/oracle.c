@@ -1,0 +1,553 @@
+/*
+ * oracle.c — CPU restatement of the Spark SQL hot-path semantics.
+ *
+ * TEST INFRASTRUCTURE ONLY. This library is the parity oracle for the GPU
+ * engine: only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline
+ * leg may call it, and only as the checker / reported CPU baseline. It is
+ * never the shipped compute path.
+ *
+ * Every function cites the reference file:line it restates
+ * (reference = apache/spark 5.0.0-SNAPSHOT at /root/reference; paths below
+ * are relative to that tree). The reference is JVM code and cannot be built
+ * in this container (no JVM — see SURVEY.md §8(c)), so this C restatement is
+ * the in-container stand-in, pinned by:
+ *   - Murmur3 known-answer vectors from
+ *     common/unsafe/src/test/java/org/apache/spark/unsafe/hash/Murmur3_x86_32Suite.java:40-56
+ *     (tests/test_oracle_murmur.py)
+ *   - the RadixSortSuite fuzz procedure re-created with the reference's own
+ *     generator (XORShiftRandom, core/src/main/scala/org/apache/spark/util/random/XORShiftRandom.scala:36-68)
+ *     and checked against an INDEPENDENT sort (numpy) in tests/test_oracle_radix.py
+ *     (procedure: core/src/test/scala/org/apache/spark/util/collection/unsafe/sort/RadixSortSuite.scala:119-200)
+ *   - property/KAT tests for aggregation and join semantics ("parity pinned
+ *     by construction" for those two — SURVEY.md §8(c) flags the same for
+ *     TPC-H since the reference ships no golden outputs runnable here).
+ *
+ * Build: gcc -O2 -shared -fPIC (oracle/Makefile). No external deps.
+ */
+
+#include <stdint.h>
+#include <stdbool.h>
+#include <string.h>
+#include <stdlib.h>
+
+#define EXPORT __attribute__((visibility("default")))
+
+/* ------------------------------------------------------------------ */
+/* Murmur3_x86_32 — restates                                           */
+/* common/unsafe/src/main/java/org/apache/spark/unsafe/hash/Murmur3_x86_32.java */
+/* ------------------------------------------------------------------ */
+
+static inline uint32_t rotl32(uint32_t x, int r) { return (x << r) | (x >> (32 - r)); }
+
+/* Murmur3_x86_32.java:125-130 mixK1 */
+static inline uint32_t mm3_mixK1(uint32_t k1) {
+  k1 *= 0xcc9e2d51u;
+  k1 = rotl32(k1, 15);
+  k1 *= 0x1b873593u;
+  return k1;
+}
+
+/* Murmur3_x86_32.java:132-137 mixH1 */
+static inline uint32_t mm3_mixH1(uint32_t h1, uint32_t k1) {
+  h1 ^= k1;
+  h1 = rotl32(h1, 13);
+  h1 = h1 * 5u + 0xe6546b64u;
+  return h1;
+}
+
+/* Murmur3_x86_32.java:140-147 fmix */
+static inline uint32_t mm3_fmix(uint32_t h1, uint32_t length) {
+  h1 ^= length;
+  h1 ^= h1 >> 16;
+  h1 *= 0x85ebca6bu;
+  h1 ^= h1 >> 13;
+  h1 *= 0xc2b2ae35u;
+  h1 ^= h1 >> 16;
+  return h1;
+}
+
+/* Murmur3_x86_32.java:45-52 hashInt */
+EXPORT int32_t mm3_hash_int(int32_t input, int32_t seed) {
+  uint32_t k1 = mm3_mixK1((uint32_t)input);
+  uint32_t h1 = mm3_mixH1((uint32_t)seed, k1);
+  return (int32_t)mm3_fmix(h1, 4);
+}
+
+/* Murmur3_x86_32.java:109-122 hashLong */
+EXPORT int32_t mm3_hash_long(int64_t input, int32_t seed) {
+  uint32_t low  = (uint32_t)((uint64_t)input);
+  uint32_t high = (uint32_t)(((uint64_t)input) >> 32);
+  uint32_t k1 = mm3_mixK1(low);
+  uint32_t h1 = mm3_mixH1((uint32_t)seed, k1);
+  k1 = mm3_mixK1(high);
+  h1 = mm3_mixH1(h1, k1);
+  return (int32_t)mm3_fmix(h1, 8);
+}
+
+/* Murmur3_x86_32.java:84-95 hashUnsafeBytes2 (little-endian ints, tail packed LE) */
+EXPORT int32_t mm3_hash_bytes2(const uint8_t* data, int32_t len, int32_t seed) {
+  int32_t lengthAligned = len - len % 4;
+  uint32_t h1 = (uint32_t)seed;
+  for (int32_t i = 0; i < lengthAligned; i += 4) {
+    uint32_t halfWord = (uint32_t)data[i] | ((uint32_t)data[i+1] << 8) |
+                        ((uint32_t)data[i+2] << 16) | ((uint32_t)data[i+3] << 24);
+    h1 = mm3_mixH1(h1, mm3_mixK1(halfWord));
+  }
+  uint32_t k1 = 0;
+  for (int32_t i = lengthAligned, shift = 0; i < len; i++, shift += 8) {
+    k1 ^= ((uint32_t)data[i] & 0xFFu) << shift;
+  }
+  h1 ^= mm3_mixK1(k1);
+  return (int32_t)mm3_fmix(h1, (uint32_t)len);
+}
+
+/* ------------------------------------------------------------------ */
+/* Partition id — restates                                             */
+/* sql/catalyst/.../plans/physical/partitioning.scala:328-330          */
+/*   partitionIdExpression = Pmod(Murmur3Hash(expressions, 42), n)     */
+/* Murmur3Hash over multiple columns chains the hash as the seed of    */
+/* the next column (sql/catalyst/.../expressions/hash.scala:849-860,   */
+/* HashExpression.eval: hash = computeHash(value, dataType, hash));    */
+/* a NULL column leaves the running hash unchanged.                    */
+/* Pmod(a, n) = ((a % n) + n) % n (arithmetic.scala Pmod.pmod).        */
+/* ------------------------------------------------------------------ */
+
+EXPORT int32_t spark_pmod(int32_t a, int32_t n) {
+  int32_t r = a % n;
+  return (r < 0) ? r + n : r;
+}
+
+/* partition ids for a single int64 key column with optional validity.
+ * validity: 1 bit per row, LSB-first within each byte (Arrow layout);
+ * NULL pass the seed through unchanged (hash.scala HashExpression.eval). */
+EXPORT void oracle_partition_ids_i64(const int64_t* keys, const uint8_t* validity,
+                                     int64_t n, int32_t num_parts, int32_t* out_pids) {
+  for (int64_t i = 0; i < n; i++) {
+    int32_t h = 42;
+    if (!validity || (validity[i >> 3] >> (i & 7)) & 1) {
+      h = mm3_hash_long(keys[i], h);
+    }
+    out_pids[i] = spark_pmod(h, num_parts);
+  }
+}
+
+/* ------------------------------------------------------------------ */
+/* Sort-prefix encodings — restates                                    */
+/* core/.../unsafe/sort/PrefixComparators.java:66-83 (double) and the  */
+/* signed-long ordering (SignedPrefixComparator :150).                 */
+/* ------------------------------------------------------------------ */
+
+/* PrefixComparators.java:72-83 DoublePrefixComparator.computePrefix */
+EXPORT uint64_t prefix_double(double value) {
+  if (value == -0.0) value = 0.0;
+  uint64_t bits;
+  /* Java Double.doubleToLongBits canonicalizes every NaN to 0x7ff8000000000000 */
+  if (value != value) {
+    bits = 0x7ff8000000000000ULL;
+  } else {
+    memcpy(&bits, &value, 8);
+  }
+  uint64_t mask = (uint64_t)(-(int64_t)(bits >> 63)) | 0x8000000000000000ULL;
+  return bits ^ mask;
+}
+
+/* ------------------------------------------------------------------ */
+/* LSB radix sort — faithful restatement of                            */
+/* core/src/main/java/org/apache/spark/util/collection/unsafe/sort/RadixSort.java */
+/* ------------------------------------------------------------------ */
+
+/* RadixSort.java:157-176 transformCountsToOffsets (element indices, not bytes) */
+static void transform_counts_to_offsets(int64_t* counts, int64_t numRecords,
+                                        int64_t outIndex, int64_t recLongs,
+                                        bool desc, bool sgn) {
+  int start = sgn ? 128 : 0; /* output the negative records first (129-255) */
+  if (desc) {
+    int64_t pos = numRecords;
+    for (int i = start; i < start + 256; i++) {
+      pos -= counts[i & 0xff];
+      counts[i & 0xff] = outIndex + pos * recLongs;
+    }
+  } else {
+    int64_t pos = 0;
+    for (int i = start; i < start + 256; i++) {
+      int64_t tmp = counts[i & 0xff];
+      counts[i & 0xff] = outIndex + pos * recLongs;
+      pos += tmp;
+    }
+  }
+}
+
+/* RadixSort.java:43-69 sort(): array has 2n long slots, data in [0,n).
+ * Returns the start index (0 or n) of the sorted data. */
+EXPORT int64_t oracle_radix_sort_longs(uint64_t* arr, int64_t n,
+                                       int32_t startByte, int32_t endByte,
+                                       bool desc, bool sgn) {
+  if (n <= 0) return 0;
+  int64_t inIndex = 0, outIndex = n;
+  /* getCounts (:108-139): skip bytes where all values agree */
+  uint64_t bitwiseMax = 0, bitwiseMin = ~0ULL;
+  for (int64_t i = 0; i < n; i++) { bitwiseMax |= arr[i]; bitwiseMin &= arr[i]; }
+  uint64_t bitsChanged = bitwiseMin ^ bitwiseMax;
+  for (int32_t b = startByte; b <= endByte; b++) {
+    if (((bitsChanged >> (b * 8)) & 0xff) == 0) continue;
+    int64_t counts[256];
+    memset(counts, 0, sizeof(counts));
+    for (int64_t i = 0; i < n; i++) counts[(arr[inIndex + i] >> (b * 8)) & 0xff]++;
+    transform_counts_to_offsets(counts, n, outIndex, 1, desc, sgn && b == endByte);
+    for (int64_t i = 0; i < n; i++) {
+      uint64_t v = arr[inIndex + i];
+      arr[counts[(v >> (b * 8)) & 0xff]++] = v;
+    }
+    int64_t t = inIndex; inIndex = outIndex; outIndex = t;
+  }
+  return inIndex;
+}
+
+/* RadixSort.java:178-213 sortKeyPrefixArray(): records are (long key, long
+ * prefix) pairs, sorted on the SECOND long. array has 4n long slots.
+ * Returns start LONG index (0 or 2n) of the sorted records. */
+EXPORT int64_t oracle_radix_sort_key_prefix(uint64_t* arr, int64_t n,
+                                            int32_t startByte, int32_t endByte,
+                                            bool desc, bool sgn) {
+  if (n <= 0) return 0;
+  int64_t inIndex = 0, outIndex = 2 * n;
+  uint64_t bitwiseMax = 0, bitwiseMin = ~0ULL;
+  for (int64_t i = 0; i < n; i++) {
+    uint64_t p = arr[2 * i + 1];
+    bitwiseMax |= p; bitwiseMin &= p;
+  }
+  uint64_t bitsChanged = bitwiseMin ^ bitwiseMax;
+  for (int32_t b = startByte; b <= endByte; b++) {
+    if (((bitsChanged >> (b * 8)) & 0xff) == 0) continue;
+    int64_t counts[256];
+    memset(counts, 0, sizeof(counts));
+    for (int64_t i = 0; i < n; i++)
+      counts[(arr[inIndex + 2 * i + 1] >> (b * 8)) & 0xff]++;
+    transform_counts_to_offsets(counts, n, outIndex, 2, desc, sgn && b == endByte);
+    for (int64_t i = 0; i < n; i++) {
+      uint64_t key = arr[inIndex + 2 * i];
+      uint64_t prefix = arr[inIndex + 2 * i + 1];
+      int64_t dest = counts[(prefix >> (b * 8)) & 0xff];
+      arr[dest] = key; arr[dest + 1] = prefix;
+      counts[(prefix >> (b * 8)) & 0xff] += 2;
+    }
+    int64_t t = inIndex; inIndex = outIndex; outIndex = t;
+  }
+  return inIndex;
+}
+
+/* ------------------------------------------------------------------ */
+/* Operator-level sort oracle: ORDER BY one int64 or float64 column.   */
+/* Restates SortExec (sql/core/.../execution/SortExec.scala:75-126):   */
+/* radix-eligible single-column sort via (rowid, prefix) pairs; NULL   */
+/* rows are kept aside in input order and placed per nullOrdering      */
+/* (SortOrder defaults: asc=>NULLS FIRST, desc=>NULLS LAST —           */
+/* catalyst/.../expressions/SortOrder.scala:35-45).                    */
+/* Emits the sorted row PERMUTATION (indices into the input).          */
+/* ------------------------------------------------------------------ */
+
+EXPORT void oracle_sort_perm_i64(const int64_t* keys, const uint8_t* validity,
+                                 int64_t n, bool desc, bool nulls_first,
+                                 int64_t* out_perm) {
+  /* pairs: (rowid, prefix). signed radix for int64 keys. */
+  uint64_t* pairs = (uint64_t*)malloc((size_t)(4 * n) * 8);
+  int64_t* nulls = (int64_t*)malloc((size_t)n * 8);
+  int64_t nn = 0, nv = 0;
+  for (int64_t i = 0; i < n; i++) {
+    if (validity && !((validity[i >> 3] >> (i & 7)) & 1)) { nulls[nn++] = i; continue; }
+    pairs[2 * nv] = (uint64_t)i;
+    pairs[2 * nv + 1] = (uint64_t)keys[i];
+    nv++;
+  }
+  int64_t off = oracle_radix_sort_key_prefix(pairs, nv, 0, 7, desc, true);
+  int64_t w = 0;
+  if (nulls_first) for (int64_t i = 0; i < nn; i++) out_perm[w++] = nulls[i];
+  for (int64_t i = 0; i < nv; i++) out_perm[w++] = (int64_t)pairs[off + 2 * i];
+  if (!nulls_first) for (int64_t i = 0; i < nn; i++) out_perm[w++] = nulls[i];
+  free(pairs); free(nulls);
+}
+
+EXPORT void oracle_sort_perm_f64(const double* keys, const uint8_t* validity,
+                                 int64_t n, bool desc, bool nulls_first,
+                                 int64_t* out_perm) {
+  uint64_t* pairs = (uint64_t*)malloc((size_t)(4 * n) * 8);
+  int64_t* nulls = (int64_t*)malloc((size_t)n * 8);
+  int64_t nn = 0, nv = 0;
+  for (int64_t i = 0; i < n; i++) {
+    if (validity && !((validity[i >> 3] >> (i & 7)) & 1)) { nulls[nn++] = i; continue; }
+    pairs[2 * nv] = (uint64_t)i;
+    pairs[2 * nv + 1] = prefix_double(keys[i]); /* unsigned radix (DOUBLE uses UnsignedPrefixComparator, PrefixComparators.java:49) */
+    nv++;
+  }
+  int64_t off = oracle_radix_sort_key_prefix(pairs, nv, 0, 7, desc, false);
+  int64_t w = 0;
+  if (nulls_first) for (int64_t i = 0; i < nn; i++) out_perm[w++] = nulls[i];
+  for (int64_t i = 0; i < nv; i++) out_perm[w++] = (int64_t)pairs[off + 2 * i];
+  if (!nulls_first) for (int64_t i = 0; i < nn; i++) out_perm[w++] = nulls[i];
+  free(pairs); free(nulls);
+}
+
+/* ------------------------------------------------------------------ */
+/* Hash aggregate oracle: GROUP BY int64 key -> SUM(float64), COUNT.   */
+/* Restates the observable semantics of HashAggregateExec              */
+/* (sql/core/.../aggregate/HashAggregateExec.scala:99-151) +           */
+/* TungstenAggregationIterator.processInputs (:206-300):               */
+/*  - every distinct key (NULL is a group) produces one output row     */
+/*  - SUM(float64): null-skipping sequential f64 accumulation in input */
+/*    order (Sum.scala:113-141, shouldTrackIsEmpty=false for double:   */
+/*    sum starts null, coalesce(add(coalesce(sum,0), v), sum));        */
+/*  - COUNT(col): number of non-null inputs (Count.scala)              */
+/*  - COUNT(*): number of rows.                                        */
+/* Groups are emitted in first-occurrence order (deterministic; the    */
+/* parity checker is order-insensitive like QueryTest.checkAnswer,     */
+/* sql/core/src/test/scala/org/apache/spark/sql/QueryTest.scala:160).  */
+/* Returns the number of groups. out_* arrays must hold >= n entries.  */
+/* out_key_valid[g]=0 marks the NULL-key group.                        */
+/* out_sum_valid[g]=0 marks an all-NULL-input SUM (result NULL).       */
+/* ------------------------------------------------------------------ */
+
+typedef struct {
+  int64_t key;
+  int64_t gidx;   /* -1 = empty */
+  uint8_t key_valid;
+} agg_slot;
+
+EXPORT int64_t oracle_hash_agg_i64_f64(const int64_t* keys, const uint8_t* key_validity,
+                                       const double* vals, const uint8_t* val_validity,
+                                       int64_t n,
+                                       int64_t* out_keys, uint8_t* out_key_valid,
+                                       double* out_sums, uint8_t* out_sum_valid,
+                                       int64_t* out_counts) {
+  uint64_t cap = 16;
+  while (cap < (uint64_t)n * 2) cap <<= 1;
+  agg_slot* table = (agg_slot*)malloc(cap * sizeof(agg_slot));
+  for (uint64_t i = 0; i < cap; i++) table[i].gidx = -1;
+  int64_t ngroups = 0;
+  int64_t null_group = -1;
+  for (int64_t i = 0; i < n; i++) {
+    bool kv = !key_validity || ((key_validity[i >> 3] >> (i & 7)) & 1);
+    int64_t g;
+    if (!kv) {
+      if (null_group < 0) {
+        null_group = ngroups++;
+        out_keys[null_group] = 0; out_key_valid[null_group] = 0;
+        out_sums[null_group] = 0.0; out_sum_valid[null_group] = 0;
+        out_counts[null_group] = 0;
+      }
+      g = null_group;
+    } else {
+      int64_t k = keys[i];
+      uint64_t h = (uint64_t)(uint32_t)mm3_hash_long(k, 42);
+      uint64_t s = h & (cap - 1);
+      for (;;) {
+        if (table[s].gidx < 0) {
+          table[s].key = k; table[s].key_valid = 1; table[s].gidx = ngroups;
+          out_keys[ngroups] = k; out_key_valid[ngroups] = 1;
+          out_sums[ngroups] = 0.0; out_sum_valid[ngroups] = 0;
+          out_counts[ngroups] = 0;
+          ngroups++;
+          break;
+        }
+        if (table[s].key == k) break;
+        s = (s + 1) & (cap - 1);
+      }
+      g = table[s].gidx;
+    }
+    bool vv = !val_validity || ((val_validity[i >> 3] >> (i & 7)) & 1);
+    if (vv) {
+      out_sums[g] += vals[i];
+      out_sum_valid[g] = 1;
+      out_counts[g] += 1;
+    }
+  }
+  free(table);
+  return ngroups;
+}
+
+/* ------------------------------------------------------------------ */
+/* Inner equi-join oracle on int64 keys.                               */
+/* Restates ShuffledHashJoinExec inner-join semantics                  */
+/* (sql/core/.../joins/ShuffledHashJoinExec.scala:103-132 +            */
+/* HashedRelation.scala LongHashedRelation :993): build a hash table   */
+/* on the build side, stream the probe side; NULL keys never match     */
+/* (HashJoin: null-aware is a separate operator).                      */
+/* Output: (probe_rid, build_rid) pairs; for each probe row in input   */
+/* order, its matches in build insertion order. Two-phase:             */
+/* oracle_join_count then oracle_join_emit with caller buffers.        */
+/* ------------------------------------------------------------------ */
+
+typedef struct {
+  uint64_t cap;
+  int64_t* slot_key;
+  int64_t* slot_head;   /* -1 empty, else build row id of first match */
+  int64_t* next;        /* per build row: next row with same key, -1 end */
+  const int64_t* bkeys;
+  const uint8_t* bvalid;
+  int64_t bn;
+} join_table;
+
+static join_table* build_join_table(const int64_t* bkeys, const uint8_t* bvalid, int64_t bn) {
+  join_table* t = (join_table*)malloc(sizeof(join_table));
+  uint64_t cap = 16;
+  while (cap < (uint64_t)bn * 2) cap <<= 1;
+  t->cap = cap;
+  t->slot_key = (int64_t*)malloc(cap * 8);
+  t->slot_head = (int64_t*)malloc(cap * 8);
+  t->next = (int64_t*)malloc((size_t)bn * 8);
+  t->bkeys = bkeys; t->bvalid = bvalid; t->bn = bn;
+  for (uint64_t i = 0; i < cap; i++) t->slot_head[i] = -1;
+  /* insert build rows in order; chains keep insertion order via tail append */
+  int64_t* slot_tail = (int64_t*)malloc(cap * 8);
+  for (int64_t i = 0; i < bn; i++) {
+    t->next[i] = -1;
+    if (bvalid && !((bvalid[i >> 3] >> (i & 7)) & 1)) continue;
+    int64_t k = bkeys[i];
+    uint64_t s = ((uint64_t)(uint32_t)mm3_hash_long(k, 42)) & (cap - 1);
+    for (;;) {
+      if (t->slot_head[s] < 0) { t->slot_head[s] = i; slot_tail[s] = i; t->slot_key[s] = k; break; }
+      if (t->slot_key[s] == k) { t->next[slot_tail[s]] = i; slot_tail[s] = i; break; }
+      s = (s + 1) & (cap - 1);
+    }
+  }
+  free(slot_tail);
+  return t;
+}
+
+static void free_join_table(join_table* t) {
+  free(t->slot_key); free(t->slot_head); free(t->next); free(t);
+}
+
+/* count + emit in one call; pass out_* = NULL to only count. */
+EXPORT int64_t oracle_join_inner_i64(const int64_t* bkeys, const uint8_t* bvalid, int64_t bn,
+                                     const int64_t* pkeys, const uint8_t* pvalid, int64_t pn,
+                                     int64_t* out_probe_rid, int64_t* out_build_rid,
+                                     int64_t out_cap) {
+  join_table* t = build_join_table(bkeys, bvalid, bn);
+  int64_t cnt = 0;
+  for (int64_t i = 0; i < pn; i++) {
+    if (pvalid && !((pvalid[i >> 3] >> (i & 7)) & 1)) continue;
+    int64_t k = pkeys[i];
+    uint64_t s = ((uint64_t)(uint32_t)mm3_hash_long(k, 42)) & (t->cap - 1);
+    for (;;) {
+      if (t->slot_head[s] < 0) break;
+      if (t->slot_key[s] == k) {
+        for (int64_t b = t->slot_head[s]; b >= 0; b = t->next[b]) {
+          if (out_probe_rid && cnt < out_cap) { out_probe_rid[cnt] = i; out_build_rid[cnt] = b; }
+          cnt++;
+        }
+        break;
+      }
+      s = (s + 1) & (t->cap - 1);
+    }
+  }
+  free_join_table(t);
+  return cnt;
+}
+
+/* ------------------------------------------------------------------ */
+/* Reference test-data generators, restated for golden-vector parity   */
+/* with the reference's own suites (NOT part of any product path).     */
+/* ------------------------------------------------------------------ */
+
+/* scala.util.hashing.MurmurHash3.bytesHash(data, seed): standard        */
+/* Murmur3 x86_32 over LE 4-byte words (== mm3_hash_bytes2 semantics).   */
+/* XORShiftRandom.hashSeed (XORShiftRandom.scala:62-68): seed bytes are  */
+/* BIG-endian (ByteBuffer.putLong), arraySeed = 0x3c074a61.              */
+static int64_t xorshift_hash_seed(int64_t init) {
+  uint8_t bytes[8];
+  for (int i = 0; i < 8; i++) bytes[i] = (uint8_t)(((uint64_t)init) >> (8 * (7 - i)));
+  int32_t low = mm3_hash_bytes2(bytes, 8, 0x3c074a61);
+  int32_t high = mm3_hash_bytes2(bytes, 8, low);
+  return ((int64_t)high << 32) | ((int64_t)low & 0xFFFFFFFFLL);
+}
+
+typedef struct { int64_t seed; } xorshift_t;
+
+EXPORT void xorshift_init(xorshift_t* r, int64_t init) { r->seed = xorshift_hash_seed(init); }
+
+/* XORShiftRandom.scala:44-51 next(bits) */
+static int32_t xorshift_next(xorshift_t* r, int bits) {
+  int64_t s = r->seed;
+  int64_t nextSeed = s ^ (int64_t)((uint64_t)s << 21);
+  nextSeed ^= (int64_t)((uint64_t)nextSeed >> 35);
+  nextSeed ^= (int64_t)((uint64_t)nextSeed << 4);
+  r->seed = nextSeed;
+  return (int32_t)(nextSeed & ((1LL << bits) - 1));
+}
+
+/* java.util.Random.nextLong: ((long)next(32) << 32) + next(32) */
+EXPORT int64_t xorshift_next_long(xorshift_t* r) {
+  int64_t hi = (int64_t)xorshift_next(r, 32);
+  int64_t lo = (int64_t)xorshift_next(r, 32); /* sign-extended add, as in Java */
+  return (hi << 32) + lo;
+}
+
+/* java.util.Random.nextInt(bound) with the overridden next() */
+EXPORT int32_t xorshift_next_int(xorshift_t* r, int32_t bound) {
+  if ((bound & -bound) == bound)
+    return (int32_t)(((int64_t)bound * (int64_t)xorshift_next(r, 31)) >> 31);
+  int32_t bits, val;
+  do {
+    bits = xorshift_next(r, 31);
+    val = bits % bound;
+  } while (bits - val + (bound - 1) < 0);
+  return val;
+}
+
+/* allocate-free helper for ctypes: state as plain int64 in/out */
+EXPORT int64_t xorshift_state_init(int64_t init) { return xorshift_hash_seed(init); }
+EXPORT int64_t xorshift_state_next_long(int64_t* state) {
+  xorshift_t r = { *state };
+  int64_t v = xorshift_next_long(&r);
+  *state = r.seed;
+  return v;
+}
+EXPORT int32_t xorshift_state_next_int(int64_t* state, int32_t bound) {
+  xorshift_t r = { *state };
+  int32_t v = xorshift_next_int(&r, bound);
+  *state = r.seed;
+  return v;
+}
+
+/* Fill an array with nextLong & mask — the RadixSortSuite generator
+ * (RadixSortSuite.scala:77-88 generateTestData / generateKeyPrefixTestData). */
+EXPORT void xorshift_fill_longs(int64_t* state, int64_t* out, int64_t n, int64_t mask) {
+  xorshift_t r = { *state };
+  for (int64_t i = 0; i < n; i++) out[i] = xorshift_next_long(&r) & mask;
+  *state = r.seed;
+}
+
+/* ------------------------------------------------------------------ */
+/* Synthetic bench data generator (OUR spec, shared bit-exactly with   */
+/* the HIP generator in spark_amd/csrc/gen.hip — see DESIGN.md §data). */
+/* splitmix64 over (seed, index).                                      */
+/* ------------------------------------------------------------------ */
+
+static inline uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ULL;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+  return x ^ (x >> 31);
+}
+
+EXPORT uint64_t gen_u64(uint64_t seed, uint64_t i) { return splitmix64(seed * 0x9E3779B97F4A7C15ULL + i); }
+
+EXPORT void gen_fill_u64(uint64_t seed, uint64_t start, int64_t n, uint64_t* out) {
+  for (int64_t i = 0; i < n; i++) out[i] = gen_u64(seed, start + (uint64_t)i);
+}
+
+/* keys uniform in [0, range): gen_u64 % range (range=0 -> full u64) */
+EXPORT void gen_fill_i64_range(uint64_t seed, uint64_t start, int64_t n, uint64_t range, int64_t* out) {
+  for (int64_t i = 0; i < n; i++) {
+    uint64_t v = gen_u64(seed, start + (uint64_t)i);
+    out[i] = (int64_t)(range ? v % range : v);
+  }
+}
+
+/* float64 uniform in [0,1): top 53 bits / 2^53 */
+EXPORT void gen_fill_f64_unit(uint64_t seed, uint64_t start, int64_t n, double* out) {
+  for (int64_t i = 0; i < n; i++) {
+    uint64_t v = gen_u64(seed, start + (uint64_t)i);
+    out[i] = (double)(v >> 11) * (1.0 / 9007199254740992.0);
+  }
+}
