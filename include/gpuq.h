@@ -1,0 +1,177 @@
+/*
+ * gpuq.h — C-ABI of the MI355X-native columnar execution engine (libgpuq.so).
+ *
+ * This is the drop-in boundary beneath Spark's columnar plugin API: a Scala
+ * host layer (SparkSessionExtensions.injectColumnar + ColumnarRule,
+ * sql/core/src/main/scala/org/apache/spark/sql/SparkSessionExtensions.scala:168
+ * and sql/core/.../execution/Columnar.scala:36-50) binds these entry points
+ * over JNI and swaps SortExec / HashAggregateExec / ShuffledHashJoinExec /
+ * ShuffleExchangeExec for GPU exec nodes whose doExecuteColumnar() calls
+ * land here. The JNI stub a Spark maintainer would add is shown in
+ * INTEGRATION.md; in this repo the same ABI is driven by the Python host
+ * mirror (spark_amd/) used for tests and benchmarks.
+ *
+ * Conventions:
+ *  - all device pointers are raw HIP device pointers on the CALLER's current
+ *    HIP device; `stream` is a hipStream_t (pass 0 for the default stream).
+ *    One executor task = one stream; calls are thread-safe per stream.
+ *  - the caller owns every buffer, including scratch workspaces, sized via
+ *    the gpuq_*_workspace_bytes() helpers. No allocation inside libgpuq.
+ *  - column buffers follow the Arrow layout used by Spark's ColumnVector
+ *    API (sql/catalyst/src/main/java/org/apache/spark/sql/vectorized/
+ *    ColumnVector.java:58-366): a dense data buffer plus an optional
+ *    validity bitmap (1 bit/row, LSB-first; NULL bitmap ptr = no nulls).
+ *  - return value: 0 on success, non-zero error code; gpuq_last_error()
+ *    returns a thread-local message (mapped to exceptions by the host,
+ *    mirroring SparkOutOfMemoryError semantics for device OOM).
+ */
+#ifndef GPUQ_H
+#define GPUQ_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- error handling ---- */
+#define GPUQ_OK 0
+#define GPUQ_ERR_HIP 1          /* underlying HIP call failed           */
+#define GPUQ_ERR_INVALID 2      /* bad argument / unsupported combo     */
+#define GPUQ_ERR_OVERFLOW 3     /* hash table or output buffer overflow */
+
+const char* gpuq_last_error(void);
+int gpuq_device_count(void);
+
+/* ---- dtypes ---- */
+#define GPUQ_INT64 0
+#define GPUQ_FLOAT64 1
+#define GPUQ_INT32 2
+
+/* One column of a ColumnarBatch (ColumnarBatch.java:61-123 access contract,
+ * device-resident). */
+typedef struct gpuq_col {
+  void* data;                 /* device ptr to dense values              */
+  const uint8_t* validity;    /* device ptr to Arrow validity bitmap or NULL */
+  int32_t dtype;              /* GPUQ_* */
+} gpuq_col;
+
+/* ---------------------------------------------------------------- */
+/* Synthetic data generation (bench only; bit-identical to the CPU  */
+/* oracle's splitmix64 generator — DESIGN.md §data).                 */
+/* ---------------------------------------------------------------- */
+int gpuq_gen_i64_range(void* stream, uint64_t seed, uint64_t start,
+                       int64_t nrows, uint64_t range, int64_t* out);
+int gpuq_gen_f64_unit(void* stream, uint64_t seed, uint64_t start,
+                      int64_t nrows, double* out);
+
+/* ---------------------------------------------------------------- */
+/* SORT — replaces SortExec (execution/SortExec.scala:75-126): the   */
+/* radix-eligible single-key path (canUseRadixSort, SortExec.scala   */
+/* :82-83) over an int64 or float64 key column. Produces the sorted  */
+/* row permutation; gather materializes payload columns.             */
+/* Key encoding on device matches PrefixComparators.java:66-83       */
+/* (double bijection) / SignedPrefixComparator (int64); sort is an   */
+/* LSB radix over 8-bit digits with the reference's skip-uniform-    */
+/* byte optimization (RadixSort.java:108-139) and is STABLE (ties    */
+/* keep row order, as the reference's (prefix,pointer) sort does).   */
+/* NULL keys are placed per nulls_first in input order               */
+/* (SortOrder.scala:35-45 defaults: asc=>first, desc=>last).         */
+/* ---------------------------------------------------------------- */
+
+/* workspace bytes for gpuq_sort_perm at nrows (nrows < 2^32) */
+int64_t gpuq_sort_workspace_bytes(int64_t nrows);
+
+/* Sort one key column; writes the sorted permutation (row ids into the
+ * input) to out_perm[nrows] (uint32) and, if out_keys != NULL, the sorted
+ * key column. key.validity may be NULL (no nulls). */
+int gpuq_sort_perm(void* stream, int64_t nrows, gpuq_col key,
+                   int32_t desc, int32_t nulls_first,
+                   uint32_t* out_perm, void* out_keys,
+                   void* workspace, int64_t workspace_bytes);
+
+/* out[i] = col[perm[i]] — payload materialization after sort/partition. */
+int gpuq_gather(void* stream, int64_t nrows, gpuq_col col,
+                const uint32_t* perm, void* out);
+
+/* ---------------------------------------------------------------- */
+/* HASH AGGREGATE — replaces HashAggregateExec                       */
+/* (execution/aggregate/HashAggregateExec.scala:99-151) for          */
+/* GROUP BY int64 key -> SUM(float64) / COUNT. Open-address table    */
+/* (the GPU analog of BytesToBytesMap, core/.../unsafe/map/          */
+/* BytesToBytesMap.java:50-56) keyed by Murmur3(key,42), linear      */
+/* probe, device-scope atomics. SUM accumulation order is device     */
+/* order => float64 parity within 1e-6 relative (north star); keys   */
+/* and COUNT bit-exact. NULL keys form one group; NULL values are    */
+/* skipped (Sum.scala:113-141).                                      */
+/* ---------------------------------------------------------------- */
+
+/* capacity must be a power of two >= 2 * expected distinct groups.   */
+int64_t gpuq_hash_agg_workspace_bytes(int64_t capacity);
+
+/* Aggregate nrows of (key,val) into the workspace table, then compact:
+ * writes ngroups rows of (key, key_valid, sum, sum_valid, count) into the
+ * caller's output arrays (each sized for max possible groups) and returns
+ * the group count via *out_ngroups. Emission order is nondeterministic
+ * (the parity checker is order-insensitive, as QueryTest.checkAnswer is).
+ * Multiple (key,val) batches can be accumulated before compaction:
+ * pass finalize=0 to accumulate only, finalize=1 to also compact. */
+int gpuq_hash_agg_i64_f64(void* stream, int64_t nrows,
+                          gpuq_col key, gpuq_col val,
+                          void* workspace, int64_t capacity, int32_t first_batch,
+                          int32_t finalize,
+                          int64_t* out_keys, uint8_t* out_key_valid,
+                          double* out_sums, uint8_t* out_sum_valid,
+                          int64_t* out_counts, int64_t* out_ngroups);
+
+/* ---------------------------------------------------------------- */
+/* PARTITION — replaces ShuffleExchangeExec's partition-id + write   */
+/* path (exchange/ShuffleExchangeExec.scala:357-470 +                */
+/* core/src/main/java/.../shuffle/sort/UnsafeShuffleWriter.java):    */
+/* pid = Pmod(Murmur3Hash(key,42), n) (partitioning.scala:328-330),  */
+/* stable radix-partition into per-partition contiguous runs. The    */
+/* cross-GPU exchange itself is RCCL all-to-all over xGMI, driven by */
+/* the host layer (torch.distributed / ncclGroup of send-recv) on    */
+/* the per-partition runs this produces.                             */
+/* ---------------------------------------------------------------- */
+
+int64_t gpuq_partition_workspace_bytes(int64_t nrows, int32_t num_parts);
+
+/* Computes the stable permutation that groups rows by partition id and the
+ * per-partition row counts. out_perm[nrows] (uint32), out_counts[num_parts]
+ * (int64, device). Gather columns with gpuq_gather afterwards.
+ * num_parts <= 65536. */
+int gpuq_partition_perm(void* stream, int64_t nrows, gpuq_col key,
+                        int32_t num_parts, uint32_t* out_perm,
+                        int64_t* out_counts,
+                        void* workspace, int64_t workspace_bytes);
+
+/* ---------------------------------------------------------------- */
+/* HASH JOIN — replaces ShuffledHashJoinExec inner join              */
+/* (joins/ShuffledHashJoinExec.scala:103-132; build side analog of   */
+/* LongHashedRelation, joins/HashedRelation.scala:993 — duplicates   */
+/* chained per key). NULL keys never match. Emission order is        */
+/* nondeterministic (checker sorts, as QueryTest does).              */
+/* ---------------------------------------------------------------- */
+
+int64_t gpuq_join_build_workspace_bytes(int64_t build_rows, int64_t capacity);
+
+/* Build the hash table over the build-side key column. capacity: power of
+ * two >= 2*build_rows. The workspace holds the table + chains and must stay
+ * alive through probes. */
+int gpuq_join_build_i64(void* stream, int64_t build_rows, gpuq_col build_key,
+                        void* workspace, int64_t capacity);
+
+/* Probe: emits matching (probe_rid, build_rid) uint32 pairs into the
+ * caller's buffers (out_cap entries each); *out_nmatches returns the total
+ * match count. If the count exceeds out_cap, returns GPUQ_ERR_OVERFLOW
+ * after setting *out_nmatches (call again with bigger buffers). */
+int gpuq_join_probe_i64(void* stream, int64_t probe_rows, gpuq_col probe_key,
+                        const void* workspace, int64_t capacity, int64_t build_rows,
+                        uint32_t* out_probe_rid, uint32_t* out_build_rid,
+                        int64_t out_cap, int64_t* out_nmatches);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* GPUQ_H */

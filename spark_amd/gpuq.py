@@ -1,0 +1,221 @@
+"""ctypes binding over libgpuq.so (include/gpuq.h) — the C-ABI boundary.
+
+Device memory, streams and the process-per-GPU launch come from PyTorch-ROCm
+(plumbing only); every compute kernel is hand-written HIP in
+spark_amd/csrc/gpuq.hip. There is NO CPU fallback here: if the extension is
+missing or a call fails, we raise — the product path must never silently run
+on an eager/CPU substitute (GPU parity would be meaningless).
+"""
+import ctypes
+import os
+
+import torch
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libgpuq.so")
+
+GPUQ_INT64 = 0
+GPUQ_FLOAT64 = 1
+GPUQ_INT32 = 2
+
+_DTYPE_MAP = {
+    torch.int64: GPUQ_INT64,
+    torch.float64: GPUQ_FLOAT64,
+    torch.int32: GPUQ_INT32,
+}
+
+
+class GpuqError(RuntimeError):
+    pass
+
+
+class _Col(ctypes.Structure):
+    _fields_ = [("data", ctypes.c_void_p),
+                ("validity", ctypes.c_void_p),
+                ("dtype", ctypes.c_int32)]
+
+
+_lib = None
+
+
+def lib() -> ctypes.CDLL:
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_SO):
+            raise GpuqError(
+                f"libgpuq.so not found at {_SO}. Build it with "
+                f"`make -C spark_amd/csrc` (or __graft_entry__.build()). "
+                f"There is no CPU fallback.")
+        L = ctypes.CDLL(_SO)
+        i32, i64, u64 = ctypes.c_int32, ctypes.c_int64, ctypes.c_uint64
+        vp = ctypes.c_void_p
+        L.gpuq_last_error.restype = ctypes.c_char_p
+        L.gpuq_device_count.restype = i32
+        L.gpuq_gen_i64_range.restype = i32
+        L.gpuq_gen_i64_range.argtypes = [vp, u64, u64, i64, u64, vp]
+        L.gpuq_gen_f64_unit.restype = i32
+        L.gpuq_gen_f64_unit.argtypes = [vp, u64, u64, i64, vp]
+        L.gpuq_sort_workspace_bytes.restype = i64
+        L.gpuq_sort_workspace_bytes.argtypes = [i64]
+        L.gpuq_sort_perm.restype = i32
+        L.gpuq_sort_perm.argtypes = [vp, i64, _Col, i32, i32, vp, vp, vp, i64]
+        L.gpuq_gather.restype = i32
+        L.gpuq_gather.argtypes = [vp, i64, _Col, vp, vp]
+        L.gpuq_hash_agg_workspace_bytes.restype = i64
+        L.gpuq_hash_agg_workspace_bytes.argtypes = [i64]
+        L.gpuq_hash_agg_i64_f64.restype = i32
+        L.gpuq_hash_agg_i64_f64.argtypes = [vp, i64, _Col, _Col, vp, i64, i32, i32,
+                                            vp, vp, vp, vp, vp, ctypes.POINTER(i64)]
+        L.gpuq_partition_workspace_bytes.restype = i64
+        L.gpuq_partition_workspace_bytes.argtypes = [i64, i32]
+        L.gpuq_partition_perm.restype = i32
+        L.gpuq_partition_perm.argtypes = [vp, i64, _Col, i32, vp, vp, vp, i64]
+        L.gpuq_join_build_workspace_bytes.restype = i64
+        L.gpuq_join_build_workspace_bytes.argtypes = [i64, i64]
+        L.gpuq_join_build_i64.restype = i32
+        L.gpuq_join_build_i64.argtypes = [vp, i64, _Col, vp, i64]
+        L.gpuq_join_probe_i64.restype = i32
+        L.gpuq_join_probe_i64.argtypes = [vp, i64, _Col, vp, i64, i64, vp, vp, i64,
+                                          ctypes.POINTER(i64)]
+        _lib = L
+    return _lib
+
+
+def _check(rc: int):
+    if rc != 0:
+        raise GpuqError(lib().gpuq_last_error().decode())
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _col(t: torch.Tensor, validity=None) -> _Col:
+    assert t.is_cuda and t.is_contiguous()
+    return _Col(t.data_ptr(), validity.data_ptr() if validity is not None else None,
+                _DTYPE_MAP[t.dtype])
+
+
+def _dp(t) -> int:
+    return t.data_ptr() if t is not None else None
+
+
+def device_count() -> int:
+    return lib().gpuq_device_count()
+
+
+def gen_i64(seed: int, n: int, range_: int = 0, start: int = 0,
+            device="cuda") -> torch.Tensor:
+    out = torch.empty(n, dtype=torch.int64, device=device)
+    _check(lib().gpuq_gen_i64_range(_stream(), seed, start, n, range_, out.data_ptr()))
+    return out
+
+
+def gen_f64_unit(seed: int, n: int, start: int = 0, device="cuda") -> torch.Tensor:
+    out = torch.empty(n, dtype=torch.float64, device=device)
+    _check(lib().gpuq_gen_f64_unit(_stream(), seed, start, n, out.data_ptr()))
+    return out
+
+
+def sort_workspace(n: int, device="cuda") -> torch.Tensor:
+    return torch.empty(lib().gpuq_sort_workspace_bytes(n), dtype=torch.uint8, device=device)
+
+
+def sort_perm(keys: torch.Tensor, desc=False, nulls_first=None, workspace=None,
+              out_keys: bool = True):
+    n = keys.numel()
+    if nulls_first is None:
+        nulls_first = not desc
+    if workspace is None:
+        workspace = sort_workspace(n, keys.device)
+    perm = torch.empty(n, dtype=torch.int32, device=keys.device)  # u32 bits
+    ok = torch.empty(n, dtype=keys.dtype, device=keys.device) if out_keys else None
+    _check(lib().gpuq_sort_perm(_stream(), n, _col(keys), int(desc), int(nulls_first),
+                                perm.data_ptr(), _dp(ok),
+                                workspace.data_ptr(), workspace.numel()))
+    return perm, ok
+
+
+def gather(col: torch.Tensor, perm: torch.Tensor) -> torch.Tensor:
+    n = perm.numel()
+    out = torch.empty(n, dtype=col.dtype, device=col.device)
+    _check(lib().gpuq_gather(_stream(), n, _col(col), perm.data_ptr(), out.data_ptr()))
+    return out
+
+
+def agg_workspace(capacity: int, device="cuda") -> torch.Tensor:
+    return torch.empty(lib().gpuq_hash_agg_workspace_bytes(capacity),
+                       dtype=torch.uint8, device=device)
+
+
+def hash_agg(keys: torch.Tensor, vals: torch.Tensor, capacity: int,
+             workspace=None, max_groups=None,
+             key_validity=None, val_validity=None):
+    """One-shot aggregate of a single batch. Returns (keys, key_valid, sums,
+    sum_valid, counts) tensors sliced to ngroups."""
+    n = keys.numel()
+    dev = keys.device
+    if workspace is None:
+        workspace = agg_workspace(capacity, dev)
+    mg = max_groups if max_groups is not None else min(n + 2, capacity + 2)
+    ok = torch.empty(mg, dtype=torch.int64, device=dev)
+    okv = torch.empty(mg, dtype=torch.uint8, device=dev)
+    osum = torch.empty(mg, dtype=torch.float64, device=dev)
+    osv = torch.empty(mg, dtype=torch.uint8, device=dev)
+    ocnt = torch.empty(mg, dtype=torch.int64, device=dev)
+    ng = ctypes.c_int64(0)
+    _check(lib().gpuq_hash_agg_i64_f64(
+        _stream(), n, _col(keys, key_validity), _col(vals, val_validity),
+        workspace.data_ptr(), capacity, 1, 1,
+        ok.data_ptr(), okv.data_ptr(), osum.data_ptr(), osv.data_ptr(),
+        ocnt.data_ptr(), ctypes.byref(ng)))
+    g = ng.value
+    return ok[:g], okv[:g], osum[:g], osv[:g], ocnt[:g]
+
+
+def partition_workspace(n: int, nparts: int, device="cuda") -> torch.Tensor:
+    return torch.empty(lib().gpuq_partition_workspace_bytes(n, nparts),
+                       dtype=torch.uint8, device=device)
+
+
+def partition_perm(keys: torch.Tensor, nparts: int, workspace=None, key_validity=None):
+    """Stable group-by-partition permutation + per-partition counts."""
+    n = keys.numel()
+    dev = keys.device
+    if workspace is None:
+        workspace = partition_workspace(n, nparts, dev)
+    perm = torch.empty(n, dtype=torch.int32, device=dev)
+    counts = torch.empty(nparts, dtype=torch.int64, device=dev)
+    _check(lib().gpuq_partition_perm(_stream(), n, _col(keys, key_validity), nparts,
+                                     perm.data_ptr(), counts.data_ptr(),
+                                     workspace.data_ptr(), workspace.numel()))
+    return perm, counts
+
+
+def join_build(build_keys: torch.Tensor, capacity: int, workspace=None,
+               key_validity=None):
+    bn = build_keys.numel()
+    dev = build_keys.device
+    if workspace is None:
+        workspace = torch.empty(lib().gpuq_join_build_workspace_bytes(bn, capacity),
+                                dtype=torch.uint8, device=dev)
+    _check(lib().gpuq_join_build_i64(_stream(), bn, _col(build_keys, key_validity),
+                                     workspace.data_ptr(), capacity))
+    return workspace
+
+
+def join_probe(probe_keys: torch.Tensor, workspace: torch.Tensor, capacity: int,
+               build_rows: int, out_cap: int, key_validity=None):
+    pn = probe_keys.numel()
+    dev = probe_keys.device
+    op = torch.empty(out_cap, dtype=torch.int32, device=dev)
+    ob = torch.empty(out_cap, dtype=torch.int32, device=dev)
+    nm = ctypes.c_int64(0)
+    rc = lib().gpuq_join_probe_i64(_stream(), pn, _col(probe_keys, key_validity),
+                                   workspace.data_ptr(), capacity, build_rows,
+                                   op.data_ptr(), ob.data_ptr(), out_cap,
+                                   ctypes.byref(nm))
+    if rc == 3:  # GPUQ_ERR_OVERFLOW: caller retries with nm.value capacity
+        return None, None, nm.value
+    _check(rc)
+    return op[:nm.value], ob[:nm.value], nm.value

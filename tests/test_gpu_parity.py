@@ -1,0 +1,211 @@
+"""GPU parity tests: the HIP engine (through the C-ABI) vs the CPU oracle on
+the same seeded inputs. Bit-exact for sort order (incl. stability), GROUP BY
+keys, COUNT, partition ids and join pairs; 1e-6 relative for float64 SUM
+(the north-star tolerance — GPU reduction order differs)."""
+import numpy as np
+import pytest
+
+import oracle
+
+torch = pytest.importorskip("torch")
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def gq():
+    from spark_amd import gpuq
+    assert torch.cuda.is_available()
+    return gpuq
+
+
+def to_dev(a: np.ndarray):
+    return torch.from_numpy(np.ascontiguousarray(a)).cuda()
+
+
+def pack_validity(valid_bool: np.ndarray):
+    return torch.from_numpy(np.packbits(valid_bool, bitorder="little")).cuda()
+
+
+# ---------- generator parity (device gen == oracle gen, bit-exact) ----------
+
+def test_gen_parity(gq):
+    n = 1 << 20
+    d = gq.gen_i64(seed=42, n=n, range_=10_000_000).cpu().numpy()
+    c = oracle.gen_i64(42, n, range_=10_000_000)
+    assert (d == c).all()
+    df = gq.gen_f64_unit(seed=7, n=n).cpu().numpy()
+    cf = oracle.gen_f64_unit(7, n)
+    assert (df == cf).all()
+
+
+# ---------- sort ----------
+
+@pytest.mark.parametrize("desc", [False, True])
+@pytest.mark.parametrize("n,rng", [(1000, 0), (100_000, 0), (1_000_000, 1000),
+                                   (4096 * 3 + 17, 50), (1, 0), (65, 2)])
+def test_sort_i64_parity(gq, desc, n, rng):
+    keys = oracle.gen_i64(seed=n + rng, n=n, range_=rng)
+    if rng == 0:
+        keys = keys  # full-range keys
+    perm, skeys = gq.sort_perm(to_dev(keys), desc=desc)
+    exp_perm = oracle.sort_perm(keys, desc=desc)
+    # STABLE sort => the whole permutation must match bit-exactly
+    assert (perm.cpu().numpy().astype(np.uint32) == exp_perm.astype(np.uint32)).all()
+    assert (skeys.cpu().numpy() == keys[exp_perm]).all()
+
+
+@pytest.mark.parametrize("desc", [False, True])
+def test_sort_f64_parity(gq, desc):
+    rng = np.random.default_rng(3)
+    n = 200_000
+    keys = rng.standard_normal(n)
+    # inject special values incl. -0.0 / NaN / inf ties
+    keys[:100] = np.repeat([0.0, -0.0, np.nan, np.inf, -np.inf], 20)
+    perm, skeys = gq.sort_perm(to_dev(keys), desc=desc)
+    exp_perm = oracle.sort_perm(keys, desc=desc)
+    assert (perm.cpu().numpy().astype(np.uint32) == exp_perm.astype(np.uint32)).all()
+    got = skeys.cpu().numpy()
+    exp = keys[exp_perm]
+    assert ((got == exp) | (np.isnan(got) & np.isnan(exp))).all()
+
+
+def test_sort_all_equal_keys(gq):
+    n = 10_000
+    keys = np.full(n, 7, dtype=np.int64)
+    perm, skeys = gq.sort_perm(to_dev(keys))
+    assert (perm.cpu().numpy() == np.arange(n)).all()  # zero passes, identity
+
+
+def test_sort_gather_payload(gq):
+    n = 300_000
+    keys = oracle.gen_i64(seed=1, n=n, range_=1000)
+    pay = oracle.gen_i64(seed=2, n=n)
+    perm, _ = gq.sort_perm(to_dev(keys))
+    out = gq.gather(to_dev(pay), perm)
+    exp = pay[oracle.sort_perm(keys)]
+    assert (out.cpu().numpy() == exp).all()
+
+
+# ---------- hash aggregate ----------
+
+def agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt, rtol=1e-6):
+    # order-insensitive compare (QueryTest.checkAnswer recipe): sort both by
+    # (key_valid, key)
+    g_order = np.lexsort((gk, gkv))
+    o_order = np.lexsort((ok, okv))
+    assert (gk[g_order] == ok[o_order]).all()
+    assert (gkv[g_order] == okv[o_order]).all()
+    assert (gc[g_order] == ocnt[o_order]).all()          # COUNT bit-exact
+    assert (gsv[g_order] == osv[o_order]).all()
+    np.testing.assert_allclose(gs[g_order], osum[o_order], rtol=rtol)
+
+
+@pytest.mark.parametrize("n,ngroups", [(1_000_000, 10_000), (100_000, 17),
+                                       (4097, 4097), (1, 1)])
+def test_agg_parity(gq, n, ngroups):
+    keys = oracle.gen_i64(seed=n, n=n, range_=ngroups)
+    vals = oracle.gen_f64_unit(seed=n + 1, n=n)
+    cap = 1 << max(4, int(np.ceil(np.log2(ngroups * 2 + 2))))
+    gk, gkv, gs, gsv, gc = (t.cpu().numpy() for t in
+                            gq.hash_agg(to_dev(keys), to_dev(vals), cap))
+    ok, okv, osum, osv, ocnt = oracle.hash_agg(keys, vals)
+    agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt)
+
+
+def test_agg_sentinel_and_null_keys(gq):
+    # keys hitting the EMPTY sentinel (-1), INT64_MIN, plus NULL keys and
+    # NULL values
+    keys = np.array([-1, 5, -1, -(2**63), 5, -1, 7], dtype=np.int64)
+    vals = np.array([1.0, 2.0, 3.0, 4.0, 5.0, 6.0, 7.0])
+    kvalid = np.array([1, 1, 1, 1, 0, 1, 1], dtype=np.uint8)  # row4 NULL key
+    vvalid = np.array([1, 1, 1, 1, 1, 0, 1], dtype=np.uint8)  # row5 NULL val
+    gk, gkv, gs, gsv, gc = (t.cpu().numpy() for t in gq.hash_agg(
+        to_dev(keys), to_dev(vals), 64,
+        key_validity=pack_validity(kvalid), val_validity=pack_validity(vvalid)))
+    ok, okv, osum, osv, ocnt = oracle.hash_agg(
+        keys, vals, np.packbits(kvalid, bitorder="little"),
+        np.packbits(vvalid, bitorder="little"))
+    agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt)
+
+
+def test_agg_overflow_detected(gq):
+    keys = oracle.gen_i64(seed=9, n=10_000, range_=0)  # ~10k distinct
+    vals = oracle.gen_f64_unit(seed=10, n=10_000)
+    from spark_amd.gpuq import GpuqError
+    with pytest.raises(GpuqError, match="overflow"):
+        gq.hash_agg(to_dev(keys), to_dev(vals), 4096)  # cap < distinct
+
+
+# ---------- partition ----------
+
+@pytest.mark.parametrize("nparts", [1, 8, 200, 256])
+def test_partition_parity(gq, nparts):
+    n = 500_000
+    keys = oracle.gen_i64(seed=77, n=n)
+    perm, counts = gq.partition_perm(to_dev(keys), nparts)
+    pids = oracle.partition_ids(keys, nparts)
+    exp_counts = np.bincount(pids, minlength=nparts)
+    assert (counts.cpu().numpy() == exp_counts).all()
+    exp_perm = np.argsort(pids, kind="stable")
+    assert (perm.cpu().numpy().astype(np.uint32) == exp_perm.astype(np.uint32)).all()
+
+
+def test_partition_null_keys(gq):
+    n = 1000
+    keys = oracle.gen_i64(seed=5, n=n)
+    valid = (oracle.gen_i64(seed=6, n=n, range_=4) != 0)
+    vbits = np.packbits(valid, bitorder="little")
+    perm, counts = gq.partition_perm(to_dev(keys), 8, key_validity=pack_validity(valid))
+    pids = oracle.partition_ids(keys, 8, validity=vbits)
+    assert (counts.cpu().numpy() == np.bincount(pids, minlength=8)).all()
+    assert (perm.cpu().numpy().astype(np.uint32)
+            == np.argsort(pids, kind="stable").astype(np.uint32)).all()
+
+
+# ---------- hash join ----------
+
+def join_compare(gp, gb, op, ob):
+    g = np.lexsort((gb, gp))
+    o = np.lexsort((ob, op))
+    assert len(gp) == len(op)
+    assert (gp[g] == op[o]).all() and (gb[g] == ob[o]).all()
+
+
+@pytest.mark.parametrize("bn,pn,rng", [(100_000, 150_000, 80_000),
+                                       (1000, 1000, 10), (1, 1, 1)])
+def test_join_parity(gq, bn, pn, rng):
+    bkeys = oracle.gen_i64(seed=bn, n=bn, range_=rng)
+    pkeys = oracle.gen_i64(seed=pn + 1, n=pn, range_=rng)
+    cap = 1 << int(np.ceil(np.log2(bn * 2 + 2)))
+    ws = gq.join_build(to_dev(bkeys), cap)
+    op_o, ob_o = oracle.join_inner(bkeys, pkeys)
+    out_cap = max(len(op_o) + 16, 16)
+    gp, gb, nm = gq.join_probe(to_dev(pkeys), ws, cap, bn, out_cap)
+    assert nm == len(op_o)
+    join_compare(gp.cpu().numpy().astype(np.uint32).astype(np.int64),
+                 gb.cpu().numpy().astype(np.uint32).astype(np.int64), op_o, ob_o)
+
+
+def test_join_sentinel_null_keys(gq):
+    bkeys = np.array([-1, 3, -1, 5], dtype=np.int64)
+    pkeys = np.array([-1, 5, 4, -1], dtype=np.int64)
+    bvalid = np.array([1, 1, 1, 0], dtype=np.uint8)   # build row3 NULL
+    pvalid = np.array([1, 0, 1, 1], dtype=np.uint8)   # probe row1 NULL
+    cap = 16
+    ws = gq.join_build(to_dev(bkeys), cap, key_validity=pack_validity(bvalid))
+    gp, gb, nm = gq.join_probe(to_dev(pkeys), ws, cap, 4, 64,
+                               key_validity=pack_validity(pvalid))
+    op_o, ob_o = oracle.join_inner(bkeys, pkeys,
+                                   np.packbits(bvalid, bitorder="little"),
+                                   np.packbits(pvalid, bitorder="little"))
+    assert nm == len(op_o)
+    join_compare(gp.cpu().numpy().astype(np.int64), gb.cpu().numpy().astype(np.int64),
+                 op_o, ob_o)
+
+
+def test_join_overflow_reports_count(gq):
+    bkeys = np.zeros(100, dtype=np.int64)
+    pkeys = np.zeros(100, dtype=np.int64)
+    ws = gq.join_build(to_dev(bkeys), 256)
+    gp, gb, nm = gq.join_probe(to_dev(pkeys), ws, 256, 100, out_cap=10)
+    assert gp is None and nm == 100 * 100
