@@ -1,0 +1,338 @@
+#!/usr/bin/env python3
+"""bench.py — measures the GPU engine on BASELINE.json's workloads.
+
+Primary line (N=1 default): config 2 — ORDER BY int64 key on 1B rows x 3 cols
+(int64 key, int64, float64) on one MI355X, whole-operator rows/s with inputs
+resident in HBM. Sub-benchmarks in the same JSON line: config 3 (GROUP BY
+1B rows / 10M groups, SUM float64) and config 4 scaled per GPU (shuffled hash
+join, 0.5B rows/side/GPU; at N>1 the real RCCL all-to-all exchange runs).
+
+N>1 (launched by the driver via torch.distributed.run): sort and agg run as
+independent per-rank batches (weak scaling — config 2/3 are single-partition
+operators; the cross-GPU exchange lives in the join workload); the join
+partitions + all-to-all exchanges + joins across all N ranks.
+
+Roofline: per-kernel HIP events (gpuq_profiling) on the launching stream give
+the dominant kernel's (radix_scatter) average launch time; achieved =
+algorithmic bytes/launch / time. PMC traffic comes from rocprofv3 runs
+committed under profiles/ (null here).
+
+CPU baseline: the C oracle (kind "port", single thread) timed on this box's
+host cores over a bounded sample of the same workload — a reported baseline,
+not the target.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+
+PEAK_HBM_GBPS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def env_rank():
+    return int(os.environ.get("RANK", "0")), int(os.environ.get("WORLD_SIZE", "1")), \
+        int(os.environ.get("LOCAL_RANK", "0"))
+
+
+def dist_setup():
+    rank, world, local = env_rank()
+    if world > 1:
+        import torch.distributed as dist
+        if not dist.is_initialized():
+            dist.init_process_group("nccl")  # nccl backend IS RCCL on ROCm
+        torch.cuda.set_device(local)
+    else:
+        torch.cuda.set_device(0)
+    return rank, world, local
+
+
+def barrier_sync(world):
+    torch.cuda.synchronize()
+    if world > 1:
+        import torch.distributed as dist
+        dist.barrier()
+        torch.cuda.synchronize()
+
+
+def max_over_ranks(x: float, world) -> float:
+    if world == 1:
+        return x
+    import torch.distributed as dist
+    t = torch.tensor([x], dtype=torch.float64, device="cuda")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
+
+
+# ---------------- workloads ----------------
+
+class SortWorkload:
+    """Config 2: ORDER BY int64 key, rows x (i64 key, i64, f64), 1 partition."""
+
+    def __init__(self, gq, rows, rank):
+        self.gq, self.rows = gq, rows
+        seed_off = rank * 3
+        self.keys = gq.gen_i64(seed=42 + seed_off, n=rows)          # full-range int64
+        self.pay1 = gq.gen_i64(seed=43 + seed_off, n=rows)
+        self.pay2 = gq.gen_f64_unit(seed=44 + seed_off, n=rows)
+        self.ws = gq.sort_workspace(rows)
+        self.out1 = torch.empty(rows, dtype=torch.int64, device="cuda")
+        self.out2 = torch.empty(rows, dtype=torch.float64, device="cuda")
+
+    def step(self):
+        gq = self.gq
+        perm, skeys = gq.sort_perm(self.keys, workspace=self.ws)
+        gq.lib().gpuq_gather(gq._stream(), self.rows, gq._col(self.pay1),
+                             perm.data_ptr(), self.out1.data_ptr())
+        gq.lib().gpuq_gather(gq._stream(), self.rows, gq._col(self.pay2),
+                             perm.data_ptr(), self.out2.data_ptr())
+
+    def free(self):
+        del self.keys, self.pay1, self.pay2, self.ws, self.out1, self.out2
+
+
+class AggWorkload:
+    """Config 3: GROUP BY int64 key (10M distinct), SUM(float64)."""
+
+    def __init__(self, gq, rows, groups, rank):
+        self.gq, self.rows = gq, rows
+        self.cap = 1 << max(4, (groups * 2 - 1).bit_length())
+        seed_off = rank * 3
+        self.keys = gq.gen_i64(seed=52 + seed_off, n=rows, range_=groups)
+        self.vals = gq.gen_f64_unit(seed=53 + seed_off, n=rows)
+        self.ws = gq.agg_workspace(self.cap)
+        mg = groups + 2
+        self.outs = [torch.empty(mg, dtype=d, device="cuda")
+                     for d in (torch.int64, torch.uint8, torch.float64,
+                               torch.uint8, torch.int64)]
+
+    def step(self):
+        import ctypes
+        gq = self.gq
+        ng = ctypes.c_int64(0)
+        rc = gq.lib().gpuq_hash_agg_i64_f64(
+            gq._stream(), self.rows, gq._col(self.keys), gq._col(self.vals),
+            self.ws.data_ptr(), self.cap, 1, 1,
+            *[t.data_ptr() for t in self.outs], ctypes.byref(ng))
+        gq._check(rc)
+        self.ngroups = ng.value
+
+    def free(self):
+        del self.keys, self.vals, self.ws, self.outs
+
+
+class JoinWorkload:
+    """Config 4 per-GPU slice: shuffled hash join of two tables,
+    rows_per_gpu per side per rank, int64 keys uniform over world*rows space
+    (~matching rate of the 4B x 4B config), one int64 payload each.
+    At world>1: radix-partition by Pmod(Murmur3,world) + RCCL all-to-all,
+    then local build+probe — the real exchange path."""
+
+    def __init__(self, gq, rows, rank, world):
+        self.gq, self.rows, self.world, self.rank = gq, rows, world, rank
+        keyspace = rows * world
+        self.bkeys = gq.gen_i64(seed=62, n=rows, range_=keyspace, start=rank * rows)
+        self.bpay = gq.gen_i64(seed=63, n=rows, range_=0, start=rank * rows)
+        self.pkeys = gq.gen_i64(seed=64, n=rows, range_=keyspace, start=rank * rows)
+        self.ppay = gq.gen_i64(seed=65, n=rows, range_=0, start=rank * rows)
+        self.part_ws = gq.partition_workspace(rows, max(world, 1))
+        # after exchange each rank holds ~rows per side (margin 1.25x)
+        self.local_cap_rows = int(rows * 1.25) + 4096
+        self.cap = 1 << max(4, (self.local_cap_rows * 2 - 1).bit_length())
+        self.join_ws = torch.empty(
+            gq.lib().gpuq_join_build_workspace_bytes(self.local_cap_rows, self.cap),
+            dtype=torch.uint8, device="cuda")
+        self.out_cap = int(rows * 2.5) + 4096
+        self.nmatches = 0
+
+    def _exchange(self, keys, pay):
+        """partition by murmur3 pmod(world) + all_to_all; returns local cols."""
+        gq = self.gq
+        if self.world == 1:
+            return keys, pay
+        import torch.distributed as dist
+        perm, counts = gq.partition_perm(keys, self.world, workspace=self.part_ws)
+        pk = gq.gather(keys, perm)
+        pp = gq.gather(pay, perm)
+        in_splits = counts.cpu().tolist()
+        out_counts = torch.empty(self.world, dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(out_counts, counts)
+        out_splits = out_counts.cpu().tolist()
+        total = sum(out_splits)
+        rk = torch.empty(total, dtype=torch.int64, device="cuda")
+        rp = torch.empty(total, dtype=torch.int64, device="cuda")
+        dist.all_to_all_single(rk, pk, out_splits, in_splits)
+        dist.all_to_all_single(rp, pp, out_splits, in_splits)
+        return rk, rp
+
+    def step(self):
+        gq = self.gq
+        bk, bp = self._exchange(self.bkeys, self.bpay)
+        pk, pp = self._exchange(self.pkeys, self.ppay)
+        bn = bk.numel()
+        gq.lib().gpuq_join_build_i64(gq._stream(), bn, gq._col(bk),
+                                     self.join_ws.data_ptr(), self.cap)
+        op, ob, nm = gq.join_probe(pk, self.join_ws, self.cap, bn, self.out_cap)
+        assert op is not None, f"join out_cap {self.out_cap} < {nm}"
+        # materialize one payload column per side (what SHJ emits)
+        gq.gather(bp, ob)
+        gq.gather(pp, op)
+        self.nmatches = nm
+
+    def free(self):
+        del self.bkeys, self.bpay, self.pkeys, self.ppay, self.part_ws, self.join_ws
+
+
+def time_workload(w, steps, warmup, world):
+    for _ in range(warmup):
+        w.step()
+    barrier_sync(world)
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        w.step()
+    barrier_sync(world)
+    dt = max_over_ranks(time.perf_counter() - t0, world)
+    return dt / steps
+
+
+def cpu_baseline_sort(sample_rows):
+    """Oracle (C restatement, 1 thread) on the same sort workload shape."""
+    import numpy as np
+    import oracle
+    keys = oracle.gen_i64(42, sample_rows)
+    pay1 = oracle.gen_i64(43, sample_rows)
+    pay2 = oracle.gen_f64_unit(44, sample_rows)
+    t0 = time.perf_counter()
+    perm = oracle.sort_perm(keys)
+    _o1 = pay1[perm]
+    _o2 = pay2[perm]
+    dt = time.perf_counter() - t0
+    return sample_rows / dt, dt
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--rows", type=int, default=1_000_000_000)
+    p.add_argument("--agg-rows", type=int, default=1_000_000_000)
+    p.add_argument("--agg-groups", type=int, default=10_000_000)
+    p.add_argument("--join-rows", type=int, default=500_000_000)
+    p.add_argument("--quick", action="store_true", help="small sizes (CI/sanity)")
+    p.add_argument("--no-cpu-baseline", action="store_true")
+    p.add_argument("--cpu-sample-rows", type=int, default=40_000_000)
+    p.add_argument("--workloads", default="sort,agg,join")
+    args = p.parse_args()
+    if args.quick:
+        args.rows, args.agg_rows, args.agg_groups, args.join_rows = \
+            20_000_000, 20_000_000, 200_000, 10_000_000
+        args.cpu_sample_rows = 2_000_000
+
+    rank, world, local = dist_setup()
+    from spark_amd import gpuq as gq
+    gq.profiling(True)
+    wl = args.workloads.split(",")
+    results = {}
+
+    if "sort" in wl:
+        w = SortWorkload(gq, args.rows, rank)
+        gq.kernel_stats_reset()
+        sec = time_workload(w, args.steps, args.warmup, world)
+        scat_ms, scat_n = gq.kernel_stats("radix_scatter")
+        hist_ms, hist_n = gq.kernel_stats("radix_hist")
+        w.free()
+        torch.cuda.empty_cache()
+        results["sort"] = {
+            "sec_per_step": sec,
+            "rows_per_sec": args.rows * world / sec,
+            "scatter_ms_avg": scat_ms / max(scat_n, 1),
+            "scatter_launches": scat_n,
+            "hist_ms_avg": hist_ms / max(hist_n, 1),
+        }
+
+    if "agg" in wl:
+        w = AggWorkload(gq, args.agg_rows, args.agg_groups, rank)
+        sec = time_workload(w, args.steps, args.warmup, world)
+        ng = w.ngroups
+        w.free()
+        torch.cuda.empty_cache()
+        results["agg"] = {"sec_per_step": sec,
+                          "rows_per_sec": args.agg_rows * world / sec,
+                          "ngroups": ng}
+
+    if "join" in wl:
+        w = JoinWorkload(gq, args.join_rows, rank, world)
+        sec = time_workload(w, args.steps, args.warmup, world)
+        nm = w.nmatches
+        w.free()
+        torch.cuda.empty_cache()
+        results["join"] = {
+            "sec_per_step": sec,
+            # rows processed = both tables, all ranks (the config's rate basis)
+            "rows_per_sec": 2 * args.join_rows * world / sec,
+            "matches_local": nm,
+            "exchange": "rccl_all_to_all" if world > 1 else "none (single GPU)",
+        }
+
+    if rank != 0:
+        return
+
+    primary = results.get("sort") or next(iter(results.values()))
+    # roofline of the dominant kernel (radix scatter): 12B read + 12B write
+    # per row per pass ((u64 encoded key, u32 rowid) pairs), algorithmic.
+    roofline = None
+    if "sort" in results:
+        r = results["sort"]
+        alg_bytes_per_launch = 24.0 * args.rows
+        ach = alg_bytes_per_launch / (r["scatter_ms_avg"] / 1e3) / 1e9 \
+            if r["scatter_ms_avg"] > 0 else 0.0
+        roofline = {"bound": "hbm", "achieved": round(ach, 1),
+                    "peak": PEAK_HBM_GBPS, "unit": "GB/s",
+                    "frac": round(ach / PEAK_HBM_GBPS, 4),
+                    "traffic": None,
+                    "kernel": "radix_scatter",
+                    "alg_bytes_per_launch": alg_bytes_per_launch}
+
+    cpu = None
+    if not args.no_cpu_baseline and world == 1 and "sort" in results:
+        rate, dt = cpu_baseline_sort(args.cpu_sample_rows)
+        cpu = {"value": round(rate, 1), "unit": "rows/s", "cores": 1,
+               "kind": "port",
+               "sample": f"oracle C sort of {args.cpu_sample_rows} rows "
+                         f"({dt:.1f}s, single thread)"}
+
+    out = {
+        "metric": "rows/sec (sort+hash-agg+join, 1B-row synth)",
+        "value": round(primary["rows_per_sec"], 1),
+        "unit": "rows/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(primary["sec_per_step"] * 1e3, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": {
+            "workload": "orderby-1b-int64x3 (BASELINE configs[1]); value = sort rows/s",
+            "rows": args.rows,
+            "cols": "int64 key + int64 + float64 payload",
+            "parallelism": ("independent per-GPU sorts (single-partition ORDER BY); "
+                            "join sub-benchmark exchanges via RCCL all-to-all"
+                            if world > 1 else "single GPU"),
+        },
+        "roofline": roofline,
+        "cpu_baseline": cpu,
+        "sub_benchmarks": results,
+    }
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

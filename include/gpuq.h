@@ -43,6 +43,15 @@ extern "C" {
 const char* gpuq_last_error(void);
 int gpuq_device_count(void);
 
+/* ---- optional per-kernel profiling (bench roofline evidence) ----
+ * When enabled, each kernel launch is bracketed by HIP events on its own
+ * stream; gpuq_kernel_stats synchronizes pending events and returns the
+ * accumulated GPU time and launch count for a kernel tag (e.g.
+ * "radix_scatter", "radix_hist", "agg_build", "join_probe"). */
+void gpuq_profiling(int enable);
+void gpuq_kernel_stats_reset(void);
+int gpuq_kernel_stats(const char* name, double* total_ms, long long* count);
+
 /* ---- dtypes ---- */
 #define GPUQ_INT64 0
 #define GPUQ_FLOAT64 1

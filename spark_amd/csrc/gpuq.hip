@@ -49,6 +49,59 @@ extern "C" int gpuq_device_count(void) {
   return n;
 }
 
+/* ---- optional per-kernel HIP-event profiling (bench roofline) ----
+ * Events are recorded on the launching stream around individual kernel
+ * launches; gpuq_kernel_stats() drains (synchronizes) and accumulates. */
+#include <vector>
+#include <map>
+#include <string>
+#include <mutex>
+
+static std::mutex g_prof_mu;
+static bool g_prof_on = false;
+struct prof_rec { const char* tag; hipEvent_t a, b; };
+static std::vector<prof_rec> g_prof_pending;
+static std::map<std::string, std::pair<double, long long>> g_prof_acc;
+
+extern "C" void gpuq_profiling(int enable) { g_prof_on = enable != 0; }
+
+static hipEvent_t prof_begin(hipStream_t s) {
+  if (!g_prof_on) return nullptr;
+  hipEvent_t e; (void)hipEventCreate(&e); (void)hipEventRecord(e, s);
+  return e;
+}
+static void prof_end(const char* tag, hipStream_t s, hipEvent_t a) {
+  if (!a) return;
+  hipEvent_t b; (void)hipEventCreate(&b); (void)hipEventRecord(b, s);
+  std::lock_guard<std::mutex> g(g_prof_mu);
+  g_prof_pending.push_back({tag, a, b});
+}
+
+extern "C" void gpuq_kernel_stats_reset(void) {
+  std::lock_guard<std::mutex> g(g_prof_mu);
+  for (auto& r : g_prof_pending) { (void)hipEventDestroy(r.a); (void)hipEventDestroy(r.b); }
+  g_prof_pending.clear();
+  g_prof_acc.clear();
+}
+
+extern "C" int gpuq_kernel_stats(const char* name, double* total_ms, long long* count) {
+  std::lock_guard<std::mutex> g(g_prof_mu);
+  for (auto& r : g_prof_pending) {
+    (void)hipEventSynchronize(r.b);
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, r.a, r.b);
+    auto& acc = g_prof_acc[r.tag];
+    acc.first += ms; acc.second += 1;
+    (void)hipEventDestroy(r.a); (void)hipEventDestroy(r.b);
+  }
+  g_prof_pending.clear();
+  auto it = g_prof_acc.find(name);
+  if (it == g_prof_acc.end()) { *total_ms = 0; *count = 0; return GPUQ_OK; }
+  *total_ms = it->second.first;
+  *count = it->second.second;
+  return GPUQ_OK;
+}
+
 /* ================= device inlines ================= */
 
 #define WAVE 64
@@ -532,12 +585,16 @@ extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
   for (int byte = 0; byte < 8; byte++) {
     if (((bits_changed >> (byte * 8)) & 0xff) == 0) continue;  /* RadixSort.java:126 skip */
     int shift = byte * 8;
+    { hipEvent_t _pe = prof_begin(s);
     k_radix_hist<<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(n, kin, shift, w.hist, (int)nb);
+    prof_end("radix_hist", s, _pe); }
     HIP_TRY(hipGetLastError());
     int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
     if (rc) return rc;
+    { hipEvent_t _pe = prof_begin(s);
     k_radix_scatter<0><<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(
         n, kin, iin, kout, iout, w.hist_scan, shift, (int)nb, 0);
+    prof_end("radix_scatter", s, _pe); }
     HIP_TRY(hipGetLastError());
     uint64_t* tk = kin; kin = kout; kout = tk;
     uint32_t* ti = iin; iin = iout; iout = ti;
@@ -607,17 +664,23 @@ extern "C" int gpuq_partition_perm(void* stream, int64_t n, gpuq_col key,
     FAIL(GPUQ_ERR_INVALID, "partition: workspace %lld < %lld", (long long)workspace_bytes, (long long)need);
   HIP_TRY(hipMemsetAsync(out_counts, 0, (size_t)nparts * 8, s));
   if (n == 0) return GPUQ_OK;
-  k_partition_pids<<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data, key.validity,
+  { hipEvent_t _pe = prof_begin(s);
+    k_partition_pids<<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data, key.validity,
                                              nparts, w.ka, w.ia,
                                              (unsigned long long*)out_counts);
+    prof_end("partition_pids", s, _pe); }
   HIP_TRY(hipGetLastError());
   int64_t nb = sort_nblocks(n);
-  k_radix_hist<<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(n, w.ka, 0, w.hist, (int)nb);
+  { hipEvent_t _pe = prof_begin(s);
+    k_radix_hist<<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(n, w.ka, 0, w.hist, (int)nb);
+    prof_end("radix_hist", s, _pe); }
   HIP_TRY(hipGetLastError());
   int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
   if (rc) return rc;
-  k_radix_scatter<0><<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(
+  { hipEvent_t _pe = prof_begin(s);
+    k_radix_scatter<0><<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(
       n, w.ka, w.ia, w.kb, w.ib, w.hist_scan, 0, (int)nb, 0);
+    prof_end("radix_scatter", s, _pe); }
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipMemcpyAsync(out_perm, w.ib, n * 4, hipMemcpyDeviceToDevice, s));
   return GPUQ_OK;
@@ -784,9 +847,11 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
     HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_special), s));
   }
   if (n > 0) {
+    { hipEvent_t _pe = prof_begin(s);
     k_agg_build<<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data, key.validity,
                                           (const double*)val.data, val.validity,
                                           w.keys, w.sums, w.cnts, w.sp, cap - 1);
+    prof_end("agg_build", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
   if (finalize) {
@@ -794,9 +859,11 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
     HIP_TRY(hipMemcpyAsync(&hsp, w.sp, sizeof(hsp), hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
     if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "agg: hash table overflow (capacity %lld)", (long long)cap);
+    { hipEvent_t _pe = prof_begin(s);
     k_agg_compact<<<grid1d(cap), 256, 0, s>>>(cap, w.keys, w.sums, w.cnts, w.sp,
                                               out_keys, out_key_valid, out_sums,
                                               out_sum_valid, out_counts);
+    prof_end("agg_compact", s, _pe); }
     HIP_TRY(hipGetLastError());
     agg_special hsp2;
     HIP_TRY(hipMemcpyAsync(&hsp2, w.sp, sizeof(hsp2), hipMemcpyDeviceToHost, s));
@@ -921,9 +988,11 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   HIP_TRY(hipMemsetAsync(w.sp, 0xFF, 4, s));             /* m1_head = NIL */
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
   if (brows > 0) {
+    { hipEvent_t _pe = prof_begin(s);
     k_join_build<<<grid1d(brows), 256, 0, s>>>(brows, (const int64_t*)bkey.data,
                                                bkey.validity, w.keys, w.heads,
                                                w.next, w.sp, cap - 1);
+    prof_end("join_build", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
   return GPUQ_OK;
@@ -939,10 +1008,12 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
   join_ws_layout(cap, brows, &w, (char*)workspace, &need);
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
   if (prows > 0) {
+    { hipEvent_t _pe = prof_begin(s);
     k_join_probe<<<grid1d(prows), 256, 0, s>>>(prows, (const int64_t*)pkey.data,
                                                pkey.validity, w.keys, w.heads,
                                                w.next, w.sp, cap - 1,
                                                out_p, out_b, out_cap);
+    prof_end("join_probe", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
   join_sp hsp;

@@ -77,8 +77,29 @@ def lib() -> ctypes.CDLL:
         L.gpuq_join_probe_i64.restype = i32
         L.gpuq_join_probe_i64.argtypes = [vp, i64, _Col, vp, i64, i64, vp, vp, i64,
                                           ctypes.POINTER(i64)]
+        L.gpuq_profiling.restype = None
+        L.gpuq_profiling.argtypes = [i32]
+        L.gpuq_kernel_stats_reset.restype = None
+        L.gpuq_kernel_stats.restype = i32
+        L.gpuq_kernel_stats.argtypes = [ctypes.c_char_p, ctypes.POINTER(ctypes.c_double),
+                                        ctypes.POINTER(ctypes.c_longlong)]
         _lib = L
     return _lib
+
+
+def profiling(enable: bool):
+    lib().gpuq_profiling(1 if enable else 0)
+
+
+def kernel_stats_reset():
+    lib().gpuq_kernel_stats_reset()
+
+
+def kernel_stats(name: str):
+    ms = ctypes.c_double(0)
+    cnt = ctypes.c_longlong(0)
+    lib().gpuq_kernel_stats(name.encode(), ctypes.byref(ms), ctypes.byref(cnt))
+    return ms.value, cnt.value
 
 
 def _check(rc: int):
