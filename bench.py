@@ -116,7 +116,7 @@ class AggWorkload:
         ng = ctypes.c_int64(0)
         rc = gq.lib().gpuq_hash_agg_i64_f64(
             gq._stream(), self.rows, gq._col(self.keys), gq._col(self.vals),
-            self.ws.data_ptr(), self.cap, 1, 1,
+            self.ws.data_ptr(), self.cap, 1, 1, 1,  # ops=SUM (config 3 shape)
             *[t.data_ptr() for t in self.outs], ctypes.byref(ng))
         gq._check(rc)
         self.ngroups = ng.value

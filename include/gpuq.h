@@ -118,17 +118,23 @@ int gpuq_gather(void* stream, int64_t nrows, gpuq_col col,
 /* capacity must be a power of two >= 2 * expected distinct groups.   */
 int64_t gpuq_hash_agg_workspace_bytes(int64_t capacity);
 
+/* aggregate ops bitmask */
+#define GPUQ_AGG_SUM 1
+#define GPUQ_AGG_COUNT 2
+
 /* Aggregate nrows of (key,val) into the workspace table, then compact:
  * writes ngroups rows of (key, key_valid, sum, sum_valid, count) into the
  * caller's output arrays (each sized for max possible groups) and returns
  * the group count via *out_ngroups. Emission order is nondeterministic
  * (the parity checker is order-insensitive, as QueryTest.checkAnswer is).
  * Multiple (key,val) batches can be accumulated before compaction:
- * pass finalize=0 to accumulate only, finalize=1 to also compact. */
+ * pass finalize=0 to accumulate only, finalize=1 to also compact.
+ * ops: GPUQ_AGG_* bitmask; SUM-only mode (no COUNT) requires non-null
+ * values (sum NULL-ness tracking needs COUNT) and allows out_counts=NULL. */
 int gpuq_hash_agg_i64_f64(void* stream, int64_t nrows,
                           gpuq_col key, gpuq_col val,
                           void* workspace, int64_t capacity, int32_t first_batch,
-                          int32_t finalize,
+                          int32_t finalize, int32_t ops,
                           int64_t* out_keys, uint8_t* out_key_valid,
                           double* out_sums, uint8_t* out_sum_valid,
                           int64_t* out_counts, int64_t* out_ngroups);

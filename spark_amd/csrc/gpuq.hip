@@ -441,10 +441,13 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
     }
     basecnt = __shfl(basecnt, leader);
     lrank[r] = (uint16_t)(basecnt + rank);
-    /* the round-r leader's LDS RMW of wave_hist[wave][bin] must be visible
-     * to round r+1's (possibly different) leader lane before it reads */
-    __syncthreads();
+    /* wave_hist rows are wave-private: the round-r leader's LDS RMW only
+     * needs to be ordered before round r+1's read within THIS wave. DS ops
+     * from one wave are serviced in issue order; the wave barrier stops the
+     * compiler from reordering them. */
+    __builtin_amdgcn_wave_barrier();
   }
+  __syncthreads();
 
   /* cross-wave exclusive prefix per bin + block-wide exclusive scan over bins.
    * thread t handles bin t (SORT_BLOCK == 256). */
@@ -705,10 +708,10 @@ struct agg_special {
   unsigned long long overflow;
 };
 
+/* Interleaved 24-byte slots [key, sum-bits, count] so one probe touches one
+ * cache line instead of three parallel arrays. */
 struct agg_ws {
-  unsigned long long* keys;
-  double* sums;
-  unsigned long long* cnts;
+  unsigned long long* tab;   /* 3 * cap u64 */
   agg_special* sp;
 };
 
@@ -719,9 +722,7 @@ static void agg_ws_layout(int64_t cap, agg_ws* w, char* base, int64_t* total) {
     off += (bytes + 255) & ~255LL;
     return p;
   };
-  w->keys = (unsigned long long*)take(cap * 8);
-  w->sums = (double*)take(cap * 8);
-  w->cnts = (unsigned long long*)take(cap * 8);
+  w->tab = (unsigned long long*)take(cap * 24);
   w->sp = (agg_special*)take(sizeof(agg_special));
   *total = off;
 }
@@ -732,10 +733,23 @@ extern "C" int64_t gpuq_hash_agg_workspace_bytes(int64_t cap) {
   return total;
 }
 
+__global__ void k_agg_init(int64_t cap, unsigned long long* tab) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < cap; i += stride) {
+    tab[3 * i] = AGG_EMPTY;
+    tab[3 * i + 1] = 0;
+    tab[3 * i + 2] = 0;
+  }
+}
+
+#define AGG_OP_SUM 1
+#define AGG_OP_COUNT 2
+
+template <int OPS>
 __global__ void k_agg_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                             const double* vals, const uint8_t* vvalid,
-                            unsigned long long* tkeys, double* tsums,
-                            unsigned long long* tcnts, agg_special* sp,
+                            unsigned long long* tab, agg_special* sp,
                             int64_t cap_mask) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -743,85 +757,80 @@ __global__ void k_agg_build(int64_t n, const int64_t* keys, const uint8_t* kvali
     bool kv = bit_valid(kvalid, i);
     bool vv = bit_valid(vvalid, i);
     double v = vv ? vals[i] : 0.0;
-    if (!kv) {
-      atomicMax(&sp->nul_seen, 1ull);
+    if (!kv || (unsigned long long)keys[i] == AGG_EMPTY) {
+      double* psum = kv ? &sp->m1_sum : &sp->nul_sum;
+      unsigned long long* pcnt = kv ? &sp->m1_cnt : &sp->nul_cnt;
+      unsigned long long* pseen = kv ? &sp->m1_seen : &sp->nul_seen;
+      atomicMax(pseen, 1ull);
       if (vv) {
-        atomicAdd(&sp->nul_sum, v);
-        atomicAdd(&sp->nul_cnt, 1ull);
+        if (OPS & AGG_OP_SUM) atomicAdd(psum, v);
+        if (OPS & AGG_OP_COUNT) atomicAdd(pcnt, 1ull);
       }
       continue;
     }
     int64_t k = keys[i];
-    if ((unsigned long long)k == AGG_EMPTY) {
-      atomicMax(&sp->m1_seen, 1ull);
-      if (vv) {
-        atomicAdd(&sp->m1_sum, v);
-        atomicAdd(&sp->m1_cnt, 1ull);
-      }
-      continue;
-    }
     uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
     for (int probes = 0;; probes++) {
-      unsigned long long cur = __hip_atomic_load(&tkeys[slot], __ATOMIC_RELAXED,
+      unsigned long long cur = __hip_atomic_load(&tab[3 * slot], __ATOMIC_RELAXED,
                                                  __HIP_MEMORY_SCOPE_AGENT);
       if (cur == (unsigned long long)k) break;
       if (cur == AGG_EMPTY) {
-        unsigned long long prev = atomicCAS(&tkeys[slot], AGG_EMPTY, (unsigned long long)k);
+        unsigned long long prev = atomicCAS(&tab[3 * slot], AGG_EMPTY, (unsigned long long)k);
         if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
       }
       slot = (slot + 1) & (uint64_t)cap_mask;
       if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
     }
     if (vv) {
-      atomicAdd(&tsums[slot], v);
-      atomicAdd(&tcnts[slot], 1ull);
-    } else {
-      /* group must exist even if value is NULL; counts stay 0 */
-      atomicMax(&tcnts[slot], 0ull);
+      if (OPS & AGG_OP_SUM) atomicAdd((double*)&tab[3 * slot + 1], v);
+      if (OPS & AGG_OP_COUNT) atomicAdd(&tab[3 * slot + 2], 1ull);
     }
   }
 }
 
-__global__ void k_agg_compact(int64_t cap, const unsigned long long* tkeys,
-                              const double* tsums, const unsigned long long* tcnts,
+template <int OPS>
+__global__ void k_agg_compact(int64_t cap, const unsigned long long* tab,
                               agg_special* sp,
                               int64_t* out_keys, uint8_t* out_kvalid,
                               double* out_sums, uint8_t* out_svalid,
                               int64_t* out_cnts) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < cap; i += stride) {
-    unsigned long long k = tkeys[i];
+  for (int64_t ib = (int64_t)blockIdx.x * blockDim.x; ib < cap;
+       ib += stride, i += stride) {
+    unsigned long long k = (i < cap) ? tab[3 * i] : AGG_EMPTY;
     bool occ = (k != AGG_EMPTY);
-    /* wave-aggregated cursor reservation (G12) */
     uint64_t mask = __ballot(occ);
     int rank = __popcll(mask & ((1ULL << (threadIdx.x & (WAVE - 1))) - 1));
     int leader = __ffsll((unsigned long long)mask) - 1;
     unsigned long long base = 0;
     if (occ && (int)(threadIdx.x & (WAVE - 1)) == leader)
       base = atomicAdd(&sp->out_cursor, (unsigned long long)__popcll(mask));
-    base = __shfl(base, leader);
+    if (mask) base = __shfl(base, leader);
     if (occ) {
       int64_t o = (int64_t)base + rank;
       out_keys[o] = (int64_t)k;
       out_kvalid[o] = 1;
-      out_sums[o] = tsums[i];
-      out_svalid[o] = tcnts[i] > 0 ? 1 : 0;
-      out_cnts[o] = (int64_t)tcnts[i];
+      out_sums[o] = __longlong_as_double((long long)tab[3 * i + 1]);
+      /* without COUNT, values were required non-null => every group has one */
+      out_svalid[o] = (OPS & AGG_OP_COUNT) ? (tab[3 * i + 2] > 0 ? 1 : 0) : 1;
+      if (out_cnts) out_cnts[o] = (int64_t)tab[3 * i + 2];
     }
   }
   if (blockIdx.x == 0 && threadIdx.x == 0) {
     if (sp->m1_seen) {
       int64_t o = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
       out_keys[o] = -1; out_kvalid[o] = 1;
-      out_sums[o] = sp->m1_sum; out_svalid[o] = sp->m1_cnt > 0 ? 1 : 0;
-      out_cnts[o] = (int64_t)sp->m1_cnt;
+      out_sums[o] = sp->m1_sum;
+      out_svalid[o] = (OPS & AGG_OP_COUNT) ? (sp->m1_cnt > 0 ? 1 : 0) : 1;
+      if (out_cnts) out_cnts[o] = (int64_t)sp->m1_cnt;
     }
     if (sp->nul_seen) {
       int64_t o = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
       out_keys[o] = 0; out_kvalid[o] = 0;
-      out_sums[o] = sp->nul_sum; out_svalid[o] = sp->nul_cnt > 0 ? 1 : 0;
-      out_cnts[o] = (int64_t)sp->nul_cnt;
+      out_sums[o] = sp->nul_sum;
+      out_svalid[o] = (OPS & AGG_OP_COUNT) ? (sp->nul_cnt > 0 ? 1 : 0) : 1;
+      if (out_cnts) out_cnts[o] = (int64_t)sp->nul_cnt;
     }
   }
 }
@@ -829,7 +838,7 @@ __global__ void k_agg_compact(int64_t cap, const unsigned long long* tkeys,
 extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
                                      gpuq_col key, gpuq_col val,
                                      void* workspace, int64_t cap, int32_t first_batch,
-                                     int32_t finalize,
+                                     int32_t finalize, int32_t ops,
                                      int64_t* out_keys, uint8_t* out_key_valid,
                                      double* out_sums, uint8_t* out_sum_valid,
                                      int64_t* out_counts, int64_t* out_ngroups) {
@@ -838,19 +847,26 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
     FAIL(GPUQ_ERR_INVALID, "agg: capacity %lld not a power of two", (long long)cap);
   if (key.dtype != GPUQ_INT64 || val.dtype != GPUQ_FLOAT64)
     FAIL(GPUQ_ERR_INVALID, "agg: expected int64 key + float64 val");
+  if (!(ops & AGG_OP_COUNT) && val.validity)
+    FAIL(GPUQ_ERR_INVALID, "agg: SUM-only mode requires non-null values "
+         "(sum NULL-ness needs COUNT)");
   agg_ws w; int64_t need;
   agg_ws_layout(cap, &w, (char*)workspace, &need);
   if (first_batch) {
-    HIP_TRY(hipMemsetAsync(w.keys, 0xFF, cap * 8, s));
-    HIP_TRY(hipMemsetAsync(w.sums, 0, cap * 8, s));
-    HIP_TRY(hipMemsetAsync(w.cnts, 0, cap * 8, s));
+    k_agg_init<<<grid1d(cap), 256, 0, s>>>(cap, w.tab);
+    HIP_TRY(hipGetLastError());
     HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_special), s));
   }
   if (n > 0) {
     { hipEvent_t _pe = prof_begin(s);
-    k_agg_build<<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data, key.validity,
-                                          (const double*)val.data, val.validity,
-                                          w.keys, w.sums, w.cnts, w.sp, cap - 1);
+    if (ops == AGG_OP_SUM)
+      k_agg_build<AGG_OP_SUM><<<grid1d(n), 256, 0, s>>>(
+          n, (const int64_t*)key.data, key.validity,
+          (const double*)val.data, val.validity, w.tab, w.sp, cap - 1);
+    else
+      k_agg_build<AGG_OP_SUM | AGG_OP_COUNT><<<grid1d(n), 256, 0, s>>>(
+          n, (const int64_t*)key.data, key.validity,
+          (const double*)val.data, val.validity, w.tab, w.sp, cap - 1);
     prof_end("agg_build", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
@@ -860,9 +876,14 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
     HIP_TRY(hipStreamSynchronize(s));
     if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "agg: hash table overflow (capacity %lld)", (long long)cap);
     { hipEvent_t _pe = prof_begin(s);
-    k_agg_compact<<<grid1d(cap), 256, 0, s>>>(cap, w.keys, w.sums, w.cnts, w.sp,
-                                              out_keys, out_key_valid, out_sums,
-                                              out_sum_valid, out_counts);
+    if (ops == AGG_OP_SUM)
+      k_agg_compact<AGG_OP_SUM><<<grid1d(cap), 256, 0, s>>>(
+          cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid,
+          out_counts);
+    else
+      k_agg_compact<AGG_OP_SUM | AGG_OP_COUNT><<<grid1d(cap), 256, 0, s>>>(
+          cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid,
+          out_counts);
     prof_end("agg_compact", s, _pe); }
     HIP_TRY(hipGetLastError());
     agg_special hsp2;
@@ -875,12 +896,13 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
 
 /* ================= hash join ================= */
 /*
- * Build: table of (key -> chain head); duplicates chained through next[]
- * (the GPU analog of LongToUnsafeRowMap's per-row next pointer,
- * HashedRelation.scala:536-600). EMPTY key sentinel -1 with a dedicated
- * chain for real -1 keys. NULL build/probe keys never match.
- * Workspace: [cap u64 keys][cap u32 heads][build_rows u32 next]
- *            [special: m1_head u32, cursor u64]
+ * Build: table of 16-byte interleaved slots {key u64, head u32, pad} (one
+ * cache line per probe); duplicates chained through next[] (the GPU analog
+ * of LongToUnsafeRowMap's per-row next pointer, HashedRelation.scala:536-600).
+ * EMPTY key sentinel -1 with a dedicated chain for real -1 keys. NULL keys
+ * never match. Probe reserves output space once per wave (aggregated
+ * atomic): count matches, wave-scan, single atomicAdd, emit.
+ * Workspace: [cap 16B slots][build_rows u32 next][special]
  */
 
 #define JOIN_NIL 0xFFFFFFFFu
@@ -888,8 +910,7 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
 struct join_sp { unsigned int m1_head; unsigned long long cursor; };
 
 struct join_ws {
-  unsigned long long* keys;
-  unsigned int* heads;
+  unsigned long long* slots;   /* 2 u64 per slot: [key][head|pad] */
   unsigned int* next;
   join_sp* sp;
 };
@@ -901,8 +922,7 @@ static void join_ws_layout(int64_t cap, int64_t brows, join_ws* w, char* base, i
     off += (bytes + 255) & ~255LL;
     return p;
   };
-  w->keys = (unsigned long long*)take(cap * 8);
-  w->heads = (unsigned int*)take(cap * 4);
+  w->slots = (unsigned long long*)take(cap * 16);
   w->next = (unsigned int*)take(brows * 4);
   w->sp = (join_sp*)take(sizeof(join_sp));
   *total = off;
@@ -915,8 +935,8 @@ extern "C" int64_t gpuq_join_build_workspace_bytes(int64_t brows, int64_t cap) {
 }
 
 __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
-                             unsigned long long* tkeys, unsigned int* heads,
-                             unsigned int* next, join_sp* sp, int64_t cap_mask) {
+                             unsigned long long* slots, unsigned int* next,
+                             join_sp* sp, int64_t cap_mask) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
@@ -929,46 +949,64 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
     }
     uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
     for (;;) {
-      unsigned long long cur = __hip_atomic_load(&tkeys[slot], __ATOMIC_RELAXED,
+      unsigned long long cur = __hip_atomic_load(&slots[2 * slot], __ATOMIC_RELAXED,
                                                  __HIP_MEMORY_SCOPE_AGENT);
       if (cur == (unsigned long long)k) break;
       if (cur == AGG_EMPTY) {
-        unsigned long long prev = atomicCAS(&tkeys[slot], AGG_EMPTY, (unsigned long long)k);
+        unsigned long long prev = atomicCAS(&slots[2 * slot], AGG_EMPTY, (unsigned long long)k);
         if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
       }
       slot = (slot + 1) & (uint64_t)cap_mask;
     }
-    unsigned int old = atomicExch(&heads[slot], (unsigned int)i);
+    unsigned int old = atomicExch((unsigned int*)&slots[2 * slot + 1], (unsigned int)i);
     next[i] = old;
   }
 }
 
 __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kvalid,
-                             const unsigned long long* tkeys, const unsigned int* heads,
-                             const unsigned int* next, join_sp* sp, int64_t cap_mask,
+                             const unsigned long long* slots, const unsigned int* next,
+                             join_sp* sp, int64_t cap_mask,
                              uint32_t* out_p, uint32_t* out_b, int64_t out_cap) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & (WAVE - 1);
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    if (!bit_valid(kvalid, i)) continue;
-    int64_t k = keys[i];
+  /* block-uniform loop bound so the whole wave stays converged for the
+   * aggregated cursor reservation */
+  for (int64_t ib = (int64_t)blockIdx.x * blockDim.x; ib < n; ib += stride) {
+    int64_t i = ib + threadIdx.x;
     unsigned int head = JOIN_NIL;
-    if ((unsigned long long)k == AGG_EMPTY) {
-      head = sp->m1_head;
-    } else {
-      uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
-      for (;;) {
-        unsigned long long cur = tkeys[slot];
-        if (cur == AGG_EMPTY) break;
-        if (cur == (unsigned long long)k) { head = heads[slot]; break; }
-        slot = (slot + 1) & (uint64_t)cap_mask;
+    if (i < n && bit_valid(kvalid, i)) {
+      int64_t k = keys[i];
+      if ((unsigned long long)k == AGG_EMPTY) {
+        head = sp->m1_head;
+      } else {
+        uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
+        for (;;) {
+          unsigned long long cur = slots[2 * slot];
+          if (cur == AGG_EMPTY) break;
+          if (cur == (unsigned long long)k) {
+            head = (unsigned int)slots[2 * slot + 1];
+            break;
+          }
+          slot = (slot + 1) & (uint64_t)cap_mask;
+        }
       }
     }
-    for (unsigned int b = head; b != JOIN_NIL; b = next[b]) {
-      unsigned long long pos = atomicAdd(&sp->cursor, 1ull);
-      if ((int64_t)pos < out_cap) {
-        out_p[pos] = (uint32_t)i;
-        out_b[pos] = b;
+    /* pass 1: count my matches */
+    uint32_t cnt = 0;
+    for (unsigned int b = head; b != JOIN_NIL; b = next[b]) cnt++;
+    /* wave-aggregated reservation: one atomic per wave */
+    uint32_t incl = wave_inclusive_scan(cnt);
+    uint32_t total = __shfl(incl, WAVE - 1);
+    unsigned long long base = 0;
+    if (lane == WAVE - 1 && total > 0)
+      base = atomicAdd(&sp->cursor, (unsigned long long)total);
+    base = __shfl(base, WAVE - 1);
+    /* pass 2: emit at base + my exclusive offset */
+    int64_t o = (int64_t)base + (incl - cnt);
+    for (unsigned int b = head; b != JOIN_NIL; b = next[b], o++) {
+      if (o < out_cap) {
+        out_p[o] = (uint32_t)i;
+        out_b[o] = b;
       }
     }
   }
@@ -983,15 +1021,14 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   if (bkey.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "join: key must be int64");
   join_ws w; int64_t need;
   join_ws_layout(cap, brows, &w, (char*)workspace, &need);
-  HIP_TRY(hipMemsetAsync(w.keys, 0xFF, cap * 8, s));
-  HIP_TRY(hipMemsetAsync(w.heads, 0xFF, cap * 4, s));
-  HIP_TRY(hipMemsetAsync(w.sp, 0xFF, 4, s));             /* m1_head = NIL */
+  HIP_TRY(hipMemsetAsync(w.slots, 0xFF, cap * 16, s));  /* keys=-1, heads=NIL */
+  HIP_TRY(hipMemsetAsync(w.sp, 0xFF, 4, s));            /* m1_head = NIL */
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
   if (brows > 0) {
     { hipEvent_t _pe = prof_begin(s);
     k_join_build<<<grid1d(brows), 256, 0, s>>>(brows, (const int64_t*)bkey.data,
-                                               bkey.validity, w.keys, w.heads,
-                                               w.next, w.sp, cap - 1);
+                                               bkey.validity, w.slots, w.next,
+                                               w.sp, cap - 1);
     prof_end("join_build", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
@@ -1010,8 +1047,8 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
   if (prows > 0) {
     { hipEvent_t _pe = prof_begin(s);
     k_join_probe<<<grid1d(prows), 256, 0, s>>>(prows, (const int64_t*)pkey.data,
-                                               pkey.validity, w.keys, w.heads,
-                                               w.next, w.sp, cap - 1,
+                                               pkey.validity, w.slots, w.next,
+                                               w.sp, cap - 1,
                                                out_p, out_b, out_cap);
     prof_end("join_probe", s, _pe); }
     HIP_TRY(hipGetLastError());

@@ -64,7 +64,7 @@ def lib() -> ctypes.CDLL:
         L.gpuq_hash_agg_workspace_bytes.restype = i64
         L.gpuq_hash_agg_workspace_bytes.argtypes = [i64]
         L.gpuq_hash_agg_i64_f64.restype = i32
-        L.gpuq_hash_agg_i64_f64.argtypes = [vp, i64, _Col, _Col, vp, i64, i32, i32,
+        L.gpuq_hash_agg_i64_f64.argtypes = [vp, i64, _Col, _Col, vp, i64, i32, i32, i32,
                                             vp, vp, vp, vp, vp, ctypes.POINTER(i64)]
         L.gpuq_partition_workspace_bytes.restype = i64
         L.gpuq_partition_workspace_bytes.argtypes = [i64, i32]
@@ -169,9 +169,13 @@ def agg_workspace(capacity: int, device="cuda") -> torch.Tensor:
                        dtype=torch.uint8, device=device)
 
 
+AGG_SUM = 1
+AGG_COUNT = 2
+
+
 def hash_agg(keys: torch.Tensor, vals: torch.Tensor, capacity: int,
              workspace=None, max_groups=None,
-             key_validity=None, val_validity=None):
+             key_validity=None, val_validity=None, ops=AGG_SUM | AGG_COUNT):
     """One-shot aggregate of a single batch. Returns (keys, key_valid, sums,
     sum_valid, counts) tensors sliced to ngroups."""
     n = keys.numel()
@@ -187,7 +191,7 @@ def hash_agg(keys: torch.Tensor, vals: torch.Tensor, capacity: int,
     ng = ctypes.c_int64(0)
     _check(lib().gpuq_hash_agg_i64_f64(
         _stream(), n, _col(keys, key_validity), _col(vals, val_validity),
-        workspace.data_ptr(), capacity, 1, 1,
+        workspace.data_ptr(), capacity, 1, 1, ops,
         ok.data_ptr(), okv.data_ptr(), osum.data_ptr(), osv.data_ptr(),
         ocnt.data_ptr(), ctypes.byref(ng)))
     g = ng.value

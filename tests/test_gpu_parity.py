@@ -128,6 +128,19 @@ def test_agg_sentinel_and_null_keys(gq):
     agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt)
 
 
+def test_agg_sum_only_mode(gq):
+    # ops=SUM (no COUNT atomic) — the config-3 shape; counts not compared
+    n, ngroups = 300_000, 1000
+    keys = oracle.gen_i64(seed=n, n=n, range_=ngroups)
+    vals = oracle.gen_f64_unit(seed=n + 1, n=n)
+    gk, gkv, gs, gsv, _ = (t.cpu().numpy() for t in
+                           gq.hash_agg(to_dev(keys), to_dev(vals), 4096, ops=gq.AGG_SUM))
+    ok, okv, osum, osv, ocnt = oracle.hash_agg(keys, vals)
+    g, o = np.argsort(gk), np.argsort(ok)
+    assert (gk[g] == ok[o]).all() and gsv.all()
+    np.testing.assert_allclose(gs[g], osum[o], rtol=1e-6)
+
+
 def test_agg_overflow_detected(gq):
     keys = oracle.gen_i64(seed=9, n=10_000, range_=0)  # ~10k distinct
     vals = oracle.gen_f64_unit(seed=10, n=10_000)
