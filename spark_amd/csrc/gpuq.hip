@@ -272,15 +272,18 @@ __global__ void k_encode(int64_t n, const void* keys, uint64_t* ek, uint32_t* id
   }
 }
 
-/* per-block histogram of digit at `shift` over the pass input */
+/* per-block histogram of digit at `shift` over the pass input;
+ * `tile` elements per block (must match the scatter geometry) */
 __global__ void k_radix_hist(int64_t n, const uint64_t* keys, int shift,
-                             uint32_t* hist /* [256][nblocks] */, int nblocks) {
+                             uint32_t* hist /* [256][nblocks] */, int nblocks,
+                             int tile) {
   __shared__ uint32_t h[256];
   if (threadIdx.x < 256) h[threadIdx.x] = 0;
   __syncthreads();
-  int64_t base = (int64_t)blockIdx.x * SORT_TILE;
-  for (int r = 0; r < SORT_ITEMS; r++) {
-    int64_t i = base + r * SORT_BLOCK + threadIdx.x;
+  int64_t base = (int64_t)blockIdx.x * tile;
+  int rounds = tile / 256;
+  for (int r = 0; r < rounds; r++) {
+    int64_t i = base + r * 256 + threadIdx.x;
     if (i < n) atomicAdd(&h[(keys[i] >> shift) & 0xff], 1u);
   }
   __syncthreads();
@@ -388,36 +391,39 @@ DEV int compute_bin(uint64_t key, int shift, int nparts) {
   return spark_pmod(mm3_hash_long(raw, 42), nparts);
 }
 
-template <int BIN_MODE>
-__global__ __launch_bounds__(SORT_BLOCK)
+template <int BIN_MODE, int BLOCK, int ITEMS>
+__global__ __launch_bounds__(BLOCK)
 void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
                      uint64_t* kout, uint32_t* iout,
                      const uint32_t* scanned /* [256][nblocks] exclusive */,
                      int shift, int nblocks, int nparts) {
-  __shared__ uint32_t wave_hist[SORT_WAVES][256];
+  constexpr int WAVES = BLOCK / WAVE;
+  constexpr int TILE = BLOCK * ITEMS;
+  __shared__ uint32_t wave_hist[WAVES][256];
   __shared__ uint32_t bin_start[256];     /* in-block exclusive start per bin */
   __shared__ uint32_t bin_gbase[256];     /* global dest minus local start    */
-  __shared__ uint64_t stage_k[SORT_TILE];
-  __shared__ uint32_t stage_i[SORT_TILE];
+  __shared__ uint32_t wtot[WAVES <= 4 ? 4 : WAVES];
+  __shared__ uint64_t stage_k[TILE];
+  __shared__ uint32_t stage_i[TILE];
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE, lane = tid & (WAVE - 1);
-  const int64_t base = (int64_t)blockIdx.x * SORT_TILE;
-  const int tile_n = (int)min((int64_t)SORT_TILE, n - base);
+  const int64_t base = (int64_t)blockIdx.x * TILE;
+  const int tile_n = (int)min((int64_t)TILE, n - base);
 
-  for (int b = tid; b < SORT_WAVES * 256; b += SORT_BLOCK)
+  for (int b = tid; b < WAVES * 256; b += BLOCK)
     ((uint32_t*)wave_hist)[b] = 0;
   __syncthreads();
 
   /* wave w owns the contiguous sub-tile [w*WAVE*ITEMS, ...): element order
    * within the block = (wave, round, lane) = linear tile order. */
-  uint64_t k[SORT_ITEMS];
-  uint32_t id[SORT_ITEMS];
-  uint16_t lrank[SORT_ITEMS];
-  uint8_t lbin[SORT_ITEMS];
+  uint64_t k[ITEMS];
+  uint32_t id[ITEMS];
+  uint16_t lrank[ITEMS];
+  uint8_t lbin[ITEMS];
 
-  const int64_t wbase = base + (int64_t)wave * WAVE * SORT_ITEMS;
-  for (int r = 0; r < SORT_ITEMS; r++) {
+  const int64_t wbase = base + (int64_t)wave * WAVE * ITEMS;
+  for (int r = 0; r < ITEMS; r++) {
     int64_t i = wbase + r * WAVE + lane;
     bool valid = i < n;
     k[r] = valid ? kin[i] : 0;
@@ -449,31 +455,33 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
   }
   __syncthreads();
 
-  /* cross-wave exclusive prefix per bin + block-wide exclusive scan over bins.
-   * thread t handles bin t (SORT_BLOCK == 256). */
-  {
+  /* cross-wave exclusive prefix per bin + block-wide exclusive scan over
+   * bins. The first 256 threads each own one bin. */
+  if (tid < 256) {
     int bin = tid;
     uint32_t acc = 0;
-    for (int w = 0; w < SORT_WAVES; w++) {
+    for (int w = 0; w < WAVES; w++) {
       uint32_t t = wave_hist[w][bin];
       wave_hist[w][bin] = acc;
       acc += t;
     }
-    /* exclusive scan of acc over the 256 bins (4 waves) */
     uint32_t inc = wave_inclusive_scan(acc);
-    __shared__ uint32_t wtot[SORT_WAVES];
     if (lane == WAVE - 1) wtot[wave] = inc;
-    __syncthreads();
+    bin_start[bin] = inc - acc;  /* provisional; add wave offsets after sync */
+  }
+  __syncthreads();
+  if (tid < 256) {
+    int bin = tid;
     uint32_t woff = 0;
     for (int w = 0; w < wave; w++) woff += wtot[w];
-    uint32_t excl = inc - acc + woff;
+    uint32_t excl = bin_start[bin] + woff;
     bin_start[bin] = excl;
     bin_gbase[bin] = scanned[(int64_t)bin * nblocks + blockIdx.x] - excl;
   }
   __syncthreads();
 
   /* stage reordered tile in LDS */
-  for (int r = 0; r < SORT_ITEMS; r++) {
+  for (int r = 0; r < ITEMS; r++) {
     int64_t i = wbase + r * WAVE + lane;
     if (i < n) {
       uint32_t pos = bin_start[lbin[r]] + wave_hist[wave][lbin[r]] + lrank[r];
@@ -484,8 +492,8 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
   __syncthreads();
 
   /* drain LDS linearly -> coalesced global runs per bin */
-  for (int r = 0; r < SORT_ITEMS; r++) {
-    int j = r * SORT_BLOCK + tid;
+  for (int r = 0; r < ITEMS; r++) {
+    int j = r * BLOCK + tid;
     if (j < tile_n) {
       uint64_t kk = stage_k[j];
       int bin = compute_bin<BIN_MODE>(kk, shift, nparts);
@@ -494,6 +502,44 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
       iout[dst] = stage_i[j];
     }
   }
+}
+
+/* runtime-selectable geometry (GPUQ_SORT_GEOM env: "BLOCKxITEMS") */
+struct scatter_geom { int block, items; };
+
+static scatter_geom get_sort_geom(void) {
+  static scatter_geom g = {0, 0};
+  if (g.block == 0) {
+    const char* e = getenv("GPUQ_SORT_GEOM");
+    int b = 0, it = 0;
+    if (e && sscanf(e, "%dx%d", &b, &it) == 2) { g.block = b; g.items = it; }
+    else { g.block = 1024; g.items = 4; }
+  }
+  return g;
+}
+
+template <int BIN_MODE>
+static void launch_scatter(hipStream_t s, scatter_geom g, int64_t nb,
+                           int64_t n, const uint64_t* kin, const uint32_t* iin,
+                           uint64_t* kout, uint32_t* iout,
+                           const uint32_t* scanned, int shift, int nparts) {
+  dim3 grid((uint32_t)nb);
+  if (g.block == 256 && g.items == 16)
+    k_radix_scatter<BIN_MODE, 256, 16><<<grid, 256, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+  else if (g.block == 512 && g.items == 8)
+    k_radix_scatter<BIN_MODE, 512, 8><<<grid, 512, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+  else if (g.block == 512 && g.items == 16)
+    k_radix_scatter<BIN_MODE, 512, 16><<<grid, 512, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+  else if (g.block == 1024 && g.items == 8)
+    k_radix_scatter<BIN_MODE, 1024, 8><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+  else if (g.block == 512 && g.items == 4)
+    k_radix_scatter<BIN_MODE, 512, 4><<<grid, 512, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+  else if (g.block == 1024 && g.items == 2)
+    k_radix_scatter<BIN_MODE, 1024, 2><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+  else if (g.block == 1024 && g.items == 6)
+    k_radix_scatter<BIN_MODE, 1024, 6><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+  else
+    k_radix_scatter<BIN_MODE, 1024, 4><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
 }
 
 /* decode sorted keys back to the output dtype */
@@ -521,10 +567,10 @@ struct sort_ws {
   unsigned long long* bits; /* [0]=and [1]=or */
 };
 
-static int64_t sort_nblocks(int64_t n) { return (n + SORT_TILE - 1) / SORT_TILE; }
+static int64_t sort_nblocks(int64_t n, int tile) { return (n + tile - 1) / tile; }
 
 static void sort_ws_layout(int64_t n, int nbins, sort_ws* w, char* basep, int64_t* total) {
-  int64_t nb = sort_nblocks(n);
+  int64_t nb = sort_nblocks(n, 1024);  /* sized for the smallest tile */
   int64_t hist_n = (int64_t)nbins * nb;
   int64_t scan_blocks = (hist_n + SCAN_TILE - 1) / SCAN_TILE + 1;
   int64_t off = 0;
@@ -582,21 +628,22 @@ extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
   HIP_TRY(hipStreamSynchronize(s));
   uint64_t bits_changed = hb[0] ^ hb[1];
 
-  int64_t nb = sort_nblocks(n);
+  scatter_geom geom = get_sort_geom();
+  int tile = geom.block * geom.items;
+  int64_t nb = sort_nblocks(n, tile);
   uint64_t *kin = w.ka, *kout = w.kb;
   uint32_t *iin = w.ia, *iout = w.ib;
   for (int byte = 0; byte < 8; byte++) {
     if (((bits_changed >> (byte * 8)) & 0xff) == 0) continue;  /* RadixSort.java:126 skip */
     int shift = byte * 8;
     { hipEvent_t _pe = prof_begin(s);
-    k_radix_hist<<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(n, kin, shift, w.hist, (int)nb);
+    k_radix_hist<<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, shift, w.hist, (int)nb, tile);
     prof_end("radix_hist", s, _pe); }
     HIP_TRY(hipGetLastError());
     int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
     if (rc) return rc;
     { hipEvent_t _pe = prof_begin(s);
-    k_radix_scatter<0><<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(
-        n, kin, iin, kout, iout, w.hist_scan, shift, (int)nb, 0);
+    launch_scatter<0>(s, geom, nb, n, kin, iin, kout, iout, w.hist_scan, shift, 0);
     prof_end("radix_scatter", s, _pe); }
     HIP_TRY(hipGetLastError());
     uint64_t* tk = kin; kin = kout; kout = tk;
@@ -673,16 +720,17 @@ extern "C" int gpuq_partition_perm(void* stream, int64_t n, gpuq_col key,
                                              (unsigned long long*)out_counts);
     prof_end("partition_pids", s, _pe); }
   HIP_TRY(hipGetLastError());
-  int64_t nb = sort_nblocks(n);
+  scatter_geom geom = get_sort_geom();
+  int tile = geom.block * geom.items;
+  int64_t nb = sort_nblocks(n, tile);
   { hipEvent_t _pe = prof_begin(s);
-    k_radix_hist<<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(n, w.ka, 0, w.hist, (int)nb);
+    k_radix_hist<<<dim3((uint32_t)nb), 256, 0, s>>>(n, w.ka, 0, w.hist, (int)nb, tile);
     prof_end("radix_hist", s, _pe); }
   HIP_TRY(hipGetLastError());
   int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
   if (rc) return rc;
   { hipEvent_t _pe = prof_begin(s);
-    k_radix_scatter<0><<<dim3((uint32_t)nb), SORT_BLOCK, 0, s>>>(
-      n, w.ka, w.ia, w.kb, w.ib, w.hist_scan, 0, (int)nb, 0);
+    launch_scatter<0>(s, geom, nb, n, w.ka, w.ia, w.kb, w.ib, w.hist_scan, 0, 0);
     prof_end("radix_scatter", s, _pe); }
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipMemcpyAsync(out_perm, w.ib, n * 4, hipMemcpyDeviceToDevice, s));
