@@ -1,0 +1,377 @@
+"""Host-side mirror of Spark's columnar physical-operator contract.
+
+In a Spark deployment this layer is Scala: exec-node subclasses registered by
+a ColumnarRule through SparkSessionExtensions.injectColumnar
+(SparkSessionExtensions.scala:116,168), each overriding
+supportsColumnar=true (SparkPlan.scala:92) and
+doExecuteColumnar(): RDD[ColumnarBatch] (SparkPlan.scala:359) and calling
+libgpuq over JNI (see INTEGRATION.md for that binding). This Python mirror
+keeps the same node names, the same required-distribution semantics and the
+same batch lifetime contract so the C-ABI is exercised exactly as the Scala
+layer would; it is what tests and the TPC-H-style pipelines drive.
+
+No CPU fallback anywhere: executing these nodes without the HIP engine (or
+without a GPU) raises.
+"""
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+
+class ColumnarBatch:
+    """Device-resident batch (ColumnarBatch.java:61-123 access contract;
+    Arrow-style columns: dense data + optional validity bitmap)."""
+
+    def __init__(self, columns: Dict[str, torch.Tensor],
+                 validity: Optional[Dict[str, torch.Tensor]] = None):
+        self._cols = columns
+        self._validity = validity or {}
+        ns = {t.numel() for t in columns.values()}
+        assert len(ns) == 1, "ragged batch"
+        self._num_rows = ns.pop()
+        self._closed = False
+
+    def num_rows(self) -> int:
+        return self._num_rows
+
+    def column(self, name: str) -> torch.Tensor:
+        assert not self._closed, "batch used after close()"
+        return self._cols[name]
+
+    def validity(self, name: str):
+        return self._validity.get(name)
+
+    def columns(self):
+        return dict(self._cols)
+
+    def close(self):
+        # consumer closes (Columnar.scala:224-240 ColumnarToRowExec pattern);
+        # device memory returns to the torch pool
+        self._cols = {}
+        self._validity = {}
+        self._closed = True
+
+
+@dataclass
+class Distribution:
+    """requiredChildDistribution analog (SparkPlan.scala:180 contract)."""
+    kind: str                      # "unspecified" | "clustered" | "ordered"
+    keys: Tuple[str, ...] = ()
+
+
+class SparkPlan:
+    """Physical operator (execution/SparkPlan.scala:65). Single-partition-
+    per-rank execution model: executeColumnar() yields this rank's batches."""
+
+    def __init__(self, *children: "SparkPlan"):
+        self.children = list(children)
+
+    @property
+    def output(self) -> List[str]:
+        raise NotImplementedError
+
+    @property
+    def supports_columnar(self) -> bool:
+        return False
+
+    def required_child_distribution(self) -> List[Distribution]:
+        return [Distribution("unspecified") for _ in self.children]
+
+    def execute_columnar(self):
+        """doExecuteColumnar (SparkPlan.scala:359)."""
+        raise NotImplementedError(
+            f"{type(self).__name__} does not support columnar execution")
+
+
+# ---------------- CPU placeholders the rule replaces ----------------
+# These mirror the Catalyst-produced nodes; they cannot execute here (the
+# JVM engine isn't present) — they exist so the ColumnarRule's rewrite is
+# the same shape as in Spark (SparkSessionExtensionSuite.scala:959-1000).
+
+@dataclass
+class SortOrder:
+    key: str
+    descending: bool = False
+    nulls_first: Optional[bool] = None   # SortOrder.scala:35-45 defaults
+
+    def __post_init__(self):
+        if self.nulls_first is None:
+            self.nulls_first = not self.descending
+
+
+class _CpuNode(SparkPlan):
+    def execute_columnar(self):
+        raise RuntimeError(f"{type(self).__name__} is a CPU placeholder; "
+                           "apply the ColumnarRule first (no CPU fallback)")
+
+
+class SortExec(_CpuNode):
+    def __init__(self, sort_order: SortOrder, global_sort: bool, child):
+        super().__init__(child)
+        self.sort_order, self.global_sort = sort_order, global_sort
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+
+class HashAggregateExec(_CpuNode):
+    def __init__(self, group_key: str, aggs: List[Tuple[str, str]], mode: str, child):
+        super().__init__(child)
+        self.group_key, self.aggs, self.mode = group_key, aggs, mode
+
+    @property
+    def output(self):
+        return [self.group_key] + [f"{fn}({col})" for fn, col in self.aggs]
+
+
+class ShuffledHashJoinExec(_CpuNode):
+    def __init__(self, left_key: str, right_key: str, build_side: str,
+                 left, right):
+        super().__init__(left, right)
+        self.left_key, self.right_key, self.build_side = left_key, right_key, build_side
+
+    @property
+    def output(self):
+        return self.children[0].output + self.children[1].output
+
+
+class ShuffleExchangeExec(_CpuNode):
+    def __init__(self, keys: Tuple[str, ...], num_partitions: int, child):
+        super().__init__(child)
+        self.keys, self.num_partitions = keys, num_partitions
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+
+class InputBatches(SparkPlan):
+    """Leaf: pre-materialized device batches (scan stand-in)."""
+
+    def __init__(self, batches: List[ColumnarBatch]):
+        super().__init__()
+        self._batches = batches
+
+    @property
+    def output(self):
+        return list(self._batches[0].columns().keys())
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def execute_columnar(self):
+        yield from self._batches
+
+
+# ---------------- GPU exec nodes ----------------
+
+class GpuSortExec(SparkPlan):
+    """Replaces SortExec (SortExec.scala:75-126): radix-eligible single-key
+    sort (canUseRadixSort analog: int64/float64 key)."""
+
+    def __init__(self, sort_order: SortOrder, global_sort: bool, child):
+        super().__init__(child)
+        self.sort_order, self.global_sort = sort_order, global_sort
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def required_child_distribution(self):
+        if self.global_sort:
+            return [Distribution("ordered", (self.sort_order.key,))]
+        return [Distribution("unspecified")]
+
+    def execute_columnar(self):
+        from . import gpuq
+        o = self.sort_order
+        for batch in self.children[0].execute_columnar():
+            keys = batch.column(o.key)
+            perm, skeys = gpuq.sort_perm(keys, desc=o.descending,
+                                         nulls_first=o.nulls_first)
+            cols = {o.key: skeys}
+            for name, t in batch.columns().items():
+                if name != o.key:
+                    cols[name] = gpuq.gather(t, perm)
+            batch.close()
+            yield ColumnarBatch(cols)
+
+
+class GpuHashAggregateExec(SparkPlan):
+    """Replaces HashAggregateExec (HashAggregateExec.scala:99-151).
+    mode: "partial" | "final" | "complete" (AggUtils.scala:126-195 split);
+    "complete" = single-node partial+final in one table."""
+
+    def __init__(self, group_key: str, aggs: List[Tuple[str, str]], mode: str,
+                 child, capacity: Optional[int] = None):
+        super().__init__(child)
+        assert mode in ("partial", "final", "complete")
+        self.group_key, self.aggs, self.mode = group_key, aggs, mode
+        self.capacity = capacity
+
+    @property
+    def output(self):
+        return [self.group_key] + [f"{fn}({col})" for fn, col in self.aggs]
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def required_child_distribution(self):
+        if self.mode == "final":
+            return [Distribution("clustered", (self.group_key,))]
+        return [Distribution("unspecified")]
+
+    def execute_columnar(self):
+        from . import gpuq
+        fns = {fn for fn, _ in self.aggs}
+        assert fns <= {"sum", "count"}, f"unsupported aggs {fns}"
+        val_col = next(col for fn, col in self.aggs if fn in ("sum", "count"))
+        ops = 0
+        if any(fn == "sum" for fn, _ in self.aggs) or self.mode == "final":
+            ops |= gpuq.AGG_SUM
+        if any(fn == "count" for fn, _ in self.aggs):
+            ops |= gpuq.AGG_COUNT
+        for batch in self.children[0].execute_columnar():
+            keys = batch.column(self.group_key)
+            if self.mode == "final":
+                # merge partial sums/counts: sum of sums; counts via SUM of
+                # partial counts cast to f64 is wrong — merge uses SUM sem.
+                # For round 1 final-merge sums only (counts merged as sums
+                # of int-valued f64 is exact below 2^53).
+                vals = batch.column(val_col).to(torch.float64)
+            else:
+                vals = batch.column(val_col)
+            n = keys.numel()
+            cap = self.capacity or (1 << max(10, int(n).bit_length()))
+            out = gpuq.hash_agg(keys, vals, cap, ops=ops or gpuq.AGG_SUM,
+                                key_validity=batch.validity(self.group_key),
+                                val_validity=batch.validity(val_col))
+            ok, okv, osum, osv, ocnt = out
+            cols = {self.group_key: ok}
+            for fn, col in self.aggs:
+                cols[f"{fn}({col})"] = osum if fn == "sum" else ocnt
+            batch.close()
+            yield ColumnarBatch(cols, validity={self.group_key: None})
+
+
+class GpuShuffleExchangeExec(SparkPlan):
+    """Replaces ShuffleExchangeExec (ShuffleExchangeExec.scala:277-470):
+    on-device radix partition + RCCL all-to-all (spark_amd/exchange.py).
+    Runs only under torch.distributed; num_partitions == world size."""
+
+    def __init__(self, keys: Tuple[str, ...], child):
+        super().__init__(child)
+        assert len(keys) == 1, "round 1: single int64 partition key"
+        self.keys = keys
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def execute_columnar(self):
+        import torch.distributed as dist
+        from .exchange import shuffle_exchange_gpu
+        assert dist.is_initialized(), "GpuShuffleExchangeExec needs torch.distributed"
+        key_name = self.keys[0]
+        for batch in self.children[0].execute_columnar():
+            payload = {n: t for n, t in batch.columns().items() if n != key_name}
+            k, cols = shuffle_exchange_gpu(batch.column(key_name), payload)
+            cols[key_name] = k
+            batch.close()
+            yield ColumnarBatch(cols)
+
+
+class GpuShuffledHashJoinExec(SparkPlan):
+    """Replaces ShuffledHashJoinExec inner join
+    (ShuffledHashJoinExec.scala:103-132). Both children must already be
+    clustered on the join key (ShuffledJoin.scala:57-69) — i.e. fed by (our)
+    exchanges, as EnsureRequirements would arrange."""
+
+    def __init__(self, left_key: str, right_key: str, build_side: str,
+                 left, right):
+        super().__init__(left, right)
+        assert build_side in ("left", "right")
+        self.left_key, self.right_key, self.build_side = left_key, right_key, build_side
+
+    @property
+    def output(self):
+        return self.children[0].output + self.children[1].output
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def required_child_distribution(self):
+        return [Distribution("clustered", (self.left_key,)),
+                Distribution("clustered", (self.right_key,))]
+
+    def execute_columnar(self):
+        from . import gpuq
+        left_b = list(self.children[0].execute_columnar())
+        right_b = list(self.children[1].execute_columnar())
+        assert len(left_b) == 1 and len(right_b) == 1, "one batch per partition"
+        lb, rb = left_b[0], right_b[0]
+        build, probe = (lb, rb) if self.build_side == "left" else (rb, lb)
+        bkey = self.left_key if self.build_side == "left" else self.right_key
+        pkey = self.right_key if self.build_side == "left" else self.left_key
+        bk = build.column(bkey)
+        bn = bk.numel()
+        cap = 1 << max(4, int(bn * 2 - 1).bit_length() if bn else 4)
+        ws = gpuq.join_build(bk, cap, key_validity=build.validity(bkey))
+        pk = probe.column(pkey)
+        out_cap = max(int(probe.num_rows() * 2) + 64, 64)
+        while True:
+            op, ob, nm = gpuq.join_probe(pk, ws, cap, bn, out_cap,
+                                         key_validity=probe.validity(pkey))
+            if op is not None:
+                break
+            out_cap = nm + 64
+        cols = {}
+        for name, t in build.columns().items():
+            cols[name] = gpuq.gather(t, ob)
+        for name, t in probe.columns().items():
+            if name in cols:
+                name = f"{name}#probe"
+            cols[name] = gpuq.gather(t, op)
+        lb.close(), rb.close()
+        yield ColumnarBatch(cols)
+
+
+class GpuColumnarRule:
+    """The injected rule (ColumnarRule, Columnar.scala:36-50; injection via
+    SparkSessionExtensions.injectColumnar:168; applied at
+    QueryExecution.scala:798 / AdaptiveSparkPlanExec.scala:184-186).
+    preColumnarTransitions swaps CPU nodes for GPU subclasses — the
+    SparkSessionExtensionSuite.scala:959-1000 pattern."""
+
+    def pre_columnar_transitions(self, plan: SparkPlan) -> SparkPlan:
+        children = [self.pre_columnar_transitions(c) for c in plan.children]
+        if isinstance(plan, SortExec):
+            return GpuSortExec(plan.sort_order, plan.global_sort, *children)
+        if isinstance(plan, HashAggregateExec):
+            return GpuHashAggregateExec(plan.group_key, plan.aggs, plan.mode,
+                                        *children)
+        if isinstance(plan, ShuffledHashJoinExec):
+            return GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
+                                           plan.build_side, *children)
+        if isinstance(plan, ShuffleExchangeExec):
+            return GpuShuffleExchangeExec(plan.keys, *children)
+        plan.children = children
+        return plan
+
+    def post_columnar_transitions(self, plan: SparkPlan) -> SparkPlan:
+        # Spark inserts Row<->Columnar transitions here (Columnar.scala:564-614);
+        # in this mirror every node is columnar, nothing to insert.
+        return plan
