@@ -41,13 +41,13 @@ def env_rank():
 
 def dist_setup():
     rank, world, local = env_rank()
+    torch.cuda.set_device(local if world > 1 else 0)
     if world > 1:
         import torch.distributed as dist
         if not dist.is_initialized():
-            dist.init_process_group("nccl")  # nccl backend IS RCCL on ROCm
-        torch.cuda.set_device(local)
-    else:
-        torch.cuda.set_device(0)
+            # nccl backend IS RCCL on ROCm; device set first so the
+            # communicator binds this rank's GPU
+            dist.init_process_group("nccl", device_id=torch.device("cuda", local))
     return rank, world, local
 
 
