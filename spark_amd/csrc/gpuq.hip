@@ -246,9 +246,17 @@ extern "C" int gpuq_gather(void* stream, int64_t n, gpuq_col col,
 
 /* encode kernel: keys -> radix-encoded u64 + identity rowids + bitwise
  * AND/OR reduction for the skip-uniform-byte decision (RadixSort.java:113-124) */
+/* encode + identity rowids + skip-byte AND/OR reduction + ALL EIGHT global
+ * byte histograms in one read (one pass replaces the reference's per-byte
+ * count loops, RadixSort.java:126-135; per-byte histograms are invariant
+ * under the passes' permutations, so one upfront count serves every pass). */
 template <int DTYPE, bool DESC>
 __global__ void k_encode(int64_t n, const void* keys, uint64_t* ek, uint32_t* idx,
-                         unsigned long long* bits_and, unsigned long long* bits_or) {
+                         unsigned long long* bits_and, unsigned long long* bits_or,
+                         uint32_t* ghist /* [8][256] */) {
+  __shared__ uint32_t h[8][256];
+  for (int b = threadIdx.x; b < 8 * 256; b += blockDim.x) ((uint32_t*)h)[b] = 0;
+  __syncthreads();
   uint64_t acc_or = 0, acc_and = ~0ULL;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -260,6 +268,8 @@ __global__ void k_encode(int64_t n, const void* keys, uint64_t* ek, uint32_t* id
     ek[i] = e;
     idx[i] = (uint32_t)i;
     acc_or |= e; acc_and &= e;
+    #pragma unroll
+    for (int b = 0; b < 8; b++) atomicAdd(&h[b][(e >> (b * 8)) & 0xff], 1u);
   }
   /* wave reduce then one atomic per wave (G12) */
   for (int off = 32; off > 0; off >>= 1) {
@@ -269,6 +279,11 @@ __global__ void k_encode(int64_t n, const void* keys, uint64_t* ek, uint32_t* id
   if ((threadIdx.x & (WAVE - 1)) == 0) {
     atomicOr(bits_or, (unsigned long long)acc_or);
     atomicAnd(bits_and, (unsigned long long)acc_and);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < 8 * 256; b += blockDim.x) {
+    uint32_t v = ((uint32_t*)h)[b];
+    if (v) atomicAdd(&ghist[b], v);
   }
 }
 
@@ -382,6 +397,9 @@ static int exclusive_scan_u32(hipStream_t s, int64_t n, const uint32_t* in,
  * BIN_MODE 1: digit = pmod(murmur3(key,42), nparts)  (partition pass;
  *             key slot carries the RAW int64 key)
  * BIN_MODE 2: digit = key >> 63 stored pid in high bits? (unused)        */
+struct scatter_geom { int block, items; };
+static scatter_geom get_sort_geom(void);
+
 template <int BIN_MODE>
 DEV int compute_bin(uint64_t key, int shift, int nparts) {
   if (BIN_MODE == 0) return (int)((key >> shift) & 0xff);
@@ -391,12 +409,25 @@ DEV int compute_bin(uint64_t key, int shift, int nparts) {
   return spark_pmod(mm3_hash_long(raw, 42), nparts);
 }
 
-template <int BIN_MODE, int BLOCK, int ITEMS>
+/* status word for decoupled lookback: [63:56] epoch, [55:54] status
+ * (1 = aggregate, 2 = inclusive prefix), [53:0] value. One 8-byte
+ * agent-scope atomic store/load per word (single-granule publish: the
+ * payload IS the flag — no fences needed). */
+#define OSW_AGG (1ULL << 54)
+#define OSW_PFX (2ULL << 54)
+#define OSW_VAL(x) ((x) & ((1ULL << 54) - 1))
+#define OSW_EPOCH(x) ((x) >> 56)
+#define OSW_SPIN_LIMIT (1u << 22)
+
+template <int BIN_MODE, int BLOCK, int ITEMS, bool LOOKBACK>
 __global__ __launch_bounds__(BLOCK)
 void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
                      uint64_t* kout, uint32_t* iout,
                      const uint32_t* scanned /* [256][nblocks] exclusive */,
-                     int shift, int nblocks, int nparts) {
+                     int shift, int nblocks, int nparts,
+                     unsigned long long* state /* [nblocks][256] */,
+                     const uint32_t* gbase /* [256] pass bin bases */,
+                     unsigned long long* err_flag, uint32_t epoch) {
   constexpr int WAVES = BLOCK / WAVE;
   constexpr int TILE = BLOCK * ITEMS;
   __shared__ uint32_t wave_hist[WAVES][256];
@@ -457,13 +488,19 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
 
   /* cross-wave exclusive prefix per bin + block-wide exclusive scan over
    * bins. The first 256 threads each own one bin. */
+  uint32_t acc = 0;
   if (tid < 256) {
     int bin = tid;
-    uint32_t acc = 0;
     for (int w = 0; w < WAVES; w++) {
       uint32_t t = wave_hist[w][bin];
       wave_hist[w][bin] = acc;
       acc += t;
+    }
+    if (LOOKBACK) {
+      /* publish own AGGREGATE early so successors can proceed */
+      __hip_atomic_store(&state[(int64_t)blockIdx.x * 256 + bin],
+                         ((unsigned long long)epoch << 56) | OSW_AGG | acc,
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     }
     uint32_t inc = wave_inclusive_scan(acc);
     if (lane == WAVE - 1) wtot[wave] = inc;
@@ -476,7 +513,36 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
     for (int w = 0; w < wave; w++) woff += wtot[w];
     uint32_t excl = bin_start[bin] + woff;
     bin_start[bin] = excl;
-    bin_gbase[bin] = scanned[(int64_t)bin * nblocks + blockIdx.x] - excl;
+    if (LOOKBACK) {
+      /* decoupled lookback: sum predecessors' counts for this bin */
+      unsigned long long pred = 0;
+      uint32_t spins = 0;
+      for (int p = (int)blockIdx.x - 1; p >= 0;) {
+        unsigned long long v = __hip_atomic_load(&state[(int64_t)p * 256 + bin],
+                                                 __ATOMIC_RELAXED,
+                                                 __HIP_MEMORY_SCOPE_AGENT);
+        if (OSW_EPOCH(v) != epoch || !(v & (OSW_AGG | OSW_PFX))) {
+          if (++spins > OSW_SPIN_LIMIT) {      /* bounded: give up, flag host */
+            __hip_atomic_store(err_flag, 1ull, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            break;
+          }
+          __builtin_amdgcn_s_sleep(1);
+          continue;
+        }
+        pred += OSW_VAL(v);
+        if (v & OSW_PFX) break;
+        p--;
+      }
+      /* publish inclusive prefix (even after timeout, to unblock others) */
+      __hip_atomic_store(&state[(int64_t)blockIdx.x * 256 + bin],
+                         ((unsigned long long)epoch << 56) | OSW_PFX |
+                             (pred + acc),
+                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      bin_gbase[bin] = gbase[bin] + (uint32_t)pred - excl;
+    } else {
+      bin_gbase[bin] = scanned[(int64_t)bin * nblocks + blockIdx.x] - excl;
+    }
   }
   __syncthreads();
 
@@ -505,8 +571,6 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
 }
 
 /* runtime-selectable geometry (GPUQ_SORT_GEOM env: "BLOCKxITEMS") */
-struct scatter_geom { int block, items; };
-
 static scatter_geom get_sort_geom(void) {
   static scatter_geom g = {0, 0};
   if (g.block == 0) {
@@ -518,28 +582,26 @@ static scatter_geom get_sort_geom(void) {
   return g;
 }
 
-template <int BIN_MODE>
+template <int BIN_MODE, bool LOOKBACK>
 static void launch_scatter(hipStream_t s, scatter_geom g, int64_t nb,
                            int64_t n, const uint64_t* kin, const uint32_t* iin,
                            uint64_t* kout, uint32_t* iout,
-                           const uint32_t* scanned, int shift, int nparts) {
+                           const uint32_t* scanned, int shift, int nparts,
+                           unsigned long long* state, const uint32_t* gbase,
+                           unsigned long long* err_flag, uint32_t epoch) {
   dim3 grid((uint32_t)nb);
-  if (g.block == 256 && g.items == 16)
-    k_radix_scatter<BIN_MODE, 256, 16><<<grid, 256, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
-  else if (g.block == 512 && g.items == 8)
-    k_radix_scatter<BIN_MODE, 512, 8><<<grid, 512, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
-  else if (g.block == 512 && g.items == 16)
-    k_radix_scatter<BIN_MODE, 512, 16><<<grid, 512, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
-  else if (g.block == 1024 && g.items == 8)
-    k_radix_scatter<BIN_MODE, 1024, 8><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
-  else if (g.block == 512 && g.items == 4)
-    k_radix_scatter<BIN_MODE, 512, 4><<<grid, 512, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
-  else if (g.block == 1024 && g.items == 2)
-    k_radix_scatter<BIN_MODE, 1024, 2><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
-  else if (g.block == 1024 && g.items == 6)
-    k_radix_scatter<BIN_MODE, 1024, 6><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
-  else
-    k_radix_scatter<BIN_MODE, 1024, 4><<<grid, 1024, 0, s>>>(n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts);
+#define LS(B, I) k_radix_scatter<BIN_MODE, B, I, LOOKBACK><<<grid, B, 0, s>>>( \
+      n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase, \
+      err_flag, epoch)
+  if (g.block == 256 && g.items == 16) LS(256, 16);
+  else if (g.block == 512 && g.items == 8) LS(512, 8);
+  else if (g.block == 512 && g.items == 16) LS(512, 16);
+  else if (g.block == 1024 && g.items == 8) LS(1024, 8);
+  else if (g.block == 512 && g.items == 4) LS(512, 4);
+  else if (g.block == 1024 && g.items == 2) LS(1024, 2);
+  else if (g.block == 1024 && g.items == 6) LS(1024, 6);
+  else LS(1024, 4);
+#undef LS
 }
 
 /* decode sorted keys back to the output dtype */
@@ -559,18 +621,24 @@ __global__ void k_decode(int64_t n, const uint64_t* ek, void* out) {
   }
 }
 
-/* workspace layout for sort/partition */
+/* workspace layout for sort/partition (geometry-aware: the GPUQ_SORT_GEOM
+ * env must not change between workspace sizing and the sort call) */
 struct sort_ws {
   uint64_t* ka; uint64_t* kb;
   uint32_t* ia; uint32_t* ib;
   uint32_t* hist; uint32_t* hist_scan; uint32_t* block_sums;
   unsigned long long* bits; /* [0]=and [1]=or */
+  uint32_t* ghist;          /* [8][256] global byte histograms */
+  uint32_t* gbase;          /* [8][256] per-pass exclusive bin bases */
+  unsigned long long* state; /* [nblocks][256] lookback status words */
+  unsigned long long* err;   /* lookback timeout flag */
 };
 
 static int64_t sort_nblocks(int64_t n, int tile) { return (n + tile - 1) / tile; }
 
 static void sort_ws_layout(int64_t n, int nbins, sort_ws* w, char* basep, int64_t* total) {
-  int64_t nb = sort_nblocks(n, 1024);  /* sized for the smallest tile */
+  int tile = get_sort_geom().block * get_sort_geom().items;
+  int64_t nb = sort_nblocks(n, tile);
   int64_t hist_n = (int64_t)nbins * nb;
   int64_t scan_blocks = (hist_n + SCAN_TILE - 1) / SCAN_TILE + 1;
   int64_t off = 0;
@@ -587,6 +655,10 @@ static void sort_ws_layout(int64_t n, int nbins, sort_ws* w, char* basep, int64_
   w->hist_scan = (uint32_t*)take(hist_n * 4);
   w->block_sums = (uint32_t*)take(scan_blocks * 4);
   w->bits = (unsigned long long*)take(16);
+  w->ghist = (uint32_t*)take(8 * 256 * 4);
+  w->gbase = (uint32_t*)take(8 * 256 * 4);
+  w->state = (unsigned long long*)take(nb * 256 * 8);
+  w->err = (unsigned long long*)take(8);
   *total = off;
 }
 
@@ -612,42 +684,99 @@ extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
     FAIL(GPUQ_ERR_INVALID, "sort: workspace %lld < %lld", (long long)workspace_bytes, (long long)need);
   if (n == 0) return GPUQ_OK;
 
-  HIP_TRY(hipMemsetAsync(w.bits, 0, 16, s));
-  HIP_TRY(hipMemsetAsync(w.bits, 0xFF, 8, s));  /* bits_and = ~0 */
-  /* encode + skip-byte reduction */
-  if (key.dtype == GPUQ_FLOAT64) {
-    if (desc) k_encode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1]);
-    else      k_encode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1]);
-  } else {
-    if (desc) k_encode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1]);
-    else      k_encode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1]);
-  }
-  HIP_TRY(hipGetLastError());
-  unsigned long long hb[2];
-  HIP_TRY(hipMemcpyAsync(hb, w.bits, 16, hipMemcpyDeviceToHost, s));
-  HIP_TRY(hipStreamSynchronize(s));
-  uint64_t bits_changed = hb[0] ^ hb[1];
-
   scatter_geom geom = get_sort_geom();
   int tile = geom.block * geom.items;
   int64_t nb = sort_nblocks(n, tile);
+  bool onesweep = getenv("GPUQ_NO_ONESWEEP") == nullptr;
+
+  HIP_TRY(hipMemsetAsync(w.bits, 0, 16, s));
+  HIP_TRY(hipMemsetAsync(w.bits, 0xFF, 8, s));  /* bits_and = ~0 */
+  HIP_TRY(hipMemsetAsync(w.ghist, 0, 8 * 256 * 4, s));
+  /* encode + skip-byte reduction + all-byte global histograms (one read) */
+  { hipEvent_t _pe = prof_begin(s);
+  if (key.dtype == GPUQ_FLOAT64) {
+    if (desc) k_encode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+    else      k_encode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+  } else {
+    if (desc) k_encode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+    else      k_encode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+  }
+  prof_end("encode", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  unsigned long long hb[2];
+  uint32_t hghist[8 * 256];
+  HIP_TRY(hipMemcpyAsync(hb, w.bits, 16, hipMemcpyDeviceToHost, s));
+  HIP_TRY(hipMemcpyAsync(hghist, w.ghist, sizeof(hghist), hipMemcpyDeviceToHost, s));
+  HIP_TRY(hipStreamSynchronize(s));
+  uint64_t bits_changed = hb[0] ^ hb[1];
+
+  int retries = 0;
+retry:
+  if (onesweep) {
+    /* per-pass exclusive bin bases, computed on host from the one-read
+     * global histograms (replaces the per-pass hist kernel + device scan) */
+    uint32_t hgbase[8 * 256];
+    for (int b = 0; b < 8; b++) {
+      uint32_t run = 0;
+      for (int bin = 0; bin < 256; bin++) {
+        hgbase[b * 256 + bin] = run;
+        run += hghist[b * 256 + bin];
+      }
+    }
+    HIP_TRY(hipMemcpyAsync(w.gbase, hgbase, sizeof(hgbase), hipMemcpyHostToDevice, s));
+    HIP_TRY(hipMemsetAsync(w.state, 0, nb * 256 * 8, s));
+    HIP_TRY(hipMemsetAsync(w.err, 0, 8, s));
+  }
+
   uint64_t *kin = w.ka, *kout = w.kb;
   uint32_t *iin = w.ia, *iout = w.ib;
   for (int byte = 0; byte < 8; byte++) {
     if (((bits_changed >> (byte * 8)) & 0xff) == 0) continue;  /* RadixSort.java:126 skip */
     int shift = byte * 8;
-    { hipEvent_t _pe = prof_begin(s);
-    k_radix_hist<<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, shift, w.hist, (int)nb, tile);
-    prof_end("radix_hist", s, _pe); }
-    HIP_TRY(hipGetLastError());
-    int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
-    if (rc) return rc;
-    { hipEvent_t _pe = prof_begin(s);
-    launch_scatter<0>(s, geom, nb, n, kin, iin, kout, iout, w.hist_scan, shift, 0);
-    prof_end("radix_scatter", s, _pe); }
-    HIP_TRY(hipGetLastError());
+    if (onesweep) {
+      { hipEvent_t _pe = prof_begin(s);
+      launch_scatter<0, true>(s, geom, nb, n, kin, iin, kout, iout, nullptr,
+                              shift, 0, w.state, w.gbase + byte * 256, w.err,
+                              (uint32_t)(byte + 1));
+      prof_end("radix_scatter", s, _pe); }
+      HIP_TRY(hipGetLastError());
+    } else {
+      { hipEvent_t _pe = prof_begin(s);
+      k_radix_hist<<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, shift, w.hist, (int)nb, tile);
+      prof_end("radix_hist", s, _pe); }
+      HIP_TRY(hipGetLastError());
+      int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
+      if (rc) return rc;
+      { hipEvent_t _pe = prof_begin(s);
+      launch_scatter<0, false>(s, geom, nb, n, kin, iin, kout, iout, w.hist_scan,
+                               shift, 0, nullptr, nullptr, nullptr, 0);
+      prof_end("radix_scatter", s, _pe); }
+      HIP_TRY(hipGetLastError());
+    }
     uint64_t* tk = kin; kin = kout; kout = tk;
     uint32_t* ti = iin; iin = iout; iout = ti;
+  }
+
+  if (onesweep) {
+    /* one bounded-spin timeout check; on timeout redo the whole sort with
+     * the hist+scan path (input untouched - passes only wrote scratch) */
+    unsigned long long herr = 0;
+    HIP_TRY(hipMemcpyAsync(&herr, w.err, 8, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (herr) {
+      if (++retries > 1) FAIL(GPUQ_ERR_HIP, "sort: lookback timed out twice");
+      onesweep = false;
+      /* re-encode (ka/ia were consumed as ping-pong scratch) */
+      if (key.dtype == GPUQ_FLOAT64) {
+        if (desc) k_encode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+        else      k_encode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+      } else {
+        if (desc) k_encode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+        else      k_encode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+      }
+      HIP_TRY(hipGetLastError());
+      goto retry;
+    }
   }
   HIP_TRY(hipMemcpyAsync(out_perm, iin, n * 4, hipMemcpyDeviceToDevice, s));
   if (out_keys) {
@@ -730,7 +859,8 @@ extern "C" int gpuq_partition_perm(void* stream, int64_t n, gpuq_col key,
   int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
   if (rc) return rc;
   { hipEvent_t _pe = prof_begin(s);
-    launch_scatter<0>(s, geom, nb, n, w.ka, w.ia, w.kb, w.ib, w.hist_scan, 0, 0);
+    launch_scatter<0, false>(s, geom, nb, n, w.ka, w.ia, w.kb, w.ib, w.hist_scan,
+                             0, 0, nullptr, nullptr, nullptr, 0);
     prof_end("radix_scatter", s, _pe); }
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipMemcpyAsync(out_perm, w.ib, n * 4, hipMemcpyDeviceToDevice, s));
