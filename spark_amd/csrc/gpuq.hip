@@ -513,40 +513,15 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
     for (int w = 0; w < wave; w++) woff += wtot[w];
     uint32_t excl = bin_start[bin] + woff;
     bin_start[bin] = excl;
-    if (LOOKBACK) {
-      /* decoupled lookback: sum predecessors' counts for this bin */
-      unsigned long long pred = 0;
-      uint32_t spins = 0;
-      for (int p = (int)blockIdx.x - 1; p >= 0;) {
-        unsigned long long v = __hip_atomic_load(&state[(int64_t)p * 256 + bin],
-                                                 __ATOMIC_RELAXED,
-                                                 __HIP_MEMORY_SCOPE_AGENT);
-        if (OSW_EPOCH(v) != epoch || !(v & (OSW_AGG | OSW_PFX))) {
-          if (++spins > OSW_SPIN_LIMIT) {      /* bounded: give up, flag host */
-            __hip_atomic_store(err_flag, 1ull, __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT);
-            break;
-          }
-          __builtin_amdgcn_s_sleep(1);
-          continue;
-        }
-        pred += OSW_VAL(v);
-        if (v & OSW_PFX) break;
-        p--;
-      }
-      /* publish inclusive prefix (even after timeout, to unblock others) */
-      __hip_atomic_store(&state[(int64_t)blockIdx.x * 256 + bin],
-                         ((unsigned long long)epoch << 56) | OSW_PFX |
-                             (pred + acc),
-                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-      bin_gbase[bin] = gbase[bin] + (uint32_t)pred - excl;
-    } else {
+    if (!LOOKBACK) {
       bin_gbase[bin] = scanned[(int64_t)bin * nblocks + blockIdx.x] - excl;
+    } else {
+      bin_gbase[bin] = excl;  /* stash; lookback resolves after staging */
     }
   }
   __syncthreads();
 
-  /* stage reordered tile in LDS */
+  /* stage reordered tile in LDS (lookback waits overlap with this) */
   for (int r = 0; r < ITEMS; r++) {
     int64_t i = wbase + r * WAVE + lane;
     if (i < n) {
@@ -554,6 +529,56 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
       stage_k[pos] = k[r];
       stage_i[pos] = id[r];
     }
+  }
+  if (LOOKBACK && tid < 256) {
+    int bin = tid;
+    uint32_t excl = bin_gbase[bin];
+    uint32_t own = acc;   /* this thread's phase-2 block count for its bin */
+    /* chunked decoupled lookback: LB predecessor words in flight per step */
+    constexpr int LB = 4;
+    unsigned long long pred = 0;
+    uint32_t spins = 0;
+    int p = (int)blockIdx.x - 1;
+    while (p >= 0) {
+      int cnt = (p + 1 < LB) ? p + 1 : LB;
+      unsigned long long v[LB];
+      #pragma unroll
+      for (int j = 0; j < LB; j++)
+        if (j < cnt)
+          v[j] = __hip_atomic_load(&state[(int64_t)(p - j) * 256 + bin],
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      /* walk newest -> oldest; stop at first PREFIX; retry at not-ready */
+      bool stall = false, done = false;
+      unsigned long long add = 0;
+      #pragma unroll
+      for (int j = 0; j < LB; j++) {
+        if (j >= cnt || done || stall) continue;
+        if (OSW_EPOCH(v[j]) != epoch || !(v[j] & (OSW_AGG | OSW_PFX))) {
+          stall = true;
+        } else {
+          add += OSW_VAL(v[j]);
+          if (v[j] & OSW_PFX) done = true;
+        }
+      }
+      if (stall) {
+        if (++spins > OSW_SPIN_LIMIT) {        /* bounded: give up, flag host */
+          __hip_atomic_store(err_flag, 1ull, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          break;
+        }
+        __builtin_amdgcn_s_sleep(1);
+        continue;                               /* retry same chunk */
+      }
+      pred += add;
+      if (done) break;
+      p -= cnt;
+    }
+    /* publish inclusive prefix (even after timeout, to unblock others) */
+    __hip_atomic_store(&state[(int64_t)blockIdx.x * 256 + bin],
+                       ((unsigned long long)epoch << 56) | OSW_PFX |
+                           (pred + own),
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    bin_gbase[bin] = gbase[bin] + (uint32_t)pred - excl;
   }
   __syncthreads();
 
