@@ -280,17 +280,50 @@ class GpuShuffleExchangeExec(SparkPlan):
     def supports_columnar(self):
         return True
 
+    # --- ShuffleExchangeLike mirror (ShuffleExchangeExec.scala:51-151) ---
+    # AQE materializes each shuffle stage and reads its map-output stats
+    # (AdaptiveSparkPlanExec.scala:784, QueryStageExec.materialize :69);
+    # these fields are what a Scala GpuShuffleExchangeExec would return from
+    # mapOutputStatisticsFuture/runtimeStatistics, with per-partition byte
+    # sizes taken from gpuq_partition_perm's counts (no per-row CPU work).
+
+    @property
+    def num_mappers(self) -> int:
+        import torch.distributed as dist
+        return dist.get_world_size() if dist.is_initialized() else 1
+
+    @property
+    def num_partitions(self) -> int:
+        return self.num_mappers
+
+    def runtime_statistics(self):
+        """(total bytes written, rows) of this rank's map output —
+        ShuffleExchangeLike.runtimeStatistics (:146)."""
+        return dict(self._stats) if hasattr(self, "_stats") else None
+
     def execute_columnar(self):
         import torch.distributed as dist
-        from .exchange import shuffle_exchange_gpu
+        from . import gpuq
+        from .exchange import exchange_columns
         assert dist.is_initialized(), "GpuShuffleExchangeExec needs torch.distributed"
         key_name = self.keys[0]
+        world = self.num_partitions
         for batch in self.children[0].execute_columnar():
-            payload = {n: t for n, t in batch.columns().items() if n != key_name}
-            k, cols = shuffle_exchange_gpu(batch.column(key_name), payload)
-            cols[key_name] = k
+            key = batch.column(key_name)
+            perm, counts = gpuq.partition_perm(key, world,
+                                               key_validity=batch.validity(key_name))
+            cols = {key_name: gpuq.gather(key, perm)}
+            row_bytes = 0
+            for name, t in batch.columns().items():
+                if name != key_name:
+                    cols[name] = gpuq.gather(t, perm)
+                row_bytes += t.element_size()
+            in_splits = counts.cpu().tolist()
+            self._stats = {"bytes_by_partition": [c * row_bytes for c in in_splits],
+                           "rows_written": int(sum(in_splits))}
+            out, _ = exchange_columns(cols, in_splits)
             batch.close()
-            yield ColumnarBatch(cols)
+            yield ColumnarBatch(out)
 
 
 class GpuShuffledHashJoinExec(SparkPlan):

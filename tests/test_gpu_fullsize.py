@@ -1,0 +1,96 @@
+"""Full-size property tests (BASELINE config scale) — the oracle cannot run
+at 1B rows in test time, so parity at full size is checked through
+size-independent properties (sortedness, permutation validity, key-multiset
+preservation, aggregation linearity, join key equality), per SURVEY §8(c).
+Scale via GPUQ_FULLSIZE_ROWS (default 1B, the config-2 size)."""
+import os
+
+import numpy as np
+import pytest
+
+torch = pytest.importorskip("torch")
+pytestmark = pytest.mark.gpu
+
+ROWS = int(os.environ.get("GPUQ_FULLSIZE_ROWS", 1_000_000_000))
+
+
+@pytest.fixture(scope="module")
+def gq():
+    from spark_amd import gpuq
+    assert torch.cuda.is_available()
+    return gpuq
+
+
+def test_sort_1b_properties(gq):
+    keys = gq.gen_i64(seed=42, n=ROWS)           # full-range int64 (config 2)
+    ws = gq.sort_workspace(ROWS)
+    perm, skeys = gq.sort_perm(keys, workspace=ws)
+    del ws
+    # (1) sorted ascending
+    assert bool((skeys[1:] >= skeys[:-1]).all())
+    # (2) perm is a permutation: every row id exactly once
+    pu = perm.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    counts = torch.zeros(ROWS, dtype=torch.int32, device="cuda")
+    counts.scatter_add_(0, pu, torch.ones(ROWS, dtype=torch.int32, device="cuda"))
+    assert bool((counts == 1).all())
+    # (3) output keys = input keys permuted (spot + checksum):
+    assert bool((skeys == keys[pu]).all())
+    # (4) stability on ties: full-range keys rarely tie; check explicitly on a
+    # low-cardinality full-size sort instead
+    del keys, skeys, perm, pu, counts
+    torch.cuda.empty_cache()
+    keys = gq.gen_i64(seed=7, n=ROWS, range_=1000)
+    perm, skeys = gq.sort_perm(keys)
+    pu = perm.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    # within equal keys, row ids must ascend: (key asc, then rowid asc) means
+    # the (key*2^32... use pairwise check: where key equal, perm increasing
+    eq = skeys[1:] == skeys[:-1]
+    inc = pu[1:] > pu[:-1]
+    assert bool((inc | ~eq).all())
+    del keys, skeys, perm, pu
+    torch.cuda.empty_cache()
+
+
+def test_agg_1b_properties(gq):
+    groups = 10_000_000                           # config 3
+    keys = gq.gen_i64(seed=52, n=ROWS, range_=groups)
+    vals = gq.gen_f64_unit(seed=53, n=ROWS)
+    total = float(vals.sum())
+    cap = 1 << (groups * 2 - 1).bit_length()
+    ok, okv, osum, osv, ocnt = gq.hash_agg(keys, vals, cap)
+    # every group distinct, all keys in range, counts sum to n (COUNT exact)
+    assert int(ocnt.sum()) == ROWS
+    assert ok.unique().numel() == ok.numel()
+    assert bool((ok >= 0).all()) and bool((ok < groups).all())
+    # linearity: sum of group sums == total sum (1e-6 relative, north star)
+    np.testing.assert_allclose(float(osum.sum()), total, rtol=1e-6)
+    del keys, vals, ok, okv, osum, osv, ocnt
+    torch.cuda.empty_cache()
+
+
+def test_join_halfbillion_properties(gq):
+    rows = ROWS // 2                              # config-4 per-GPU slice
+    keyspace = rows
+    bkeys = gq.gen_i64(seed=62, n=rows, range_=keyspace)
+    pkeys = gq.gen_i64(seed=64, n=rows, range_=keyspace)
+    cap = 1 << (rows * 2 - 1).bit_length()
+    ws = gq.join_build(bkeys, cap)
+    out_cap = int(rows * 2.5)
+    op, ob, nm = gq.join_probe(pkeys, ws, cap, rows, out_cap)
+    assert op is not None
+    # every emitted pair joins equal keys
+    pu_p = op.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    pu_b = ob.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    assert bool((pkeys[pu_p] == bkeys[pu_b]).all())
+    # match count == sum over keys of cb(k)*cp(k), computed via the agg
+    # kernel's COUNT on both sides (independent code path from the join)
+    ones = torch.zeros(rows, dtype=torch.float64, device="cuda")
+    bk_k, _, _, _, bk_c = gq.hash_agg(bkeys, ones, cap)
+    pk_k, _, _, _, pk_c = gq.hash_agg(pkeys, ones, cap)
+    bs, bo = bk_k.sort()
+    ps, po = pk_k.sort()
+    idx = torch.searchsorted(bs, ps)
+    idx_c = idx.clamp(max=bs.numel() - 1)
+    hit = bs[idx_c] == ps
+    expected = int((bk_c[bo][idx_c][hit] * pk_c[po][hit]).sum())
+    assert nm == expected

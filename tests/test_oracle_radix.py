@@ -156,3 +156,17 @@ def test_sort_perm_f64_special_values():
     gotd = keys[permd]
     assert np.isnan(gotd[0]) and np.isnan(gotd[1]) and gotd[2] == np.inf
     assert gotd[-1] == -np.inf
+
+
+def test_xorshift_restatement_lock():
+    """Regression lock for the XORShiftRandom restatement (fixture header
+    explains provenance)."""
+    import json, os
+    fix = json.load(open(os.path.join(os.path.dirname(__file__), "golden",
+                                      "xorshift_restatement_lock.json")))
+    assert oracle.XorShiftRandom(123).fill_longs(64).tolist() == \
+        fix["seed123_nextLong_first64"]
+    assert oracle.XorShiftRandom(123).fill_longs(32, mask=0xFF).tolist() == \
+        fix["seed123_masked_ff_first32"]
+    r = oracle.XorShiftRandom(1)
+    assert [r.next_int(5) for _ in range(16)] == fix["seed1_nextInt5_first16"]
