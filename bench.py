@@ -139,10 +139,11 @@ class JoinWorkload:
         self.bpay = gq.gen_i64(seed=63, n=rows, range_=0, start=rank * rows)
         self.pkeys = gq.gen_i64(seed=64, n=rows, range_=keyspace, start=rank * rows)
         self.ppay = gq.gen_i64(seed=65, n=rows, range_=0, start=rank * rows)
-        self.part_ws = gq.partition_workspace(rows, max(world, 1))
+        self.part_ws = gq.partition_workspace(rows, world) if world > 1 else None
         # after exchange each rank holds ~rows per side (margin 1.25x)
         self.local_cap_rows = int(rows * 1.25) + 4096
-        self.cap = 1 << max(4, (self.local_cap_rows * 2 - 1).bit_length())
+        # ~0.6 max load factor: linear probing stays short, table stays small
+        self.cap = 1 << max(4, int(self.local_cap_rows * 1.6 - 1).bit_length())
         self.join_ws = torch.empty(
             gq.lib().gpuq_join_build_workspace_bytes(self.local_cap_rows, self.cap),
             dtype=torch.uint8, device="cuda")
