@@ -1137,6 +1137,24 @@ extern "C" int64_t gpuq_join_build_workspace_bytes(int64_t brows, int64_t cap) {
   return total;
 }
 
+/* chain-head word: bit 31 = MULTI (slot holds >1 row) so single-match
+ * probes never touch next[]; low 31 bits = first build row id */
+#define JOIN_MULTI 0x80000000u
+
+DEV void join_push_head(unsigned int* headp, unsigned int i, unsigned int* next) {
+  unsigned int old = __hip_atomic_load(headp, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+  for (;;) {
+    unsigned int newv = i | (old != JOIN_NIL ? JOIN_MULTI : 0u);
+    unsigned int prev = atomicCAS(headp, old, newv);
+    if (prev == old) {
+      next[i] = (old == JOIN_NIL) ? JOIN_NIL : (old & ~JOIN_MULTI);
+      break;
+    }
+    old = prev;
+  }
+}
+
 __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                              unsigned long long* slots, unsigned int* next,
                              join_sp* sp, int64_t cap_mask) {
@@ -1146,8 +1164,7 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
     if (!bit_valid(kvalid, i)) continue;  /* NULL never matches (inner join) */
     int64_t k = keys[i];
     if ((unsigned long long)k == AGG_EMPTY) {
-      unsigned int old = atomicExch(&sp->m1_head, (unsigned int)i);
-      next[i] = old;
+      join_push_head(&sp->m1_head, (unsigned int)i, next);
       continue;
     }
     uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
@@ -1161,8 +1178,7 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
       }
       slot = (slot + 1) & (uint64_t)cap_mask;
     }
-    unsigned int old = atomicExch((unsigned int*)&slots[2 * slot + 1], (unsigned int)i);
-    next[i] = old;
+    join_push_head((unsigned int*)&slots[2 * slot + 1], (unsigned int)i, next);
   }
 }
 
@@ -1194,9 +1210,21 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
         }
       }
     }
-    /* pass 1: count my matches */
+    /* count matches; the single-match fast path (no MULTI bit) never reads
+     * next[]; the first two chain entries are cached in registers so the
+     * emit pass rarely re-walks */
     uint32_t cnt = 0;
-    for (unsigned int b = head; b != JOIN_NIL; b = next[b]) cnt++;
+    unsigned int c0 = JOIN_NIL, c1 = JOIN_NIL;
+    if (head != JOIN_NIL) {
+      if (!(head & JOIN_MULTI)) {
+        c0 = head; cnt = 1;
+      } else {
+        for (unsigned int b = head & ~JOIN_MULTI; b != JOIN_NIL; b = next[b]) {
+          if (cnt == 0) c0 = b; else if (cnt == 1) c1 = b;
+          cnt++;
+        }
+      }
+    }
     /* wave-aggregated reservation: one atomic per wave */
     uint32_t incl = wave_inclusive_scan(cnt);
     uint32_t total = __shfl(incl, WAVE - 1);
@@ -1204,12 +1232,13 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
     if (lane == WAVE - 1 && total > 0)
       base = atomicAdd(&sp->cursor, (unsigned long long)total);
     base = __shfl(base, WAVE - 1);
-    /* pass 2: emit at base + my exclusive offset */
     int64_t o = (int64_t)base + (incl - cnt);
-    for (unsigned int b = head; b != JOIN_NIL; b = next[b], o++) {
-      if (o < out_cap) {
-        out_p[o] = (uint32_t)i;
-        out_b[o] = b;
+    if (cnt >= 1 && o < out_cap) { out_p[o] = (uint32_t)i; out_b[o] = c0; }
+    if (cnt >= 2 && o + 1 < out_cap) { out_p[o + 1] = (uint32_t)i; out_b[o + 1] = c1; }
+    if (cnt > 2) {
+      unsigned int b = next[c1];
+      for (uint32_t j = 2; j < cnt; j++, b = next[b]) {
+        if (o + j < out_cap) { out_p[o + j] = (uint32_t)i; out_b[o + j] = b; }
       }
     }
   }
@@ -1220,7 +1249,7 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   hipStream_t s = (hipStream_t)stream;
   if (cap <= 0 || (cap & (cap - 1)))
     FAIL(GPUQ_ERR_INVALID, "join: capacity %lld not a power of two", (long long)cap);
-  if (brows > 0xFFFFFFFELL) FAIL(GPUQ_ERR_INVALID, "join: build side too large for u32 rowids");
+  if (brows > 0x7FFFFFFELL) FAIL(GPUQ_ERR_INVALID, "join: build side too large for 31-bit rowids");
   if (bkey.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "join: key must be int64");
   join_ws w; int64_t need;
   join_ws_layout(cap, brows, &w, (char*)workspace, &need);
