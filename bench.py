@@ -260,20 +260,27 @@ def main():
         w = AggWorkload(gq, args.agg_rows, args.agg_groups, rank)
         sec = time_workload(w, args.steps, args.warmup, world)
         ng = w.ngroups
+        ab_ms, ab_n = gq.kernel_stats("agg_build")
         w.free()
         torch.cuda.empty_cache()
         results["agg"] = {"sec_per_step": sec,
                           "rows_per_sec": args.agg_rows * world / sec,
+                          "agg_build_ms_avg": ab_ms / max(ab_n, 1),
                           "ngroups": ng}
 
     if "join" in wl:
         w = JoinWorkload(gq, args.join_rows, rank, world)
+        gq.kernel_stats_reset()
         sec = time_workload(w, args.steps, args.warmup, world)
         nm = w.nmatches
         w.free()
         torch.cuda.empty_cache()
+        jb_ms, jb_n = gq.kernel_stats("join_build")
+        jp_ms, jp_n = gq.kernel_stats("join_probe")
         results["join"] = {
             "sec_per_step": sec,
+            "build_ms_avg": jb_ms / max(jb_n, 1),
+            "probe_ms_avg": jp_ms / max(jp_n, 1),
             # rows processed = both tables, all ranks (the config's rate basis)
             "rows_per_sec": 2 * args.join_rows * world / sec,
             "matches_local": nm,

@@ -183,6 +183,21 @@ __global__ void k_gen_f64_unit(uint64_t seed, uint64_t start, int64_t n, double*
   }
 }
 
+/* grid shape for random-access hash kernels (A/B via GPUQ_HASH_GRID:
+ * 0 = one element per thread, N = cap at N blocks with grid-stride) */
+static dim3 hash_grid(int64_t n, int block = 256) {
+  static int cap = -2;
+  if (cap == -2) {
+    const char* e = getenv("GPUQ_HASH_GRID");
+    cap = e ? atoi(e) : 2048;
+  }
+  int64_t b = (n + block - 1) / block;
+  if (cap > 0 && b > cap) b = cap;
+  if (b > 0x7FFFFFFF) b = 0x7FFFFFFF;
+  if (b < 1) b = 1;
+  return dim3((uint32_t)b);
+}
+
 static dim3 grid1d(int64_t n, int block = 256) {
   int64_t b = (n + block - 1) / block;
   if (b > 2048) b = 2048;  /* grid-stride beyond (G11: cap + stride) */
@@ -1063,11 +1078,11 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
   if (n > 0) {
     { hipEvent_t _pe = prof_begin(s);
     if (ops == AGG_OP_SUM)
-      k_agg_build<AGG_OP_SUM><<<grid1d(n), 256, 0, s>>>(
+      k_agg_build<AGG_OP_SUM><<<hash_grid(n), 256, 0, s>>>(
           n, (const int64_t*)key.data, key.validity,
           (const double*)val.data, val.validity, w.tab, w.sp, cap - 1);
     else
-      k_agg_build<AGG_OP_SUM | AGG_OP_COUNT><<<grid1d(n), 256, 0, s>>>(
+      k_agg_build<AGG_OP_SUM | AGG_OP_COUNT><<<hash_grid(n), 256, 0, s>>>(
           n, (const int64_t*)key.data, key.validity,
           (const double*)val.data, val.validity, w.tab, w.sp, cap - 1);
     prof_end("agg_build", s, _pe); }
@@ -1258,7 +1273,7 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
   if (brows > 0) {
     { hipEvent_t _pe = prof_begin(s);
-    k_join_build<<<grid1d(brows), 256, 0, s>>>(brows, (const int64_t*)bkey.data,
+    k_join_build<<<hash_grid(brows), 256, 0, s>>>(brows, (const int64_t*)bkey.data,
                                                bkey.validity, w.slots, w.next,
                                                w.sp, cap - 1);
     prof_end("join_build", s, _pe); }
@@ -1278,7 +1293,7 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
   if (prows > 0) {
     { hipEvent_t _pe = prof_begin(s);
-    k_join_probe<<<grid1d(prows), 256, 0, s>>>(prows, (const int64_t*)pkey.data,
+    k_join_probe<<<hash_grid(prows), 256, 0, s>>>(prows, (const int64_t*)pkey.data,
                                                pkey.validity, w.slots, w.next,
                                                w.sp, cap - 1,
                                                out_p, out_b, out_cap);
