@@ -147,6 +147,7 @@ class JoinWorkload:
         self.join_ws = torch.empty(
             gq.lib().gpuq_join_build_workspace_bytes(self.local_cap_rows, self.cap),
             dtype=torch.uint8, device="cuda")
+        self.probe_ws = gq.join_probe_workspace(self.local_cap_rows)
         self.out_cap = int(rows * 2.5) + 4096
         self.nmatches = 0
 
@@ -177,7 +178,8 @@ class JoinWorkload:
         bn = bk.numel()
         gq.lib().gpuq_join_build_i64(gq._stream(), bn, gq._col(bk),
                                      self.join_ws.data_ptr(), self.cap)
-        op, ob, nm = gq.join_probe(pk, self.join_ws, self.cap, bn, self.out_cap)
+        op, ob, nm = gq.join_probe(pk, self.join_ws, self.cap, bn, self.out_cap,
+                                   probe_ws=self.probe_ws)
         assert op is not None, f"join out_cap {self.out_cap} < {nm}"
         # materialize one payload column per side (what SHJ emits)
         gq.gather(bp, ob)
@@ -185,7 +187,7 @@ class JoinWorkload:
         self.nmatches = nm
 
     def free(self):
-        del self.bkeys, self.bpay, self.pkeys, self.ppay, self.part_ws, self.join_ws
+        del self.bkeys, self.bpay, self.pkeys, self.ppay, self.part_ws, self.join_ws, self.probe_ws
 
 
 def time_workload(w, steps, warmup, world):

@@ -170,19 +170,24 @@ int gpuq_partition_perm(void* stream, int64_t nrows, gpuq_col key,
 /* ---------------------------------------------------------------- */
 
 int64_t gpuq_join_build_workspace_bytes(int64_t build_rows, int64_t capacity);
+int64_t gpuq_join_probe_workspace_bytes(int64_t probe_rows);
 
 /* Build the hash table over the build-side key column. capacity: power of
- * two >= 2*build_rows. The workspace holds the table + chains and must stay
- * alive through probes. */
+ * two >= ~1.6*build_rows. The workspace holds the table + chains (and, for
+ * large NULL-free tables, the hash-ordered copy of the build side that
+ * keeps probes L3-local) and must stay alive through probes. */
 int gpuq_join_build_i64(void* stream, int64_t build_rows, gpuq_col build_key,
                         void* workspace, int64_t capacity);
 
 /* Probe: emits matching (probe_rid, build_rid) uint32 pairs into the
  * caller's buffers (out_cap entries each); *out_nmatches returns the total
  * match count. If the count exceeds out_cap, returns GPUQ_ERR_OVERFLOW
- * after setting *out_nmatches (call again with bigger buffers). */
+ * after setting *out_nmatches (call again with bigger buffers).
+ * probe_workspace (gpuq_join_probe_workspace_bytes) enables the
+ * hash-ordered probe stream; pass NULL to probe in input order. */
 int gpuq_join_probe_i64(void* stream, int64_t probe_rows, gpuq_col probe_key,
                         const void* workspace, int64_t capacity, int64_t build_rows,
+                        void* probe_workspace, int64_t probe_ws_bytes,
                         uint32_t* out_probe_rid, uint32_t* out_build_rid,
                         int64_t out_cap, int64_t* out_nmatches);
 

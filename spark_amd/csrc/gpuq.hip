@@ -302,8 +302,25 @@ __global__ void k_encode(int64_t n, const void* keys, uint64_t* ek, uint32_t* id
   }
 }
 
+/* ---- ranked scatter pass ----
+ * BIN_MODE 0: digit = (key >> shift) & 0xff  (radix sort pass)
+ * BIN_MODE 2: digit = top 8 bits of Murmur3(key,42) (hash-order
+ *             bucketing for the locality-ordered hash join)             */
+struct scatter_geom { int block, items; };
+static scatter_geom get_sort_geom(void);
+
+template <int BIN_MODE>
+DEV int compute_bin(uint64_t key, int shift, int nparts) {
+  (void)nparts;
+  if (BIN_MODE == 0) return (int)((key >> shift) & 0xff);
+  /* BIN_MODE 2: top 8 bits of Murmur3(key,42) — hash-order bucketing so a
+   * bucketed stream sweeps a hash-ordered table monotonically (join) */
+  return (int)(((uint32_t)mm3_hash_long((int64_t)key, 42)) >> 24);
+}
+
 /* per-block histogram of digit at `shift` over the pass input;
  * `tile` elements per block (must match the scatter geometry) */
+template <int BIN_MODE>
 __global__ void k_radix_hist(int64_t n, const uint64_t* keys, int shift,
                              uint32_t* hist /* [256][nblocks] */, int nblocks,
                              int tile) {
@@ -314,7 +331,7 @@ __global__ void k_radix_hist(int64_t n, const uint64_t* keys, int shift,
   int rounds = tile / 256;
   for (int r = 0; r < rounds; r++) {
     int64_t i = base + r * 256 + threadIdx.x;
-    if (i < n) atomicAdd(&h[(keys[i] >> shift) & 0xff], 1u);
+    if (i < n) atomicAdd(&h[compute_bin<BIN_MODE>(keys[i], shift, 0)], 1u);
   }
   __syncthreads();
   if (threadIdx.x < 256)
@@ -405,23 +422,6 @@ static int exclusive_scan_u32(hipStream_t s, int64_t n, const uint32_t* in,
   k_scan_add<<<dim3((uint32_t)nblocks), SCAN_BLOCK, 0, s>>>(n, in, out, block_sums);
   HIP_TRY(hipGetLastError());
   return GPUQ_OK;
-}
-
-/* ---- ranked scatter pass ----
- * BIN_MODE 0: digit = (key >> shift) & 0xff  (radix sort pass)
- * BIN_MODE 1: digit = pmod(murmur3(key,42), nparts)  (partition pass;
- *             key slot carries the RAW int64 key)
- * BIN_MODE 2: digit = key >> 63 stored pid in high bits? (unused)        */
-struct scatter_geom { int block, items; };
-static scatter_geom get_sort_geom(void);
-
-template <int BIN_MODE>
-DEV int compute_bin(uint64_t key, int shift, int nparts) {
-  if (BIN_MODE == 0) return (int)((key >> shift) & 0xff);
-  /* partition: low 63 bits = key, top bit = null flag */
-  if (key & SIGNBIT) return spark_pmod(42, nparts); /* NULL key: hash stays seed */
-  int64_t raw = (int64_t)(key << 1) >> 1; /* sign-extend low 63 bits */
-  return spark_pmod(mm3_hash_long(raw, 42), nparts);
 }
 
 /* status word for decoupled lookback: [63:56] epoch, [55:54] status
@@ -782,7 +782,7 @@ retry:
       HIP_TRY(hipGetLastError());
     } else {
       { hipEvent_t _pe = prof_begin(s);
-      k_radix_hist<<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, shift, w.hist, (int)nb, tile);
+      k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, shift, w.hist, (int)nb, tile);
       prof_end("radix_hist", s, _pe); }
       HIP_TRY(hipGetLastError());
       int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
@@ -893,7 +893,7 @@ extern "C" int gpuq_partition_perm(void* stream, int64_t n, gpuq_col key,
   int tile = geom.block * geom.items;
   int64_t nb = sort_nblocks(n, tile);
   { hipEvent_t _pe = prof_begin(s);
-    k_radix_hist<<<dim3((uint32_t)nb), 256, 0, s>>>(n, w.ka, 0, w.hist, (int)nb, tile);
+    k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, w.ka, 0, w.hist, (int)nb, tile);
     prof_end("radix_hist", s, _pe); }
   HIP_TRY(hipGetLastError());
   int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
@@ -1114,26 +1114,49 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
 
 /* ================= hash join ================= */
 /*
- * Build: table of 16-byte interleaved slots {key u64, head u32, pad} (one
- * cache line per probe); duplicates chained through next[] (the GPU analog
- * of LongToUnsafeRowMap's per-row next pointer, HashedRelation.scala:536-600).
- * EMPTY key sentinel -1 with a dedicated chain for real -1 keys. NULL keys
- * never match. Probe reserves output space once per wave (aggregated
- * atomic): count matches, wave-scan, single atomicAdd, emit.
- * Workspace: [cap 16B slots][build_rows u32 next][special]
+ * Build: table of 16-byte interleaved slots {key u64, head u32, pad};
+ * duplicates chained through next[] (the GPU analog of LongToUnsafeRowMap's
+ * per-row next pointer, HashedRelation.scala:536-600), chain heads carry a
+ * MULTI bit so single-match probes never read next[]. EMPTY key sentinel -1
+ * with a dedicated chain for real -1 keys. NULL keys never match.
+ *
+ * Locality design (MI355X: random HBM lines are ~7x slower than L3-resident
+ * access — measured in tools/diag_hash.py): the slot index is the TOP bits
+ * of Murmur3(key,42) (multiplicative map, monotone in hash), and when keys
+ * have no NULLs both sides are first radix-partitioned into hash order
+ * (one ranked-scatter pass, BIN_MODE 2). Blocks then process contiguous
+ * chunks of the hash-ordered stream, so the live table region at any
+ * instant is a small hash-contiguous window that stays L3-resident, and
+ * next[]/rid side arrays are bucket-local. Probe reserves output space
+ * with one wave-aggregated atomic.
  */
 
 #define JOIN_NIL 0xFFFFFFFFu
+#define JOIN_CHUNK 4096
 
-struct join_sp { unsigned int m1_head; unsigned long long cursor; };
+struct join_sp { unsigned int m1_head; unsigned int bucketed; unsigned long long cursor; };
+
+/* hash -> slot: monotone multiplicative map onto [0, cap) */
+DEV uint64_t join_slot(int64_t key, int64_t cap_mask) {
+  uint32_t h = (uint32_t)mm3_hash_long(key, 42);
+  return ((uint64_t)h * (uint64_t)(cap_mask + 1)) >> 32;
+}
 
 struct join_ws {
   unsigned long long* slots;   /* 2 u64 per slot: [key][head|pad] */
   unsigned int* next;
+  unsigned int* brid;          /* partitioned row id map (bucketed path) */
+  uint64_t* pk_a; uint32_t* pi_a;   /* partition scratch: pairs in */
+  uint64_t* pk_b;                   /* partitioned keys out (ids go to brid) */
+  uint32_t* hist; uint32_t* hist_scan; uint32_t* block_sums;
   join_sp* sp;
 };
 
 static void join_ws_layout(int64_t cap, int64_t brows, join_ws* w, char* base, int64_t* total) {
+  int tile = get_sort_geom().block * get_sort_geom().items;
+  int64_t nb = sort_nblocks(brows, tile);
+  int64_t hist_n = 256 * nb;
+  int64_t scan_blocks = (hist_n + SCAN_TILE - 1) / SCAN_TILE + 1;
   int64_t off = 0;
   auto take = [&](int64_t bytes) {
     char* p = base ? base + off : nullptr;
@@ -1142,6 +1165,13 @@ static void join_ws_layout(int64_t cap, int64_t brows, join_ws* w, char* base, i
   };
   w->slots = (unsigned long long*)take(cap * 16);
   w->next = (unsigned int*)take(brows * 4);
+  w->brid = (unsigned int*)take(brows * 4);
+  w->pk_a = (uint64_t*)take(brows * 8);
+  w->pi_a = (uint32_t*)take(brows * 4);
+  w->pk_b = (uint64_t*)take(brows * 8);
+  w->hist = (uint32_t*)take(hist_n * 4);
+  w->hist_scan = (uint32_t*)take(hist_n * 4);
+  w->block_sums = (uint32_t*)take(scan_blocks * 4);
   w->sp = (join_sp*)take(sizeof(join_sp));
   *total = off;
 }
@@ -1152,8 +1182,77 @@ extern "C" int64_t gpuq_join_build_workspace_bytes(int64_t brows, int64_t cap) {
   return total;
 }
 
+/* probe-side scratch (hash-ordered probe stream) */
+struct probe_ws {
+  uint64_t* pk_a; uint32_t* pi_a;
+  uint64_t* pk_b; uint32_t* pi_b;
+  uint32_t* hist; uint32_t* hist_scan; uint32_t* block_sums;
+};
+
+static void probe_ws_layout(int64_t prows, probe_ws* w, char* base, int64_t* total) {
+  int tile = get_sort_geom().block * get_sort_geom().items;
+  int64_t nb = sort_nblocks(prows, tile);
+  int64_t hist_n = 256 * nb;
+  int64_t scan_blocks = (hist_n + SCAN_TILE - 1) / SCAN_TILE + 1;
+  int64_t off = 0;
+  auto take = [&](int64_t bytes) {
+    char* p = base ? base + off : nullptr;
+    off += (bytes + 255) & ~255LL;
+    return p;
+  };
+  w->pk_a = (uint64_t*)take(prows * 8);
+  w->pi_a = (uint32_t*)take(prows * 4);
+  w->pk_b = (uint64_t*)take(prows * 8);
+  w->pi_b = (uint32_t*)take(prows * 4);
+  w->hist = (uint32_t*)take(hist_n * 4);
+  w->hist_scan = (uint32_t*)take(hist_n * 4);
+  w->block_sums = (uint32_t*)take(scan_blocks * 4);
+  *total = off;
+}
+
+extern "C" int64_t gpuq_join_probe_workspace_bytes(int64_t prows) {
+  probe_ws w; int64_t total;
+  probe_ws_layout(prows, &w, nullptr, &total);
+  return total;
+}
+
+__global__ void k_make_pairs(int64_t n, const int64_t* keys, uint64_t* out_k,
+                             uint32_t* out_i) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    out_k[i] = (uint64_t)keys[i];
+    out_i[i] = (uint32_t)i;
+  }
+}
+
+/* one hash-order partition pass over (key, rowid) pairs */
+static int hash_order_pairs(hipStream_t s, int64_t n, const int64_t* keys,
+                            uint64_t* tmp_k, uint32_t* tmp_i,
+                            uint64_t* out_k, uint32_t* out_i,
+                            uint32_t* hist, uint32_t* hist_scan,
+                            uint32_t* block_sums) {
+  scatter_geom geom = get_sort_geom();
+  int tile = geom.block * geom.items;
+  int64_t nb = sort_nblocks(n, tile);
+  k_make_pairs<<<grid1d(n), 256, 0, s>>>(n, keys, tmp_k, tmp_i);
+  HIP_TRY(hipGetLastError());
+  { hipEvent_t _pe = prof_begin(s);
+  k_radix_hist<2><<<dim3((uint32_t)nb), 256, 0, s>>>(n, tmp_k, 0, hist, (int)nb, tile);
+  prof_end("join_part_hist", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  int rc = exclusive_scan_u32(s, 256 * nb, hist, hist_scan, block_sums);
+  if (rc) return rc;
+  { hipEvent_t _pe = prof_begin(s);
+  launch_scatter<2, false>(s, geom, nb, n, tmp_k, tmp_i, out_k, out_i,
+                           hist_scan, 0, 0, nullptr, nullptr, nullptr, 0);
+  prof_end("join_part_scatter", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
 /* chain-head word: bit 31 = MULTI (slot holds >1 row) so single-match
- * probes never touch next[]; low 31 bits = first build row id */
+ * probes never touch next[]; low 31 bits = first build row index */
 #define JOIN_MULTI 0x80000000u
 
 DEV void join_push_head(unsigned int* headp, unsigned int i, unsigned int* next) {
@@ -1170,19 +1269,22 @@ DEV void join_push_head(unsigned int* headp, unsigned int i, unsigned int* next)
   }
 }
 
+/* contiguous-chunk mapping: block b owns rows [b*CHUNK, (b+1)*CHUNK) so the
+ * live window of a hash-ordered stream is hash-contiguous (L3-resident) */
 __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                              unsigned long long* slots, unsigned int* next,
                              join_sp* sp, int64_t cap_mask) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
+  int64_t base = (int64_t)blockIdx.x * JOIN_CHUNK;
+  for (int r = 0; r < JOIN_CHUNK / 256; r++) {
+    int64_t i = base + r * 256 + threadIdx.x;
+    if (i >= n) return;
     if (!bit_valid(kvalid, i)) continue;  /* NULL never matches (inner join) */
     int64_t k = keys[i];
     if ((unsigned long long)k == AGG_EMPTY) {
       join_push_head(&sp->m1_head, (unsigned int)i, next);
       continue;
     }
-    uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
+    uint64_t slot = join_slot(k, cap_mask);
     for (;;) {
       unsigned long long cur = __hip_atomic_load(&slots[2 * slot], __ATOMIC_RELAXED,
                                                  __HIP_MEMORY_SCOPE_AGENT);
@@ -1199,21 +1301,21 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
 
 __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                              const unsigned long long* slots, const unsigned int* next,
+                             const unsigned int* brid_map, const unsigned int* prid_map,
                              join_sp* sp, int64_t cap_mask,
                              uint32_t* out_p, uint32_t* out_b, int64_t out_cap) {
   const int lane = threadIdx.x & (WAVE - 1);
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  /* block-uniform loop bound so the whole wave stays converged for the
-   * aggregated cursor reservation */
-  for (int64_t ib = (int64_t)blockIdx.x * blockDim.x; ib < n; ib += stride) {
-    int64_t i = ib + threadIdx.x;
+  int64_t base = (int64_t)blockIdx.x * JOIN_CHUNK;
+  for (int r = 0; r < JOIN_CHUNK / 256; r++) {
+    int64_t i = base + r * 256 + threadIdx.x;
+    if (base + r * 256 >= n) return;      /* block-uniform exit */
     unsigned int head = JOIN_NIL;
     if (i < n && bit_valid(kvalid, i)) {
       int64_t k = keys[i];
       if ((unsigned long long)k == AGG_EMPTY) {
         head = sp->m1_head;
       } else {
-        uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
+        uint64_t slot = join_slot(k, cap_mask);
         for (;;) {
           unsigned long long cur = slots[2 * slot];
           if (cur == AGG_EMPTY) break;
@@ -1225,9 +1327,8 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
         }
       }
     }
-    /* count matches; the single-match fast path (no MULTI bit) never reads
-     * next[]; the first two chain entries are cached in registers so the
-     * emit pass rarely re-walks */
+    /* count matches; single-match fast path (no MULTI bit) never reads
+     * next[]; first two chain entries cached in registers */
     uint32_t cnt = 0;
     unsigned int c0 = JOIN_NIL, c1 = JOIN_NIL;
     if (head != JOIN_NIL) {
@@ -1243,17 +1344,27 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
     /* wave-aggregated reservation: one atomic per wave */
     uint32_t incl = wave_inclusive_scan(cnt);
     uint32_t total = __shfl(incl, WAVE - 1);
-    unsigned long long base = 0;
+    unsigned long long obase = 0;
     if (lane == WAVE - 1 && total > 0)
-      base = atomicAdd(&sp->cursor, (unsigned long long)total);
-    base = __shfl(base, WAVE - 1);
-    int64_t o = (int64_t)base + (incl - cnt);
-    if (cnt >= 1 && o < out_cap) { out_p[o] = (uint32_t)i; out_b[o] = c0; }
-    if (cnt >= 2 && o + 1 < out_cap) { out_p[o + 1] = (uint32_t)i; out_b[o + 1] = c1; }
+      obase = atomicAdd(&sp->cursor, (unsigned long long)total);
+    obase = __shfl(obase, WAVE - 1);
+    int64_t o = (int64_t)obase + (incl - cnt);
+    uint32_t pr = (cnt && prid_map) ? prid_map[i] : (uint32_t)i;
+    if (cnt >= 1 && o < out_cap) {
+      out_p[o] = pr;
+      out_b[o] = brid_map ? brid_map[c0] : c0;
+    }
+    if (cnt >= 2 && o + 1 < out_cap) {
+      out_p[o + 1] = pr;
+      out_b[o + 1] = brid_map ? brid_map[c1] : c1;
+    }
     if (cnt > 2) {
       unsigned int b = next[c1];
       for (uint32_t j = 2; j < cnt; j++, b = next[b]) {
-        if (o + j < out_cap) { out_p[o + j] = (uint32_t)i; out_b[o + j] = b; }
+        if (o + j < out_cap) {
+          out_p[o + j] = pr;
+          out_b[o + j] = brid_map ? brid_map[b] : b;
+        }
       }
     }
   }
@@ -1270,20 +1381,34 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   join_ws_layout(cap, brows, &w, (char*)workspace, &need);
   HIP_TRY(hipMemsetAsync(w.slots, 0xFF, cap * 16, s));  /* keys=-1, heads=NIL */
   HIP_TRY(hipMemsetAsync(w.sp, 0xFF, 4, s));            /* m1_head = NIL */
+  HIP_TRY(hipMemsetAsync(&w.sp->bucketed, 0, 4, s));
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
-  if (brows > 0) {
-    { hipEvent_t _pe = prof_begin(s);
-    k_join_build<<<hash_grid(brows), 256, 0, s>>>(brows, (const int64_t*)bkey.data,
-                                               bkey.validity, w.slots, w.next,
-                                               w.sp, cap - 1);
-    prof_end("join_build", s, _pe); }
-    HIP_TRY(hipGetLastError());
+  if (brows == 0) return GPUQ_OK;
+  /* bucketed (hash-ordered) path when there are no NULLs and the table is
+   * past L3 size; flat otherwise */
+  bool bucketed = bkey.validity == nullptr && cap * 16 > (256LL << 20) &&
+                  getenv("GPUQ_NO_BUCKET_JOIN") == nullptr;
+  const int64_t* keys = (const int64_t*)bkey.data;
+  if (bucketed) {
+    int rc = hash_order_pairs(s, brows, keys, w.pk_a, w.pi_a, w.pk_b, w.brid,
+                              w.hist, w.hist_scan, w.block_sums);
+    if (rc) return rc;
+    keys = (const int64_t*)w.pk_b;
+    HIP_TRY(hipMemsetAsync(&w.sp->bucketed, 1, 1, s));
   }
+  int64_t nchunks = (brows + JOIN_CHUNK - 1) / JOIN_CHUNK;
+  { hipEvent_t _pe = prof_begin(s);
+  k_join_build<<<dim3((uint32_t)nchunks), 256, 0, s>>>(
+      brows, keys, bucketed ? nullptr : bkey.validity, w.slots, w.next, w.sp,
+      cap - 1);
+  prof_end("join_build", s, _pe); }
+  HIP_TRY(hipGetLastError());
   return GPUQ_OK;
 }
 
 extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
                                    const void* workspace, int64_t cap, int64_t brows,
+                                   void* probe_workspace, int64_t probe_ws_bytes,
                                    uint32_t* out_p, uint32_t* out_b,
                                    int64_t out_cap, int64_t* out_nmatches) {
   hipStream_t s = (hipStream_t)stream;
@@ -1292,11 +1417,36 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
   join_ws_layout(cap, brows, &w, (char*)workspace, &need);
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
   if (prows > 0) {
+    join_sp hsp0;
+    HIP_TRY(hipMemcpyAsync(&hsp0, w.sp, sizeof(hsp0), hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    bool build_bucketed = hsp0.bucketed != 0;
+    const int64_t* pkeys = (const int64_t*)pkey.data;
+    const uint8_t* pvalid = pkey.validity;
+    const unsigned int* prid_map = nullptr;
+    probe_ws pw;
+    /* order the probe stream by hash too (same locality argument) */
+    bool probe_bucketed = build_bucketed && pvalid == nullptr &&
+                          probe_workspace != nullptr;
+    if (probe_bucketed) {
+      int64_t pneed;
+      probe_ws_layout(prows, &pw, (char*)probe_workspace, &pneed);
+      if (probe_ws_bytes < pneed)
+        FAIL(GPUQ_ERR_INVALID, "join: probe workspace %lld < %lld",
+             (long long)probe_ws_bytes, (long long)pneed);
+      int rc = hash_order_pairs(s, prows, pkeys, pw.pk_a, pw.pi_a, pw.pk_b,
+                                pw.pi_b, pw.hist, pw.hist_scan, pw.block_sums);
+      if (rc) return rc;
+      pkeys = (const int64_t*)pw.pk_b;
+      prid_map = pw.pi_b;
+      pvalid = nullptr;
+    }
+    int64_t nchunks = (prows + JOIN_CHUNK - 1) / JOIN_CHUNK;
     { hipEvent_t _pe = prof_begin(s);
-    k_join_probe<<<hash_grid(prows), 256, 0, s>>>(prows, (const int64_t*)pkey.data,
-                                               pkey.validity, w.slots, w.next,
-                                               w.sp, cap - 1,
-                                               out_p, out_b, out_cap);
+    k_join_probe<<<dim3((uint32_t)nchunks), 256, 0, s>>>(
+        prows, pkeys, pvalid, w.slots, w.next,
+        build_bucketed ? w.brid : nullptr, prid_map, w.sp, cap - 1,
+        out_p, out_b, out_cap);
     prof_end("join_probe", s, _pe); }
     HIP_TRY(hipGetLastError());
   }

@@ -72,11 +72,13 @@ def lib() -> ctypes.CDLL:
         L.gpuq_partition_perm.argtypes = [vp, i64, _Col, i32, vp, vp, vp, i64]
         L.gpuq_join_build_workspace_bytes.restype = i64
         L.gpuq_join_build_workspace_bytes.argtypes = [i64, i64]
+        L.gpuq_join_probe_workspace_bytes.restype = i64
+        L.gpuq_join_probe_workspace_bytes.argtypes = [i64]
         L.gpuq_join_build_i64.restype = i32
         L.gpuq_join_build_i64.argtypes = [vp, i64, _Col, vp, i64]
         L.gpuq_join_probe_i64.restype = i32
-        L.gpuq_join_probe_i64.argtypes = [vp, i64, _Col, vp, i64, i64, vp, vp, i64,
-                                          ctypes.POINTER(i64)]
+        L.gpuq_join_probe_i64.argtypes = [vp, i64, _Col, vp, i64, i64, vp, i64,
+                                          vp, vp, i64, ctypes.POINTER(i64)]
         L.gpuq_profiling.restype = None
         L.gpuq_profiling.argtypes = [i32]
         L.gpuq_kernel_stats_reset.restype = None
@@ -229,15 +231,24 @@ def join_build(build_keys: torch.Tensor, capacity: int, workspace=None,
     return workspace
 
 
+def join_probe_workspace(probe_rows: int, device="cuda") -> torch.Tensor:
+    return torch.empty(lib().gpuq_join_probe_workspace_bytes(probe_rows),
+                       dtype=torch.uint8, device=device)
+
+
 def join_probe(probe_keys: torch.Tensor, workspace: torch.Tensor, capacity: int,
-               build_rows: int, out_cap: int, key_validity=None):
+               build_rows: int, out_cap: int, key_validity=None, probe_ws=None):
     pn = probe_keys.numel()
     dev = probe_keys.device
+    if probe_ws is None and key_validity is None:
+        probe_ws = join_probe_workspace(pn, dev)
     op = torch.empty(out_cap, dtype=torch.int32, device=dev)
     ob = torch.empty(out_cap, dtype=torch.int32, device=dev)
     nm = ctypes.c_int64(0)
     rc = lib().gpuq_join_probe_i64(_stream(), pn, _col(probe_keys, key_validity),
                                    workspace.data_ptr(), capacity, build_rows,
+                                   _dp(probe_ws),
+                                   probe_ws.numel() if probe_ws is not None else 0,
                                    op.data_ptr(), ob.data_ptr(), out_cap,
                                    ctypes.byref(nm))
     if rc == 3:  # GPUQ_ERR_OVERFLOW: caller retries with nm.value capacity
