@@ -1132,7 +1132,9 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
  */
 
 #define JOIN_NIL 0xFFFFFFFFu
+#ifndef JOIN_CHUNK
 #define JOIN_CHUNK 4096
+#endif
 
 struct join_sp { unsigned int m1_head; unsigned int bucketed; unsigned long long cursor; };
 
@@ -1304,11 +1306,22 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
                              const unsigned int* brid_map, const unsigned int* prid_map,
                              join_sp* sp, int64_t cap_mask,
                              uint32_t* out_p, uint32_t* out_b, int64_t out_cap) {
+  constexpr int ROUNDS = JOIN_CHUNK / 256;
+  __shared__ unsigned long long block_base;
+  __shared__ uint32_t wtot[JOIN_CHUNK / 256 > 4 ? 16 : 16];
   const int lane = threadIdx.x & (WAVE - 1);
-  int64_t base = (int64_t)blockIdx.x * JOIN_CHUNK;
-  for (int r = 0; r < JOIN_CHUNK / 256; r++) {
+  const int wave = threadIdx.x / WAVE;
+  const int64_t base = (int64_t)blockIdx.x * JOIN_CHUNK;
+
+  /* phase A: all lookups; match counts + first two matches stay in
+   * registers. ONE output-cursor atomic per block (a single global cursor
+   * word saturates near ~88 updates/us — the per-wave-iteration version
+   * was the whole kernel's bottleneck). */
+  uint32_t cnt[ROUNDS];
+  unsigned int c0[ROUNDS], c1[ROUNDS];
+  uint32_t lane_total = 0;
+  for (int r = 0; r < ROUNDS; r++) {
     int64_t i = base + r * 256 + threadIdx.x;
-    if (base + r * 256 >= n) return;      /* block-uniform exit */
     unsigned int head = JOIN_NIL;
     if (i < n && bit_valid(kvalid, i)) {
       int64_t k = keys[i];
@@ -1327,44 +1340,45 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
         }
       }
     }
-    /* count matches; single-match fast path (no MULTI bit) never reads
-     * next[]; first two chain entries cached in registers */
-    uint32_t cnt = 0;
-    unsigned int c0 = JOIN_NIL, c1 = JOIN_NIL;
+    cnt[r] = 0; c0[r] = JOIN_NIL; c1[r] = JOIN_NIL;
     if (head != JOIN_NIL) {
       if (!(head & JOIN_MULTI)) {
-        c0 = head; cnt = 1;
+        c0[r] = head; cnt[r] = 1;
       } else {
         for (unsigned int b = head & ~JOIN_MULTI; b != JOIN_NIL; b = next[b]) {
-          if (cnt == 0) c0 = b; else if (cnt == 1) c1 = b;
-          cnt++;
+          if (cnt[r] == 0) c0[r] = b; else if (cnt[r] == 1) c1[r] = b;
+          cnt[r]++;
         }
       }
     }
-    /* wave-aggregated reservation: one atomic per wave */
-    uint32_t incl = wave_inclusive_scan(cnt);
-    uint32_t total = __shfl(incl, WAVE - 1);
-    unsigned long long obase = 0;
-    if (lane == WAVE - 1 && total > 0)
-      obase = atomicAdd(&sp->cursor, (unsigned long long)total);
-    obase = __shfl(obase, WAVE - 1);
-    int64_t o = (int64_t)obase + (incl - cnt);
-    uint32_t pr = (cnt && prid_map) ? prid_map[i] : (uint32_t)i;
-    if (cnt >= 1 && o < out_cap) {
-      out_p[o] = pr;
-      out_b[o] = brid_map ? brid_map[c0] : c0;
+    lane_total += cnt[r];
+  }
+  /* phase B: block-level reservation (emit order within the block is
+   * arbitrary — join output order is nondeterministic by contract) */
+  uint32_t incl = wave_inclusive_scan(lane_total);
+  if (lane == WAVE - 1) wtot[wave] = incl;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint32_t tot = 0;
+    for (int w = 0; w < (int)(blockDim.x / WAVE); w++) {
+      uint32_t t = wtot[w]; wtot[w] = tot; tot += t;
     }
-    if (cnt >= 2 && o + 1 < out_cap) {
-      out_p[o + 1] = pr;
-      out_b[o + 1] = brid_map ? brid_map[c1] : c1;
-    }
-    if (cnt > 2) {
-      unsigned int b = next[c1];
-      for (uint32_t j = 2; j < cnt; j++, b = next[b]) {
-        if (o + j < out_cap) {
-          out_p[o + j] = pr;
-          out_b[o + j] = brid_map ? brid_map[b] : b;
-        }
+    block_base = tot ? atomicAdd(&sp->cursor, (unsigned long long)tot) : 0;
+  }
+  __syncthreads();
+  int64_t o = (int64_t)block_base + wtot[wave] + (incl - lane_total);
+  for (int r = 0; r < ROUNDS; r++) {
+    if (!cnt[r]) continue;
+    int64_t i = base + r * 256 + threadIdx.x;
+    uint32_t pr = prid_map ? prid_map[i] : (uint32_t)i;
+    if (o < out_cap) { out_p[o] = pr; out_b[o] = brid_map ? brid_map[c0[r]] : c0[r]; }
+    o++;
+    if (cnt[r] >= 2) {
+      if (o < out_cap) { out_p[o] = pr; out_b[o] = brid_map ? brid_map[c1[r]] : c1[r]; }
+      o++;
+      unsigned int b = (cnt[r] > 2) ? next[c1[r]] : JOIN_NIL;
+      for (uint32_t j = 2; j < cnt[r]; j++, b = next[b], o++) {
+        if (o < out_cap) { out_p[o] = pr; out_b[o] = brid_map ? brid_map[b] : b; }
       }
     }
   }
