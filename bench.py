@@ -147,7 +147,7 @@ class JoinWorkload:
         self.join_ws = torch.empty(
             gq.lib().gpuq_join_build_workspace_bytes(self.local_cap_rows, self.cap),
             dtype=torch.uint8, device="cuda")
-        self.probe_ws = gq.join_probe_workspace(self.local_cap_rows)
+        self.probe_ws = None  # hash-order probe bucketing is opt-in (see gpuq.hip)
         self.out_cap = int(rows * 2.5) + 4096
         self.nmatches = 0
 

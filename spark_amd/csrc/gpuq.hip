@@ -1006,49 +1006,66 @@ __global__ void k_agg_build(int64_t n, const int64_t* keys, const uint8_t* kvali
   }
 }
 
+#define AGGC_CHUNK 8192
+
 template <int OPS>
 __global__ void k_agg_compact(int64_t cap, const unsigned long long* tab,
                               agg_special* sp,
                               int64_t* out_keys, uint8_t* out_kvalid,
                               double* out_sums, uint8_t* out_svalid,
                               int64_t* out_cnts) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t ib = (int64_t)blockIdx.x * blockDim.x; ib < cap;
-       ib += stride, i += stride) {
-    unsigned long long k = (i < cap) ? tab[3 * i] : AGG_EMPTY;
-    bool occ = (k != AGG_EMPTY);
-    uint64_t mask = __ballot(occ);
-    int rank = __popcll(mask & ((1ULL << (threadIdx.x & (WAVE - 1))) - 1));
-    int leader = __ffsll((unsigned long long)mask) - 1;
-    unsigned long long base = 0;
-    if (occ && (int)(threadIdx.x & (WAVE - 1)) == leader)
-      base = atomicAdd(&sp->out_cursor, (unsigned long long)__popcll(mask));
-    if (mask) base = __shfl(base, leader);
-    if (occ) {
-      int64_t o = (int64_t)base + rank;
-      out_keys[o] = (int64_t)k;
-      out_kvalid[o] = 1;
-      out_sums[o] = __longlong_as_double((long long)tab[3 * i + 1]);
-      /* without COUNT, values were required non-null => every group has one */
-      out_svalid[o] = (OPS & AGG_OP_COUNT) ? (tab[3 * i + 2] > 0 ? 1 : 0) : 1;
-      if (out_cnts) out_cnts[o] = (int64_t)tab[3 * i + 2];
-    }
+  /* ONE output-cursor atomic per block (a single cursor word saturates
+   * near ~88 updates/us): count occupied slots in this block's chunk,
+   * block-scan, reserve once, emit. */
+  constexpr int ROUNDS = AGGC_CHUNK / 256;
+  __shared__ unsigned long long block_base;
+  __shared__ uint32_t wtot[4];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int64_t base = (int64_t)blockIdx.x * AGGC_CHUNK;
+  uint32_t occ_mask[ROUNDS / 32 + 1];
+  for (int m = 0; m < ROUNDS / 32 + 1; m++) occ_mask[m] = 0;
+  uint32_t lane_total = 0;
+  for (int r = 0; r < ROUNDS; r++) {
+    int64_t i = base + r * 256 + threadIdx.x;
+    bool occ = i < cap && tab[3 * i] != AGG_EMPTY;
+    if (occ) { occ_mask[r / 32] |= 1u << (r & 31); lane_total++; }
+  }
+  uint32_t incl = wave_inclusive_scan(lane_total);
+  if (lane == WAVE - 1) wtot[wave] = incl;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint32_t tot = 0;
+    for (int w = 0; w < 4; w++) { uint32_t t = wtot[w]; wtot[w] = tot; tot += t; }
+    block_base = tot ? atomicAdd(&sp->out_cursor, (unsigned long long)tot) : 0;
+  }
+  __syncthreads();
+  int64_t o = (int64_t)block_base + wtot[wave] + (incl - lane_total);
+  for (int r = 0; r < ROUNDS; r++) {
+    if (!((occ_mask[r / 32] >> (r & 31)) & 1)) continue;
+    int64_t i = base + r * 256 + threadIdx.x;
+    unsigned long long k = tab[3 * i];
+    out_keys[o] = (int64_t)k;
+    out_kvalid[o] = 1;
+    out_sums[o] = __longlong_as_double((long long)tab[3 * i + 1]);
+    out_svalid[o] = (OPS & AGG_OP_COUNT) ? (tab[3 * i + 2] > 0 ? 1 : 0) : 1;
+    if (out_cnts) out_cnts[o] = (int64_t)tab[3 * i + 2];
+    o++;
   }
   if (blockIdx.x == 0 && threadIdx.x == 0) {
     if (sp->m1_seen) {
-      int64_t o = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
-      out_keys[o] = -1; out_kvalid[o] = 1;
-      out_sums[o] = sp->m1_sum;
-      out_svalid[o] = (OPS & AGG_OP_COUNT) ? (sp->m1_cnt > 0 ? 1 : 0) : 1;
-      if (out_cnts) out_cnts[o] = (int64_t)sp->m1_cnt;
+      int64_t q = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
+      out_keys[q] = -1; out_kvalid[q] = 1;
+      out_sums[q] = sp->m1_sum;
+      out_svalid[q] = (OPS & AGG_OP_COUNT) ? (sp->m1_cnt > 0 ? 1 : 0) : 1;
+      if (out_cnts) out_cnts[q] = (int64_t)sp->m1_cnt;
     }
     if (sp->nul_seen) {
-      int64_t o = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
-      out_keys[o] = 0; out_kvalid[o] = 0;
-      out_sums[o] = sp->nul_sum;
-      out_svalid[o] = (OPS & AGG_OP_COUNT) ? (sp->nul_cnt > 0 ? 1 : 0) : 1;
-      if (out_cnts) out_cnts[o] = (int64_t)sp->nul_cnt;
+      int64_t q = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
+      out_keys[q] = 0; out_kvalid[q] = 0;
+      out_sums[q] = sp->nul_sum;
+      out_svalid[q] = (OPS & AGG_OP_COUNT) ? (sp->nul_cnt > 0 ? 1 : 0) : 1;
+      if (out_cnts) out_cnts[q] = (int64_t)sp->nul_cnt;
     }
   }
 }
@@ -1094,12 +1111,13 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
     HIP_TRY(hipStreamSynchronize(s));
     if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "agg: hash table overflow (capacity %lld)", (long long)cap);
     { hipEvent_t _pe = prof_begin(s);
+    dim3 cgrid((uint32_t)((cap + AGGC_CHUNK - 1) / AGGC_CHUNK));
     if (ops == AGG_OP_SUM)
-      k_agg_compact<AGG_OP_SUM><<<grid1d(cap), 256, 0, s>>>(
+      k_agg_compact<AGG_OP_SUM><<<cgrid, 256, 0, s>>>(
           cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid,
           out_counts);
     else
-      k_agg_compact<AGG_OP_SUM | AGG_OP_COUNT><<<grid1d(cap), 256, 0, s>>>(
+      k_agg_compact<AGG_OP_SUM | AGG_OP_COUNT><<<cgrid, 256, 0, s>>>(
           cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid,
           out_counts);
     prof_end("agg_compact", s, _pe); }
@@ -1400,8 +1418,11 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   if (brows == 0) return GPUQ_OK;
   /* bucketed (hash-ordered) path when there are no NULLs and the table is
    * past L3 size; flat otherwise */
+  /* hash-order bucketing measured as net overhead once the output-cursor
+   * bottleneck was removed (probe is latency-, not line-refetch-bound);
+   * kept behind an opt-in env for future asymmetric shapes */
   bool bucketed = bkey.validity == nullptr && cap * 16 > (256LL << 20) &&
-                  getenv("GPUQ_NO_BUCKET_JOIN") == nullptr;
+                  getenv("GPUQ_BUCKET_JOIN") != nullptr;
   const int64_t* keys = (const int64_t*)bkey.data;
   if (bucketed) {
     int rc = hash_order_pairs(s, brows, keys, w.pk_a, w.pi_a, w.pk_b, w.brid,

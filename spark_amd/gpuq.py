@@ -240,8 +240,6 @@ def join_probe(probe_keys: torch.Tensor, workspace: torch.Tensor, capacity: int,
                build_rows: int, out_cap: int, key_validity=None, probe_ws=None):
     pn = probe_keys.numel()
     dev = probe_keys.device
-    if probe_ws is None and key_validity is None:
-        probe_ws = join_probe_workspace(pn, dev)
     op = torch.empty(out_cap, dtype=torch.int32, device=dev)
     ob = torch.empty(out_cap, dtype=torch.int32, device=dev)
     nm = ctypes.c_int64(0)
