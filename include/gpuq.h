@@ -93,7 +93,9 @@ int64_t gpuq_sort_workspace_bytes(int64_t nrows);
 
 /* Sort one key column; writes the sorted permutation (row ids into the
  * input) to out_perm[nrows] (uint32) and, if out_keys != NULL, the sorted
- * key column. key.validity may be NULL (no nulls). */
+ * key column. With key.validity set, NULL rows are placed per nulls_first
+ * in input order and out_keys carries the input's raw values at those
+ * positions (callers track NULL-ness via the permuted validity). */
 int gpuq_sort_perm(void* stream, int64_t nrows, gpuq_col key,
                    int32_t desc, int32_t nulls_first,
                    uint32_t* out_perm, void* out_keys,

@@ -266,7 +266,8 @@ extern "C" int gpuq_gather(void* stream, int64_t n, gpuq_col col,
  * count loops, RadixSort.java:126-135; per-byte histograms are invariant
  * under the passes' permutations, so one upfront count serves every pass). */
 template <int DTYPE, bool DESC>
-__global__ void k_encode(int64_t n, const void* keys, uint64_t* ek, uint32_t* idx,
+__global__ void k_encode(int64_t n, const void* keys, const uint32_t* rowmap,
+                         uint64_t* ek, uint32_t* idx,
                          unsigned long long* bits_and, unsigned long long* bits_or,
                          uint32_t* ghist /* [8][256] */) {
   __shared__ uint32_t h[8][256];
@@ -277,11 +278,12 @@ __global__ void k_encode(int64_t n, const void* keys, uint64_t* ek, uint32_t* id
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
     uint64_t e;
-    if (DTYPE == GPUQ_FLOAT64) e = encode_f64(((const double*)keys)[i]);
-    else e = encode_i64(((const int64_t*)keys)[i]);
+    uint32_t row = rowmap ? rowmap[i] : (uint32_t)i;
+    if (DTYPE == GPUQ_FLOAT64) e = encode_f64(((const double*)keys)[row]);
+    else e = encode_i64(((const int64_t*)keys)[row]);
     if (DESC) e = ~e;
     ek[i] = e;
-    idx[i] = (uint32_t)i;
+    idx[i] = row;
     acc_or |= e; acc_and &= e;
     #pragma unroll
     for (int b = 0; b < 8; b++) atomicAdd(&h[b][(e >> (b * 8)) & 0xff], 1u);
@@ -316,6 +318,29 @@ DEV int compute_bin(uint64_t key, int shift, int nparts) {
   /* BIN_MODE 2: top 8 bits of Murmur3(key,42) — hash-order bucketing so a
    * bucketed stream sweeps a hash-ordered table monotonically (join) */
   return (int)(((uint32_t)mm3_hash_long((int64_t)key, 42)) >> 24);
+}
+
+/* validity split for sort-with-nulls: pair key = 0 for the group that
+ * sorts first (nulls when nulls_first), 1 for the other; also counts the
+ * null rows (SortExec null ordering, SortOrder.scala:35-45) */
+__global__ void k_validity_pids(int64_t n, const uint8_t* validity, int nulls_first,
+                                uint64_t* pid_as_key, uint32_t* idx,
+                                unsigned long long* null_count) {
+  __shared__ unsigned int h;
+  if (threadIdx.x == 0) h = 0;
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  unsigned int mine = 0;
+  for (; i < n; i += stride) {
+    bool valid = (validity[i >> 3] >> (i & 7)) & 1;
+    pid_as_key[i] = (uint64_t)(valid == (bool)nulls_first);
+    idx[i] = (uint32_t)i;
+    if (!valid) mine++;
+  }
+  if (mine) atomicAdd(&h, mine);
+  __syncthreads();
+  if (threadIdx.x == 0 && h) atomicAdd(null_count, (unsigned long long)h);
 }
 
 /* per-block histogram of digit at `shift` over the pass input;
@@ -708,16 +733,25 @@ extern "C" int64_t gpuq_sort_workspace_bytes(int64_t n) {
   return total;
 }
 
+static void launch_encode(hipStream_t s, int64_t n, const void* keys,
+                          const uint32_t* rowmap, int dtype, int desc, sort_ws* w) {
+  if (dtype == GPUQ_FLOAT64) {
+    if (desc) k_encode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, keys, rowmap, w->ka, w->ia, &w->bits[0], &w->bits[1], w->ghist);
+    else      k_encode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, keys, rowmap, w->ka, w->ia, &w->bits[0], &w->bits[1], w->ghist);
+  } else {
+    if (desc) k_encode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, keys, rowmap, w->ka, w->ia, &w->bits[0], &w->bits[1], w->ghist);
+    else      k_encode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, keys, rowmap, w->ka, w->ia, &w->bits[0], &w->bits[1], w->ghist);
+  }
+}
+
 extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
                               int32_t desc, int32_t nulls_first,
                               uint32_t* out_perm, void* out_keys,
                               void* workspace, int64_t workspace_bytes) {
   hipStream_t s = (hipStream_t)stream;
   if (n > 0xFFFFFFFFLL) FAIL(GPUQ_ERR_INVALID, "sort: nrows %lld > 2^32", (long long)n);
-  if (key.validity) FAIL(GPUQ_ERR_INVALID, "sort: validity not yet supported (round 1)");
   if (key.dtype != GPUQ_INT64 && key.dtype != GPUQ_FLOAT64)
     FAIL(GPUQ_ERR_INVALID, "sort: unsupported dtype %d", key.dtype);
-  (void)nulls_first;
   sort_ws w; int64_t need;
   sort_ws_layout(n, 256, &w, (char*)workspace, &need);
   if (workspace_bytes < need)
@@ -726,21 +760,56 @@ extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
 
   scatter_geom geom = get_sort_geom();
   int tile = geom.block * geom.items;
-  int64_t nb = sort_nblocks(n, tile);
-  bool onesweep = getenv("GPUQ_NO_ONESWEEP") == nullptr;
+  static __thread int force_classic = 0;
+  bool onesweep = !force_classic && getenv("GPUQ_NO_ONESWEEP") == nullptr;
 
+  /* NULL keys: stable split into [nulls][valids] (or the reverse) per
+   * nullOrdering (SortOrder.scala:35-45), then radix-sort the valid subset
+   * through a row map; the final permutation is the concatenation. */
+  int64_t n_sort = n;           /* rows entering the radix passes */
+  int64_t sorted_at = 0;        /* offset of the sorted segment in out_perm */
+  const uint32_t* rowmap = nullptr;
+  if (key.validity) {
+    int64_t nbv = sort_nblocks(n, tile);
+    HIP_TRY(hipMemsetAsync(w.err, 0, 8, s));
+    k_validity_pids<<<grid1d(n), 256, 0, s>>>(n, key.validity, nulls_first,
+                                              w.ka, w.ia, w.err);
+    HIP_TRY(hipGetLastError());
+    k_radix_hist<0><<<dim3((uint32_t)nbv), 256, 0, s>>>(n, w.ka, 0, w.hist, (int)nbv, tile);
+    HIP_TRY(hipGetLastError());
+    int rc = exclusive_scan_u32(s, 256 * nbv, w.hist, w.hist_scan, w.block_sums);
+    if (rc) return rc;
+    launch_scatter<0, false>(s, geom, nbv, n, w.ka, w.ia, w.kb, w.ib, w.hist_scan,
+                             0, 0, nullptr, nullptr, nullptr, 0);
+    HIP_TRY(hipGetLastError());
+    unsigned long long hnull = 0;
+    HIP_TRY(hipMemcpyAsync(&hnull, w.err, 8, hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    int64_t nnull = (int64_t)hnull;
+    n_sort = n - nnull;
+    sorted_at = nulls_first ? nnull : 0;
+    int64_t nulls_at_out = nulls_first ? 0 : n_sort;      /* in out_perm */
+    int64_t nulls_at_ib = nulls_first ? 0 : n_sort;       /* in w.ib (same order) */
+    if (nnull > 0)
+      HIP_TRY(hipMemcpyAsync(out_perm + nulls_at_out, w.ib + nulls_at_ib,
+                             nnull * 4, hipMemcpyDeviceToDevice, s));
+    rowmap = w.ib + sorted_at;   /* valid rows, input order; consumed by encode */
+    if (n_sort == 0) {
+      if (out_keys) {
+        gpuq_col kc = key; kc.validity = nullptr;
+        return gpuq_gather(stream, n, kc, out_perm, out_keys);
+      }
+      return GPUQ_OK;
+    }
+  }
+
+  int64_t nb = sort_nblocks(n_sort, tile);
   HIP_TRY(hipMemsetAsync(w.bits, 0, 16, s));
   HIP_TRY(hipMemsetAsync(w.bits, 0xFF, 8, s));  /* bits_and = ~0 */
   HIP_TRY(hipMemsetAsync(w.ghist, 0, 8 * 256 * 4, s));
   /* encode + skip-byte reduction + all-byte global histograms (one read) */
   { hipEvent_t _pe = prof_begin(s);
-  if (key.dtype == GPUQ_FLOAT64) {
-    if (desc) k_encode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
-    else      k_encode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
-  } else {
-    if (desc) k_encode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
-    else      k_encode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
-  }
+  launch_encode(s, n_sort, key.data, rowmap, key.dtype, desc, &w);
   prof_end("encode", s, _pe); }
   HIP_TRY(hipGetLastError());
   unsigned long long hb[2];
@@ -753,8 +822,8 @@ extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
   int retries = 0;
 retry:
   if (onesweep) {
-    /* per-pass exclusive bin bases, computed on host from the one-read
-     * global histograms (replaces the per-pass hist kernel + device scan) */
+    /* per-pass exclusive bin bases from the one-read global histograms
+     * (replaces the per-pass hist kernel + device scan) */
     uint32_t hgbase[8 * 256];
     for (int b = 0; b < 8; b++) {
       uint32_t run = 0;
@@ -775,20 +844,20 @@ retry:
     int shift = byte * 8;
     if (onesweep) {
       { hipEvent_t _pe = prof_begin(s);
-      launch_scatter<0, true>(s, geom, nb, n, kin, iin, kout, iout, nullptr,
+      launch_scatter<0, true>(s, geom, nb, n_sort, kin, iin, kout, iout, nullptr,
                               shift, 0, w.state, w.gbase + byte * 256, w.err,
                               (uint32_t)(byte + 1));
       prof_end("radix_scatter", s, _pe); }
       HIP_TRY(hipGetLastError());
     } else {
       { hipEvent_t _pe = prof_begin(s);
-      k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, shift, w.hist, (int)nb, tile);
+      k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n_sort, kin, shift, w.hist, (int)nb, tile);
       prof_end("radix_hist", s, _pe); }
       HIP_TRY(hipGetLastError());
       int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
       if (rc) return rc;
       { hipEvent_t _pe = prof_begin(s);
-      launch_scatter<0, false>(s, geom, nb, n, kin, iin, kout, iout, w.hist_scan,
+      launch_scatter<0, false>(s, geom, nb, n_sort, kin, iin, kout, iout, w.hist_scan,
                                shift, 0, nullptr, nullptr, nullptr, 0);
       prof_end("radix_scatter", s, _pe); }
       HIP_TRY(hipGetLastError());
@@ -798,36 +867,47 @@ retry:
   }
 
   if (onesweep) {
-    /* one bounded-spin timeout check; on timeout redo the whole sort with
-     * the hist+scan path (input untouched - passes only wrote scratch) */
+    /* one bounded-spin timeout check; on timeout redo with hist+scan (the
+     * input was only read; pass buffers are scratch) */
     unsigned long long herr = 0;
     HIP_TRY(hipMemcpyAsync(&herr, w.err, 8, hipMemcpyDeviceToHost, s));
     HIP_TRY(hipStreamSynchronize(s));
     if (herr) {
       if (++retries > 1) FAIL(GPUQ_ERR_HIP, "sort: lookback timed out twice");
       onesweep = false;
-      /* re-encode (ka/ia were consumed as ping-pong scratch) */
-      if (key.dtype == GPUQ_FLOAT64) {
-        if (desc) k_encode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
-        else      k_encode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
-      } else {
-        if (desc) k_encode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
-        else      k_encode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, key.data, w.ka, w.ia, &w.bits[0], &w.bits[1], w.ghist);
+      /* re-encode (ka/ia were consumed as ping-pong scratch; with a validity
+       * split the rowmap in w.ib was also consumed -> rebuild is only needed
+       * for the null-free path; with validity, fall back to a full redo) */
+      if (key.validity) {
+        force_classic = 1;
+        int rc = gpuq_sort_perm(stream, n, key, desc, nulls_first, out_perm,
+                                out_keys, workspace, workspace_bytes);
+        force_classic = 0;
+        return rc;
       }
+      launch_encode(s, n_sort, key.data, nullptr, key.dtype, desc, &w);
       HIP_TRY(hipGetLastError());
       goto retry;
     }
   }
-  HIP_TRY(hipMemcpyAsync(out_perm, iin, n * 4, hipMemcpyDeviceToDevice, s));
+  HIP_TRY(hipMemcpyAsync(out_perm + sorted_at, iin, n_sort * 4, hipMemcpyDeviceToDevice, s));
   if (out_keys) {
-    if (key.dtype == GPUQ_FLOAT64) {
-      if (desc) k_decode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
-      else      k_decode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
+    if (key.validity) {
+      /* null rows' data values are whatever the input held — gather the
+       * whole column by the final permutation */
+      gpuq_col kc = key; kc.validity = nullptr;
+      int rc = gpuq_gather(stream, n, kc, out_perm, out_keys);
+      if (rc) return rc;
     } else {
-      if (desc) k_decode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
-      else      k_decode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
+      if (key.dtype == GPUQ_FLOAT64) {
+        if (desc) k_decode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
+        else      k_decode<GPUQ_FLOAT64, false><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
+      } else {
+        if (desc) k_decode<GPUQ_INT64, true><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
+        else      k_decode<GPUQ_INT64, false><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
+      }
+      HIP_TRY(hipGetLastError());
     }
-    HIP_TRY(hipGetLastError());
   }
   return GPUQ_OK;
 }
@@ -1305,14 +1385,11 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
       continue;
     }
     uint64_t slot = join_slot(k, cap_mask);
+    /* CAS-first: on the build side most inserts claim an empty slot, so one
+     * CAS does probe+claim in a single round trip */
     for (;;) {
-      unsigned long long cur = __hip_atomic_load(&slots[2 * slot], __ATOMIC_RELAXED,
-                                                 __HIP_MEMORY_SCOPE_AGENT);
-      if (cur == (unsigned long long)k) break;
-      if (cur == AGG_EMPTY) {
-        unsigned long long prev = atomicCAS(&slots[2 * slot], AGG_EMPTY, (unsigned long long)k);
-        if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
-      }
+      unsigned long long prev = atomicCAS(&slots[2 * slot], AGG_EMPTY, (unsigned long long)k);
+      if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
       slot = (slot + 1) & (uint64_t)cap_mask;
     }
     join_push_head((unsigned int*)&slots[2 * slot + 1], (unsigned int)i, next);

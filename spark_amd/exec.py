@@ -195,7 +195,8 @@ class GpuSortExec(SparkPlan):
         for batch in self.children[0].execute_columnar():
             keys = batch.column(o.key)
             perm, skeys = gpuq.sort_perm(keys, desc=o.descending,
-                                         nulls_first=o.nulls_first)
+                                         nulls_first=o.nulls_first,
+                                         key_validity=batch.validity(o.key))
             cols = {o.key: skeys}
             for name, t in batch.columns().items():
                 if name != o.key:

@@ -145,7 +145,7 @@ def sort_workspace(n: int, device="cuda") -> torch.Tensor:
 
 
 def sort_perm(keys: torch.Tensor, desc=False, nulls_first=None, workspace=None,
-              out_keys: bool = True):
+              out_keys: bool = True, key_validity=None):
     n = keys.numel()
     if nulls_first is None:
         nulls_first = not desc
@@ -153,7 +153,7 @@ def sort_perm(keys: torch.Tensor, desc=False, nulls_first=None, workspace=None,
         workspace = sort_workspace(n, keys.device)
     perm = torch.empty(n, dtype=torch.int32, device=keys.device)  # u32 bits
     ok = torch.empty(n, dtype=keys.dtype, device=keys.device) if out_keys else None
-    _check(lib().gpuq_sort_perm(_stream(), n, _col(keys), int(desc), int(nulls_first),
+    _check(lib().gpuq_sort_perm(_stream(), n, _col(keys, key_validity), int(desc), int(nulls_first),
                                 perm.data_ptr(), _dp(ok),
                                 workspace.data_ptr(), workspace.numel()))
     return perm, ok

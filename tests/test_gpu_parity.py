@@ -222,3 +222,26 @@ def test_join_overflow_reports_count(gq):
     ws = gq.join_build(to_dev(bkeys), 256)
     gp, gb, nm = gq.join_probe(to_dev(pkeys), ws, 256, 100, out_cap=10)
     assert gp is None and nm == 100 * 100
+
+
+@pytest.mark.parametrize("desc", [False, True])
+def test_sort_i64_with_nulls(gq, desc):
+    rng = np.random.default_rng(11)
+    n = 300_000
+    keys = rng.integers(-1000, 1000, n).astype(np.int64)
+    valid = rng.random(n) > 0.15
+    vbits = np.packbits(valid, bitorder="little")
+    perm, skeys = gq.sort_perm(to_dev(keys), desc=desc,
+                               key_validity=pack_validity(valid))
+    exp = oracle.sort_perm(keys, desc=desc, validity=vbits)
+    assert (perm.cpu().numpy().astype(np.uint32) == exp.astype(np.uint32)).all()
+    assert (skeys.cpu().numpy() == keys[exp]).all()
+
+
+def test_sort_all_nulls(gq):
+    n = 1000
+    keys = oracle.gen_i64(seed=3, n=n)
+    valid = np.zeros(n, dtype=bool)
+    perm, skeys = gq.sort_perm(to_dev(keys), key_validity=pack_validity(valid))
+    assert (perm.cpu().numpy() == np.arange(n)).all()
+    assert (skeys.cpu().numpy() == keys).all()
