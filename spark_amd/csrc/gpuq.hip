@@ -1385,11 +1385,14 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
       continue;
     }
     uint64_t slot = join_slot(k, cap_mask);
-    /* CAS-first: on the build side most inserts claim an empty slot, so one
-     * CAS does probe+claim in a single round trip */
     for (;;) {
-      unsigned long long prev = atomicCAS(&slots[2 * slot], AGG_EMPTY, (unsigned long long)k);
-      if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
+      unsigned long long cur = __hip_atomic_load(&slots[2 * slot], __ATOMIC_RELAXED,
+                                                 __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == (unsigned long long)k) break;
+      if (cur == AGG_EMPTY) {
+        unsigned long long prev = atomicCAS(&slots[2 * slot], AGG_EMPTY, (unsigned long long)k);
+        if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
+      }
       slot = (slot + 1) & (uint64_t)cap_mask;
     }
     join_push_head((unsigned int*)&slots[2 * slot + 1], (unsigned int)i, next);
