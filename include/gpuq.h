@@ -193,6 +193,37 @@ int gpuq_join_probe_i64(void* stream, int64_t probe_rows, gpuq_col probe_key,
                         uint32_t* out_probe_rid, uint32_t* out_build_rid,
                         int64_t out_cap, int64_t* out_nmatches);
 
+/* ---------------------------------------------------------------- */
+/* FILTER / PROJECT — SURVEY §8(f).2: FilterExec / ProjectExec on    */
+/* columnar batches, so multi-operator plans stay on-device.         */
+/* Filter: single comparison col OP literal; WHERE keeps only TRUE   */
+/* (NULL comparison result drops the row). Stable compaction: the    */
+/* first *out_count entries of out_perm are the passing rows in      */
+/* input order. Project: elementwise a OP b (column or literal).     */
+/* ---------------------------------------------------------------- */
+
+#define GPUQ_CMP_EQ 0
+#define GPUQ_CMP_LT 1
+#define GPUQ_CMP_LE 2
+#define GPUQ_CMP_GT 3
+#define GPUQ_CMP_GE 4
+#define GPUQ_CMP_NE 5
+
+int64_t gpuq_filter_workspace_bytes(int64_t nrows);
+int gpuq_filter_cmp(void* stream, int64_t nrows, gpuq_col col, int32_t op,
+                    double lit_f64, int64_t lit_i64,
+                    uint32_t* out_perm, int64_t* out_count /* device, u64 */,
+                    void* workspace, int64_t workspace_bytes);
+
+#define GPUQ_BINOP_ADD 0
+#define GPUQ_BINOP_SUB 1
+#define GPUQ_BINOP_MUL 2
+#define GPUQ_BINOP_DIV 3
+
+int gpuq_project_binop(void* stream, int64_t nrows, gpuq_col a,
+                       const void* b /* second column's data or NULL for literal */,
+                       double lit_f64, int64_t lit_i64, int32_t op, void* out);
+
 #ifdef __cplusplus
 }
 #endif

@@ -1574,3 +1574,154 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
          (long long)hsp.cursor, (long long)out_cap);
   return GPUQ_OK;
 }
+
+/* ================= filter / project (SURVEY (f).2) ================= */
+/*
+ * FILTER — replaces FilterExec for single-comparison predicates
+ * (col OP literal): stable compaction via the same ranked-scatter machinery
+ * as the sort's validity split (pass=group 0, fail=group 1; WHERE keeps
+ * only TRUE — a NULL comparison result drops the row, SQL 3VL).
+ * PROJECT — replaces ProjectExec for elementwise binary arithmetic
+ * (col OP col / col OP literal) on int64/float64.
+ */
+
+DEV bool filter_cmp_f64(double v, int op, double lit) {
+  switch (op) {
+    case 0: return v == lit;
+    case 1: return v < lit;
+    case 2: return v <= lit;
+    case 3: return v > lit;
+    case 4: return v >= lit;
+    default: return v != lit;
+  }
+}
+DEV bool filter_cmp_i64(int64_t v, int op, int64_t lit) {
+  switch (op) {
+    case 0: return v == lit;
+    case 1: return v < lit;
+    case 2: return v <= lit;
+    case 3: return v > lit;
+    case 4: return v >= lit;
+    default: return v != lit;
+  }
+}
+
+template <int DTYPE>
+__global__ void k_filter_pred(int64_t n, const void* data, const uint8_t* validity,
+                              int op, double lit_f, int64_t lit_i,
+                              uint64_t* pid_as_key, uint32_t* idx,
+                              unsigned long long* pass_count) {
+  __shared__ unsigned int h;
+  if (threadIdx.x == 0) h = 0;
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  unsigned int mine = 0;
+  for (; i < n; i += stride) {
+    bool pass = bit_valid(validity, i);
+    if (pass) {
+      if (DTYPE == GPUQ_FLOAT64) pass = filter_cmp_f64(((const double*)data)[i], op, lit_f);
+      else pass = filter_cmp_i64(((const int64_t*)data)[i], op, lit_i);
+    }
+    pid_as_key[i] = pass ? 0ull : 1ull;
+    idx[i] = (uint32_t)i;
+    if (pass) mine++;
+  }
+  if (mine) atomicAdd(&h, mine);
+  __syncthreads();
+  if (threadIdx.x == 0 && h) atomicAdd(pass_count, (unsigned long long)h);
+}
+
+extern "C" int64_t gpuq_filter_workspace_bytes(int64_t n) {
+  return gpuq_sort_workspace_bytes(n);
+}
+
+extern "C" int gpuq_filter_cmp(void* stream, int64_t n, gpuq_col col, int32_t op,
+                               double lit_f, int64_t lit_i,
+                               uint32_t* out_perm, int64_t* out_count,
+                               void* workspace, int64_t workspace_bytes) {
+  hipStream_t s = (hipStream_t)stream;
+  if (n > 0xFFFFFFFFLL) FAIL(GPUQ_ERR_INVALID, "filter: nrows %lld > 2^32", (long long)n);
+  if (op < 0 || op > 5) FAIL(GPUQ_ERR_INVALID, "filter: bad op %d", op);
+  if (col.dtype != GPUQ_INT64 && col.dtype != GPUQ_FLOAT64)
+    FAIL(GPUQ_ERR_INVALID, "filter: unsupported dtype %d", col.dtype);
+  sort_ws w; int64_t need;
+  sort_ws_layout(n, 256, &w, (char*)workspace, &need);
+  if (workspace_bytes < need)
+    FAIL(GPUQ_ERR_INVALID, "filter: workspace %lld < %lld", (long long)workspace_bytes, (long long)need);
+  HIP_TRY(hipMemsetAsync(out_count, 0, 8, s));
+  if (n == 0) return GPUQ_OK;
+  scatter_geom geom = get_sort_geom();
+  int tile = geom.block * geom.items;
+  int64_t nb = sort_nblocks(n, tile);
+  { hipEvent_t _pe = prof_begin(s);
+  if (col.dtype == GPUQ_FLOAT64)
+    k_filter_pred<GPUQ_FLOAT64><<<grid1d(n), 256, 0, s>>>(
+        n, col.data, col.validity, op, lit_f, lit_i, w.ka, w.ia,
+        (unsigned long long*)out_count);
+  else
+    k_filter_pred<GPUQ_INT64><<<grid1d(n), 256, 0, s>>>(
+        n, col.data, col.validity, op, lit_f, lit_i, w.ka, w.ia,
+        (unsigned long long*)out_count);
+  prof_end("filter_pred", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, w.ka, 0, w.hist, (int)nb, tile);
+  HIP_TRY(hipGetLastError());
+  int rc = exclusive_scan_u32(s, 256 * nb, w.hist, w.hist_scan, w.block_sums);
+  if (rc) return rc;
+  { hipEvent_t _pe = prof_begin(s);
+  launch_scatter<0, false>(s, geom, nb, n, w.ka, w.ia, w.kb, w.ib, w.hist_scan,
+                           0, 0, nullptr, nullptr, nullptr, 0);
+  prof_end("filter_scatter", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipMemcpyAsync(out_perm, w.ib, n * 4, hipMemcpyDeviceToDevice, s));
+  return GPUQ_OK;
+}
+
+/* out = a OP b (b = column or broadcast literal) */
+template <int DTYPE, int OP, bool B_IS_LIT>
+__global__ void k_project_binop(int64_t n, const void* a, const void* b,
+                                double lit_f, int64_t lit_i, void* out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    if (DTYPE == GPUQ_FLOAT64) {
+      double x = ((const double*)a)[i];
+      double y = B_IS_LIT ? lit_f : ((const double*)b)[i];
+      double r = OP == 0 ? x + y : OP == 1 ? x - y : OP == 2 ? x * y : x / y;
+      ((double*)out)[i] = r;
+    } else {
+      int64_t x = ((const int64_t*)a)[i];
+      int64_t y = B_IS_LIT ? lit_i : ((const int64_t*)b)[i];
+      int64_t r = OP == 0 ? x + y : OP == 1 ? x - y : OP == 2 ? x * y
+                  : (y == 0 ? 0 : x / y);
+      ((int64_t*)out)[i] = r;
+    }
+  }
+}
+
+extern "C" int gpuq_project_binop(void* stream, int64_t n, gpuq_col a,
+                                  const void* b /* col data or NULL */,
+                                  double lit_f, int64_t lit_i, int32_t op,
+                                  void* out) {
+  hipStream_t s = (hipStream_t)stream;
+  if (op < 0 || op > 3) FAIL(GPUQ_ERR_INVALID, "project: bad op %d", op);
+  if (a.validity) FAIL(GPUQ_ERR_INVALID, "project: validity not yet supported");
+  dim3 g = grid1d(n);
+#define PJ(DT, OPV) do { \
+    if (b) k_project_binop<DT, OPV, false><<<g, 256, 0, s>>>(n, a.data, b, lit_f, lit_i, out); \
+    else   k_project_binop<DT, OPV, true><<<g, 256, 0, s>>>(n, a.data, b, lit_f, lit_i, out); \
+  } while (0)
+  if (a.dtype == GPUQ_FLOAT64) {
+    if (op == 0) PJ(GPUQ_FLOAT64, 0); else if (op == 1) PJ(GPUQ_FLOAT64, 1);
+    else if (op == 2) PJ(GPUQ_FLOAT64, 2); else PJ(GPUQ_FLOAT64, 3);
+  } else if (a.dtype == GPUQ_INT64) {
+    if (op == 0) PJ(GPUQ_INT64, 0); else if (op == 1) PJ(GPUQ_INT64, 1);
+    else if (op == 2) PJ(GPUQ_INT64, 2); else PJ(GPUQ_INT64, 3);
+  } else {
+    FAIL(GPUQ_ERR_INVALID, "project: unsupported dtype %d", a.dtype);
+  }
+#undef PJ
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}

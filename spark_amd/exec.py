@@ -147,6 +147,29 @@ class ShuffleExchangeExec(_CpuNode):
         return self.children[0].output
 
 
+class FilterExec(_CpuNode):
+    def __init__(self, col: str, op: str, literal, child):
+        super().__init__(child)
+        self.col, self.op, self.literal = col, op, literal
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+
+class ProjectExec(_CpuNode):
+    """projections: list of (out_name, a, op, b_or_None, literal_or_None) plus
+    pass-through column names."""
+
+    def __init__(self, projections, child):
+        super().__init__(child)
+        self.projections = projections
+
+    @property
+    def output(self):
+        return [p if isinstance(p, str) else p[0] for p in self.projections]
+
+
 class InputBatches(SparkPlan):
     """Leaf: pre-materialized device batches (scan stand-in)."""
 
@@ -383,6 +406,68 @@ class GpuShuffledHashJoinExec(SparkPlan):
         yield ColumnarBatch(cols)
 
 
+class GpuFilterExec(SparkPlan):
+    """Replaces FilterExec for col OP literal predicates:
+    stable compaction on device, then payload gather by the passing-row
+    permutation."""
+
+    def __init__(self, col: str, op: str, literal, child):
+        super().__init__(child)
+        self.col, self.op, self.literal = col, op, literal
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def execute_columnar(self):
+        from . import gpuq
+        for batch in self.children[0].execute_columnar():
+            perm, cnt = gpuq.filter_cmp(batch.column(self.col), self.op,
+                                        self.literal,
+                                        validity=batch.validity(self.col))
+            cols = {name: gpuq.gather(t, perm)
+                    for name, t in batch.columns().items()}
+            batch.close()
+            yield ColumnarBatch(cols)
+
+
+class GpuProjectExec(SparkPlan):
+    """Replaces ProjectExec for elementwise binary
+    arithmetic; pass-through entries keep their column."""
+
+    def __init__(self, projections, child):
+        super().__init__(child)
+        self.projections = projections
+
+    @property
+    def output(self):
+        return [p if isinstance(p, str) else p[0] for p in self.projections]
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def execute_columnar(self):
+        from . import gpuq
+        for batch in self.children[0].execute_columnar():
+            cols = {}
+            for p in self.projections:
+                if isinstance(p, str):
+                    cols[p] = batch.column(p)
+                else:
+                    out_name, a, op, b, lit = p
+                    cols[out_name] = gpuq.project_binop(
+                        batch.column(a), op,
+                        b=batch.column(b) if b is not None else None,
+                        literal=lit)
+            batch.close()
+            yield ColumnarBatch(cols)
+
+
 class GpuColumnarRule:
     """The injected rule (ColumnarRule, Columnar.scala:36-50; injection via
     SparkSessionExtensions.injectColumnar:168; applied at
@@ -402,6 +487,10 @@ class GpuColumnarRule:
                                            plan.build_side, *children)
         if isinstance(plan, ShuffleExchangeExec):
             return GpuShuffleExchangeExec(plan.keys, *children)
+        if isinstance(plan, FilterExec):
+            return GpuFilterExec(plan.col, plan.op, plan.literal, *children)
+        if isinstance(plan, ProjectExec):
+            return GpuProjectExec(plan.projections, *children)
         plan.children = children
         return plan
 

@@ -79,6 +79,14 @@ def lib() -> ctypes.CDLL:
         L.gpuq_join_probe_i64.restype = i32
         L.gpuq_join_probe_i64.argtypes = [vp, i64, _Col, vp, i64, i64, vp, i64,
                                           vp, vp, i64, ctypes.POINTER(i64)]
+        L.gpuq_filter_workspace_bytes.restype = i64
+        L.gpuq_filter_workspace_bytes.argtypes = [i64]
+        L.gpuq_filter_cmp.restype = i32
+        L.gpuq_filter_cmp.argtypes = [vp, i64, _Col, i32, ctypes.c_double, i64,
+                                      vp, vp, vp, i64]
+        L.gpuq_project_binop.restype = i32
+        L.gpuq_project_binop.argtypes = [vp, i64, _Col, vp, ctypes.c_double, i64,
+                                         i32, vp]
         L.gpuq_profiling.restype = None
         L.gpuq_profiling.argtypes = [i32]
         L.gpuq_kernel_stats_reset.restype = None
@@ -253,3 +261,37 @@ def join_probe(probe_keys: torch.Tensor, workspace: torch.Tensor, capacity: int,
         return None, None, nm.value
     _check(rc)
     return op[:nm.value], ob[:nm.value], nm.value
+
+
+CMP = {"==": 0, "<": 1, "<=": 2, ">": 3, ">=": 4, "!=": 5}
+BINOP = {"+": 0, "-": 1, "*": 2, "/": 3}
+
+
+def filter_cmp(col: torch.Tensor, op: str, literal, workspace=None, validity=None):
+    """Stable filter: returns (perm[:count], count) — passing rows in input
+    order (FilterExec replacement for col OP literal predicates)."""
+    n = col.numel()
+    dev = col.device
+    if workspace is None:
+        workspace = torch.empty(lib().gpuq_filter_workspace_bytes(n),
+                                dtype=torch.uint8, device=dev)
+    perm = torch.empty(n, dtype=torch.int32, device=dev)
+    cnt = torch.zeros(1, dtype=torch.int64, device=dev)
+    _check(lib().gpuq_filter_cmp(_stream(), n, _col(col, validity), CMP[op],
+                                 float(literal), int(literal),
+                                 perm.data_ptr(), cnt.data_ptr(),
+                                 workspace.data_ptr(), workspace.numel()))
+    c = int(cnt.item())
+    return perm[:c], c
+
+
+def project_binop(a: torch.Tensor, op: str, b=None, literal=None):
+    """Elementwise a OP b (b: tensor or None with literal)."""
+    n = a.numel()
+    out = torch.empty(n, dtype=a.dtype, device=a.device)
+    lit_f = float(literal) if literal is not None else 0.0
+    lit_i = int(literal) if literal is not None else 0
+    _check(lib().gpuq_project_binop(_stream(), n, _col(a),
+                                    b.data_ptr() if b is not None else None,
+                                    lit_f, lit_i, BINOP[op], out.data_ptr()))
+    return out

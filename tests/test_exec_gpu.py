@@ -88,3 +88,37 @@ def test_full_pipeline_join_then_agg_then_sort():
     perm = oracle.sort_perm(ok)
     assert (got["k"] == ok[perm]).all()
     np.testing.assert_allclose(got["sum(v)"], osum[perm], rtol=1e-6)
+
+
+def test_filter_project_pipeline():
+    """filter -> project -> agg: the SURVEY §8(f).2 shape (keeps a Q1-style
+    plan on-device between GPU exec nodes)."""
+    n = 200_000
+    keys = oracle.gen_i64(seed=20, n=n, range_=100)
+    qty = oracle.gen_f64_unit(seed=21, n=n)
+    price = oracle.gen_f64_unit(seed=22, n=n)
+    scan = gx.InputBatches([dev_batch(k=keys, qty=qty, price=price)])
+    plan = gx.HashAggregateExec(
+        "k", [("sum", "revenue"), ("count", "revenue")], "complete",
+        gx.ProjectExec(["k", ("revenue", "qty", "*", "price", None)],
+                       gx.FilterExec("qty", "<", 0.5, scan)))
+    got = run_plan(plan)
+    mask = qty < 0.5
+    rev = qty[mask] * price[mask]
+    ok, _, osum, _, ocnt = oracle.hash_agg(keys[mask], rev)
+    g, o = np.argsort(got["k"]), np.argsort(ok)
+    assert (got["k"][g] == ok[o]).all()
+    assert (got["count(revenue)"][g] == ocnt[o]).all()
+    np.testing.assert_allclose(got["sum(revenue)"][g], osum[o], rtol=1e-6)
+
+
+def test_filter_stability_and_literal_ops():
+    n = 50_000
+    vals = oracle.gen_i64(seed=30, n=n, range_=1000)
+    from spark_amd import gpuq
+    for op, fn in [("<", np.less), (">=", np.greater_equal), ("==", np.equal),
+                   ("!=", np.not_equal)]:
+        perm, cnt = gpuq.filter_cmp(torch.from_numpy(vals).cuda(), op, 500)
+        exp = np.flatnonzero(fn(vals, 500))
+        assert cnt == len(exp)
+        assert (perm.cpu().numpy().astype(np.uint32) == exp.astype(np.uint32)).all()
