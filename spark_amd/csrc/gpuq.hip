@@ -1725,3 +1725,18 @@ extern "C" int gpuq_project_binop(void* stream, int64_t n, gpuq_col a,
   HIP_TRY(hipGetLastError());
   return GPUQ_OK;
 }
+
+/* int64 -> float64 cast (AVG = SUM/COUNT evaluation; Average.scala
+ * evaluateExpression divides sum by count cast to double) */
+__global__ void k_cast_i64_f64(int64_t n, const int64_t* in, double* out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = (double)in[i];
+}
+
+extern "C" int gpuq_cast_i64_f64(void* stream, int64_t n, const int64_t* in,
+                                 double* out) {
+  k_cast_i64_f64<<<grid1d(n), 256, 0, (hipStream_t)stream>>>(n, in, out);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
