@@ -224,6 +224,9 @@ int gpuq_project_binop(void* stream, int64_t nrows, gpuq_col a,
                        const void* b /* second column's data or NULL for literal */,
                        double lit_f64, int64_t lit_i64, int32_t op, void* out);
 
+/* int64 -> float64 cast (AVG evaluation: sum / cast(count)) */
+int gpuq_cast_i64_f64(void* stream, int64_t nrows, const int64_t* in, double* out);
+
 #ifdef __cplusplus
 }
 #endif

@@ -256,12 +256,12 @@ class GpuHashAggregateExec(SparkPlan):
     def execute_columnar(self):
         from . import gpuq
         fns = {fn for fn, _ in self.aggs}
-        assert fns <= {"sum", "count"}, f"unsupported aggs {fns}"
-        val_col = next(col for fn, col in self.aggs if fn in ("sum", "count"))
+        assert fns <= {"sum", "count", "avg"}, f"unsupported aggs {fns}"
+        val_col = next(col for fn, col in self.aggs)
         ops = 0
-        if any(fn == "sum" for fn, _ in self.aggs) or self.mode == "final":
+        if fns & {"sum", "avg"} or self.mode == "final":
             ops |= gpuq.AGG_SUM
-        if any(fn == "count" for fn, _ in self.aggs):
+        if fns & {"count", "avg"}:
             ops |= gpuq.AGG_COUNT
         for batch in self.children[0].execute_columnar():
             keys = batch.column(self.group_key)
@@ -281,7 +281,13 @@ class GpuHashAggregateExec(SparkPlan):
             ok, okv, osum, osv, ocnt = out
             cols = {self.group_key: ok}
             for fn, col in self.aggs:
-                cols[f"{fn}({col})"] = osum if fn == "sum" else ocnt
+                if fn == "avg":
+                    # Average.evaluateExpression: sum / cast(count)
+                    # (catalyst/.../expressions/aggregate/Average.scala)
+                    cols[f"{fn}({col})"] = gpuq.project_binop(
+                        osum, "/", b=gpuq.cast_i64_f64(ocnt))
+                else:
+                    cols[f"{fn}({col})"] = osum if fn == "sum" else ocnt
             batch.close()
             yield ColumnarBatch(cols, validity={self.group_key: None})
 

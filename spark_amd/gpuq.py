@@ -87,6 +87,8 @@ def lib() -> ctypes.CDLL:
         L.gpuq_project_binop.restype = i32
         L.gpuq_project_binop.argtypes = [vp, i64, _Col, vp, ctypes.c_double, i64,
                                          i32, vp]
+        L.gpuq_cast_i64_f64.restype = i32
+        L.gpuq_cast_i64_f64.argtypes = [vp, i64, vp, vp]
         L.gpuq_profiling.restype = None
         L.gpuq_profiling.argtypes = [i32]
         L.gpuq_kernel_stats_reset.restype = None
@@ -294,4 +296,10 @@ def project_binop(a: torch.Tensor, op: str, b=None, literal=None):
     _check(lib().gpuq_project_binop(_stream(), n, _col(a),
                                     b.data_ptr() if b is not None else None,
                                     lit_f, lit_i, BINOP[op], out.data_ptr()))
+    return out
+
+
+def cast_i64_f64(t: torch.Tensor) -> torch.Tensor:
+    out = torch.empty(t.numel(), dtype=torch.float64, device=t.device)
+    _check(lib().gpuq_cast_i64_f64(_stream(), t.numel(), t.data_ptr(), out.data_ptr()))
     return out

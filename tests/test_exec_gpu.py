@@ -122,3 +122,14 @@ def test_filter_stability_and_literal_ops():
         exp = np.flatnonzero(fn(vals, 500))
         assert cnt == len(exp)
         assert (perm.cpu().numpy().astype(np.uint32) == exp.astype(np.uint32)).all()
+
+
+def test_avg_aggregate():
+    n = 100_000
+    keys = oracle.gen_i64(seed=40, n=n, range_=333)
+    vals = oracle.gen_f64_unit(seed=41, n=n)
+    scan = gx.InputBatches([dev_batch(k=keys, v=vals)])
+    got = run_plan(gx.HashAggregateExec("k", [("avg", "v")], "complete", scan))
+    ok, _, osum, _, ocnt = oracle.hash_agg(keys, vals)
+    g, o = np.argsort(got["k"]), np.argsort(ok)
+    np.testing.assert_allclose(got["avg(v)"][g], osum[o] / ocnt[o], rtol=1e-6)
