@@ -137,6 +137,22 @@ class ShuffledHashJoinExec(_CpuNode):
         return self.children[0].output + self.children[1].output
 
 
+class SortMergeJoinExec(_CpuNode):
+    """CPU placeholder (joins/SortMergeJoinExec.scala:135). The GPU plan
+    replaces SMJ with the hash join — both fill the same equi-join plan slot
+    with identical ClusteredDistribution requirements (ShuffledJoin.scala:
+    57-69), and on GPU the hash build/probe dominates sort+merge for
+    co-partitioned batches."""
+
+    def __init__(self, left_key: str, right_key: str, left, right):
+        super().__init__(left, right)
+        self.left_key, self.right_key = left_key, right_key
+
+    @property
+    def output(self):
+        return self.children[0].output + self.children[1].output
+
+
 class ShuffleExchangeExec(_CpuNode):
     def __init__(self, keys: Tuple[str, ...], num_partitions: int, child):
         super().__init__(child)
@@ -491,6 +507,11 @@ class GpuColumnarRule:
         if isinstance(plan, ShuffledHashJoinExec):
             return GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
                                            plan.build_side, *children)
+        if isinstance(plan, SortMergeJoinExec):
+            # SMJ -> GPU hash join (same slot, same required distribution;
+            # build on the right side as SHJ's default would choose)
+            return GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
+                                           "right", *children)
         if isinstance(plan, ShuffleExchangeExec):
             return GpuShuffleExchangeExec(plan.keys, *children)
         if isinstance(plan, FilterExec):

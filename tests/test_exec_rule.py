@@ -59,3 +59,15 @@ def test_batch_lifetime():
     b.close()
     with pytest.raises(AssertionError):
         b.column("k")
+
+
+def test_sort_merge_join_replaced_by_gpu_hash_join():
+    # SMJ and SHJ fill the same equi-join plan slot; the
+    # rule maps both to the GPU hash join
+    scan = gx.InputBatches.__new__(gx.InputBatches)
+    gx.SparkPlan.__init__(scan)
+    scan._batches = []
+    smj = gx.SortMergeJoinExec("a", "b", scan, scan)
+    plan = gx.GpuColumnarRule().pre_columnar_transitions(smj)
+    assert isinstance(plan, gx.GpuShuffledHashJoinExec)
+    assert plan.left_key == "a" and plan.right_key == "b"

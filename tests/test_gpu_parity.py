@@ -245,3 +245,38 @@ def test_sort_all_nulls(gq):
     perm, skeys = gq.sort_perm(to_dev(keys), key_validity=pack_validity(valid))
     assert (perm.cpu().numpy() == np.arange(n)).all()
     assert (skeys.cpu().numpy() == keys).all()
+
+
+def test_agg_multi_batch_accumulation(gq):
+    """first_batch/finalize contract: accumulate two batches into one table
+    (the Partial-mode shape: TungstenAggregationIterator consumes a whole
+    partition before emitting)."""
+    import ctypes
+    n = 100_000
+    keys1 = oracle.gen_i64(seed=81, n=n, range_=500)
+    vals1 = oracle.gen_f64_unit(seed=82, n=n)
+    keys2 = oracle.gen_i64(seed=83, n=n, range_=500)
+    vals2 = oracle.gen_f64_unit(seed=84, n=n)
+    cap = 2048
+    ws = gq.agg_workspace(cap)
+    outs = [torch.empty(600, dtype=d, device="cuda")
+            for d in (torch.int64, torch.uint8, torch.float64, torch.uint8,
+                      torch.int64)]
+    ng = ctypes.c_int64(0)
+    gq._check(gq.lib().gpuq_hash_agg_i64_f64(
+        gq._stream(), n, gq._col(to_dev(keys1)), gq._col(to_dev(vals1)),
+        ws.data_ptr(), cap, 1, 0, 3, *[t.data_ptr() for t in outs],
+        ctypes.byref(ng)))
+    gq._check(gq.lib().gpuq_hash_agg_i64_f64(
+        gq._stream(), n, gq._col(to_dev(keys2)), gq._col(to_dev(vals2)),
+        ws.data_ptr(), cap, 0, 1, 3, *[t.data_ptr() for t in outs],
+        ctypes.byref(ng)))
+    g = ng.value
+    gk = outs[0][:g].cpu().numpy()
+    gs = outs[2][:g].cpu().numpy()
+    gc = outs[4][:g].cpu().numpy()
+    ok, _, osum, _, ocnt = oracle.hash_agg(
+        np.concatenate([keys1, keys2]), np.concatenate([vals1, vals2]))
+    go, oo = np.argsort(gk), np.argsort(ok)
+    assert (gk[go] == ok[oo]).all() and (gc[go] == ocnt[oo]).all()
+    np.testing.assert_allclose(gs[go], osum[oo], rtol=1e-6)
