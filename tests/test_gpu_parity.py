@@ -263,12 +263,14 @@ def test_agg_multi_batch_accumulation(gq):
             for d in (torch.int64, torch.uint8, torch.float64, torch.uint8,
                       torch.int64)]
     ng = ctypes.c_int64(0)
+    # keep device tensors alive through the async kernel launches
+    dk1, dv1, dk2, dv2 = to_dev(keys1), to_dev(vals1), to_dev(keys2), to_dev(vals2)
     gq._check(gq.lib().gpuq_hash_agg_i64_f64(
-        gq._stream(), n, gq._col(to_dev(keys1)), gq._col(to_dev(vals1)),
+        gq._stream(), n, gq._col(dk1), gq._col(dv1),
         ws.data_ptr(), cap, 1, 0, 3, *[t.data_ptr() for t in outs],
         ctypes.byref(ng)))
     gq._check(gq.lib().gpuq_hash_agg_i64_f64(
-        gq._stream(), n, gq._col(to_dev(keys2)), gq._col(to_dev(vals2)),
+        gq._stream(), n, gq._col(dk2), gq._col(dv2),
         ws.data_ptr(), cap, 0, 1, 3, *[t.data_ptr() for t in outs],
         ctypes.byref(ng)))
     g = ng.value
