@@ -1006,10 +1006,20 @@ struct agg_special {
   unsigned long long overflow;
 };
 
-/* Interleaved 24-byte slots [key, sum-bits, count] so one probe touches one
- * cache line instead of three parallel arrays. */
+/* Interleaved slots [key, sum-bits, count(, pad)] so one probe touches one
+ * cache line instead of three parallel arrays. Stride (u64 words/slot) is
+ * 3 (24 B, default) or 4 (32 B, never line-straddling) via GPUQ_AGG_STRIDE. */
+static int agg_stride(void) {
+  static int s = 0;
+  if (!s) {
+    const char* e = getenv("GPUQ_AGG_STRIDE");
+    s = (e && atoi(e) == 4) ? 4 : 3;
+  }
+  return s;
+}
+
 struct agg_ws {
-  unsigned long long* tab;   /* 3 * cap u64 */
+  unsigned long long* tab;   /* stride * cap u64 */
   agg_special* sp;
 };
 
@@ -1020,7 +1030,7 @@ static void agg_ws_layout(int64_t cap, agg_ws* w, char* base, int64_t* total) {
     off += (bytes + 255) & ~255LL;
     return p;
   };
-  w->tab = (unsigned long long*)take(cap * 24);
+  w->tab = (unsigned long long*)take(cap * 8 * 4);  /* sized for stride 4 */
   w->sp = (agg_special*)take(sizeof(agg_special));
   *total = off;
 }
@@ -1031,20 +1041,21 @@ extern "C" int64_t gpuq_hash_agg_workspace_bytes(int64_t cap) {
   return total;
 }
 
+template <int SLOT>
 __global__ void k_agg_init(int64_t cap, unsigned long long* tab) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < cap; i += stride) {
-    tab[3 * i] = AGG_EMPTY;
-    tab[3 * i + 1] = 0;
-    tab[3 * i + 2] = 0;
+    tab[SLOT * i] = AGG_EMPTY;
+    tab[SLOT * i + 1] = 0;
+    tab[SLOT * i + 2] = 0;
   }
 }
 
 #define AGG_OP_SUM 1
 #define AGG_OP_COUNT 2
 
-template <int OPS>
+template <int OPS, int SLOT>
 __global__ void k_agg_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                             const double* vals, const uint8_t* vvalid,
                             unsigned long long* tab, agg_special* sp,
@@ -1069,26 +1080,26 @@ __global__ void k_agg_build(int64_t n, const int64_t* keys, const uint8_t* kvali
     int64_t k = keys[i];
     uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
     for (int probes = 0;; probes++) {
-      unsigned long long cur = __hip_atomic_load(&tab[3 * slot], __ATOMIC_RELAXED,
+      unsigned long long cur = __hip_atomic_load(&tab[SLOT * slot], __ATOMIC_RELAXED,
                                                  __HIP_MEMORY_SCOPE_AGENT);
       if (cur == (unsigned long long)k) break;
       if (cur == AGG_EMPTY) {
-        unsigned long long prev = atomicCAS(&tab[3 * slot], AGG_EMPTY, (unsigned long long)k);
+        unsigned long long prev = atomicCAS(&tab[SLOT * slot], AGG_EMPTY, (unsigned long long)k);
         if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
       }
       slot = (slot + 1) & (uint64_t)cap_mask;
       if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
     }
     if (vv) {
-      if (OPS & AGG_OP_SUM) atomicAdd((double*)&tab[3 * slot + 1], v);
-      if (OPS & AGG_OP_COUNT) atomicAdd(&tab[3 * slot + 2], 1ull);
+      if (OPS & AGG_OP_SUM) atomicAdd((double*)&tab[SLOT * slot + 1], v);
+      if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], 1ull);
     }
   }
 }
 
 #define AGGC_CHUNK 8192
 
-template <int OPS>
+template <int OPS, int SLOT>
 __global__ void k_agg_compact(int64_t cap, const unsigned long long* tab,
                               agg_special* sp,
                               int64_t* out_keys, uint8_t* out_kvalid,
@@ -1108,7 +1119,7 @@ __global__ void k_agg_compact(int64_t cap, const unsigned long long* tab,
   uint32_t lane_total = 0;
   for (int r = 0; r < ROUNDS; r++) {
     int64_t i = base + r * 256 + threadIdx.x;
-    bool occ = i < cap && tab[3 * i] != AGG_EMPTY;
+    bool occ = i < cap && tab[SLOT * i] != AGG_EMPTY;
     if (occ) { occ_mask[r / 32] |= 1u << (r & 31); lane_total++; }
   }
   uint32_t incl = wave_inclusive_scan(lane_total);
@@ -1124,12 +1135,12 @@ __global__ void k_agg_compact(int64_t cap, const unsigned long long* tab,
   for (int r = 0; r < ROUNDS; r++) {
     if (!((occ_mask[r / 32] >> (r & 31)) & 1)) continue;
     int64_t i = base + r * 256 + threadIdx.x;
-    unsigned long long k = tab[3 * i];
+    unsigned long long k = tab[SLOT * i];
     out_keys[o] = (int64_t)k;
     out_kvalid[o] = 1;
-    out_sums[o] = __longlong_as_double((long long)tab[3 * i + 1]);
-    out_svalid[o] = (OPS & AGG_OP_COUNT) ? (tab[3 * i + 2] > 0 ? 1 : 0) : 1;
-    if (out_cnts) out_cnts[o] = (int64_t)tab[3 * i + 2];
+    out_sums[o] = __longlong_as_double((long long)tab[SLOT * i + 1]);
+    out_svalid[o] = (OPS & AGG_OP_COUNT) ? (tab[SLOT * i + 2] > 0 ? 1 : 0) : 1;
+    if (out_cnts) out_cnts[o] = (int64_t)tab[SLOT * i + 2];
     o++;
   }
   if (blockIdx.x == 0 && threadIdx.x == 0) {
@@ -1168,20 +1179,19 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
   agg_ws w; int64_t need;
   agg_ws_layout(cap, &w, (char*)workspace, &need);
   if (first_batch) {
-    k_agg_init<<<grid1d(cap), 256, 0, s>>>(cap, w.tab);
+    if (agg_stride() == 4) k_agg_init<4><<<grid1d(cap), 256, 0, s>>>(cap, w.tab);
+    else k_agg_init<3><<<grid1d(cap), 256, 0, s>>>(cap, w.tab);
     HIP_TRY(hipGetLastError());
     HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_special), s));
   }
   if (n > 0) {
     { hipEvent_t _pe = prof_begin(s);
-    if (ops == AGG_OP_SUM)
-      k_agg_build<AGG_OP_SUM><<<hash_grid(n), 256, 0, s>>>(
-          n, (const int64_t*)key.data, key.validity,
-          (const double*)val.data, val.validity, w.tab, w.sp, cap - 1);
-    else
-      k_agg_build<AGG_OP_SUM | AGG_OP_COUNT><<<hash_grid(n), 256, 0, s>>>(
-          n, (const int64_t*)key.data, key.validity,
-          (const double*)val.data, val.validity, w.tab, w.sp, cap - 1);
+#define AGB(OPS, SL) k_agg_build<OPS, SL><<<hash_grid(n), 256, 0, s>>>( \
+        n, (const int64_t*)key.data, key.validity, \
+        (const double*)val.data, val.validity, w.tab, w.sp, cap - 1)
+    if (agg_stride() == 4) { if (ops == AGG_OP_SUM) AGB(AGG_OP_SUM, 4); else AGB(3, 4); }
+    else { if (ops == AGG_OP_SUM) AGB(AGG_OP_SUM, 3); else AGB(3, 3); }
+#undef AGB
     prof_end("agg_build", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
@@ -1192,14 +1202,12 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
     if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "agg: hash table overflow (capacity %lld)", (long long)cap);
     { hipEvent_t _pe = prof_begin(s);
     dim3 cgrid((uint32_t)((cap + AGGC_CHUNK - 1) / AGGC_CHUNK));
-    if (ops == AGG_OP_SUM)
-      k_agg_compact<AGG_OP_SUM><<<cgrid, 256, 0, s>>>(
-          cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid,
-          out_counts);
-    else
-      k_agg_compact<AGG_OP_SUM | AGG_OP_COUNT><<<cgrid, 256, 0, s>>>(
-          cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid,
-          out_counts);
+#define AGC(OPS, SL) k_agg_compact<OPS, SL><<<cgrid, 256, 0, s>>>( \
+        cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid, \
+        out_counts)
+    if (agg_stride() == 4) { if (ops == AGG_OP_SUM) AGC(AGG_OP_SUM, 4); else AGC(3, 4); }
+    else { if (ops == AGG_OP_SUM) AGC(AGG_OP_SUM, 3); else AGC(3, 3); }
+#undef AGC
     prof_end("agg_compact", s, _pe); }
     HIP_TRY(hipGetLastError());
     agg_special hsp2;
