@@ -74,6 +74,11 @@ int gpuq_gen_i64_range(void* stream, uint64_t seed, uint64_t start,
 int gpuq_gen_f64_unit(void* stream, uint64_t seed, uint64_t start,
                       int64_t nrows, double* out);
 
+/* RangeExec scan feed (basicPhysicalOperators.scala:630):
+ * out[i] = start + i*step */
+int gpuq_range_i64(void* stream, int64_t nrows, int64_t start, int64_t step,
+                   int64_t* out);
+
 /* ---------------------------------------------------------------- */
 /* SORT — replaces SortExec (execution/SortExec.scala:75-126): the   */
 /* radix-eligible single-key path (canUseRadixSort, SortExec.scala   */

@@ -1161,6 +1161,82 @@ __global__ void k_agg_compact(int64_t cap, const unsigned long long* tab,
   }
 }
 
+/* small-cardinality build: per-block LDS table (classic two-level
+ * aggregation), merged once into the global table per block. Removes the
+ * global-atomic contention wall on few-group aggregates (measured 607 ms
+ * for 1B rows / 1K groups on the direct path). cap <= AGG_LDS_CAP slots. */
+#define AGG_LDS_CAP 2048
+
+template <int OPS, int SLOT>
+__global__ __launch_bounds__(256)
+void k_agg_build_lds(int64_t n, const int64_t* keys, const uint8_t* kvalid,
+                     const double* vals, const uint8_t* vvalid,
+                     unsigned long long* tab, agg_special* sp,
+                     int64_t cap_mask) {
+  __shared__ unsigned long long lt[AGG_LDS_CAP * 3];
+  const int64_t lcap = cap_mask + 1;
+  for (int j = threadIdx.x; j < (int)lcap; j += blockDim.x) {
+    lt[3 * j] = AGG_EMPTY; lt[3 * j + 1] = 0; lt[3 * j + 2] = 0;
+  }
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    bool kv = bit_valid(kvalid, i);
+    bool vv = bit_valid(vvalid, i);
+    double v = vv ? vals[i] : 0.0;
+    if (!kv || (unsigned long long)keys[i] == AGG_EMPTY) {
+      double* psum = kv ? &sp->m1_sum : &sp->nul_sum;
+      unsigned long long* pcnt = kv ? &sp->m1_cnt : &sp->nul_cnt;
+      unsigned long long* pseen = kv ? &sp->m1_seen : &sp->nul_seen;
+      atomicMax(pseen, 1ull);
+      if (vv) {
+        if (OPS & AGG_OP_SUM) atomicAdd(psum, v);
+        if (OPS & AGG_OP_COUNT) atomicAdd(pcnt, 1ull);
+      }
+      continue;
+    }
+    int64_t k = keys[i];
+    uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
+    for (int probes = 0;; probes++) {
+      unsigned long long cur = lt[3 * slot];
+      if (cur == (unsigned long long)k) break;
+      if (cur == AGG_EMPTY) {
+        unsigned long long prev = atomicCAS(&lt[3 * slot], AGG_EMPTY,
+                                            (unsigned long long)k);
+        if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
+      }
+      slot = (slot + 1) & (uint64_t)cap_mask;
+      if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
+    }
+    if (vv) {
+      if (OPS & AGG_OP_SUM) atomicAdd((double*)&lt[3 * slot + 1], v);
+      if (OPS & AGG_OP_COUNT) atomicAdd(&lt[3 * slot + 2], 1ull);
+    }
+  }
+  __syncthreads();
+  /* merge this block's LDS table into the global one (same probe scheme) */
+  for (int j = threadIdx.x; j < (int)lcap; j += blockDim.x) {
+    unsigned long long k = lt[3 * j];
+    if (k == AGG_EMPTY) continue;
+    uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)k, 42)) & (uint64_t)cap_mask;
+    for (;;) {
+      unsigned long long cur = __hip_atomic_load(&tab[SLOT * slot], __ATOMIC_RELAXED,
+                                                 __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == k) break;
+      if (cur == AGG_EMPTY) {
+        unsigned long long prev = atomicCAS(&tab[SLOT * slot], AGG_EMPTY, k);
+        if (prev == AGG_EMPTY || prev == k) break;
+      }
+      slot = (slot + 1) & (uint64_t)cap_mask;
+    }
+    if (OPS & AGG_OP_SUM)
+      atomicAdd((double*)&tab[SLOT * slot + 1],
+                __longlong_as_double((long long)lt[3 * j + 1]));
+    if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], lt[3 * j + 2]);
+  }
+}
+
 extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
                                      gpuq_col key, gpuq_col val,
                                      void* workspace, int64_t cap, int32_t first_batch,
@@ -1186,12 +1262,22 @@ extern "C" int gpuq_hash_agg_i64_f64(void* stream, int64_t n,
   }
   if (n > 0) {
     { hipEvent_t _pe = prof_begin(s);
+    bool lds_path = cap <= AGG_LDS_CAP && getenv("GPUQ_NO_LDS_AGG") == nullptr;
 #define AGB(OPS, SL) k_agg_build<OPS, SL><<<hash_grid(n), 256, 0, s>>>( \
         n, (const int64_t*)key.data, key.validity, \
         (const double*)val.data, val.validity, w.tab, w.sp, cap - 1)
-    if (agg_stride() == 4) { if (ops == AGG_OP_SUM) AGB(AGG_OP_SUM, 4); else AGB(3, 4); }
-    else { if (ops == AGG_OP_SUM) AGB(AGG_OP_SUM, 3); else AGB(3, 3); }
+#define AGBL(OPS, SL) k_agg_build_lds<OPS, SL><<<hash_grid(n), 256, 0, s>>>( \
+        n, (const int64_t*)key.data, key.validity, \
+        (const double*)val.data, val.validity, w.tab, w.sp, cap - 1)
+    if (lds_path) {
+      if (agg_stride() == 4) { if (ops == AGG_OP_SUM) AGBL(AGG_OP_SUM, 4); else AGBL(3, 4); }
+      else { if (ops == AGG_OP_SUM) AGBL(AGG_OP_SUM, 3); else AGBL(3, 3); }
+    } else {
+      if (agg_stride() == 4) { if (ops == AGG_OP_SUM) AGB(AGG_OP_SUM, 4); else AGB(3, 4); }
+      else { if (ops == AGG_OP_SUM) AGB(AGG_OP_SUM, 3); else AGB(3, 3); }
+    }
 #undef AGB
+#undef AGBL
     prof_end("agg_build", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
@@ -1745,6 +1831,20 @@ __global__ void k_cast_i64_f64(int64_t n, const int64_t* in, double* out) {
 extern "C" int gpuq_cast_i64_f64(void* stream, int64_t n, const int64_t* in,
                                  double* out) {
   k_cast_i64_f64<<<grid1d(n), 256, 0, (hipStream_t)stream>>>(n, in, out);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+/* RangeExec scan feed (basicPhysicalOperators.scala:630): id = start + i*step */
+__global__ void k_range_i64(int64_t n, int64_t start, int64_t step, int64_t* out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) out[i] = start + i * step;
+}
+
+extern "C" int gpuq_range_i64(void* stream, int64_t n, int64_t start,
+                              int64_t step, int64_t* out) {
+  k_range_i64<<<grid1d(n), 256, 0, (hipStream_t)stream>>>(n, start, step, out);
   HIP_TRY(hipGetLastError());
   return GPUQ_OK;
 }

@@ -186,6 +186,18 @@ class ProjectExec(_CpuNode):
         return [p if isinstance(p, str) else p[0] for p in self.projections]
 
 
+class RangeExec(_CpuNode):
+    """CPU placeholder for RangeExec (basicPhysicalOperators.scala:630)."""
+
+    def __init__(self, n: int, start: int = 0, step: int = 1, name: str = "id"):
+        super().__init__()
+        self.n, self.start, self.step, self.name = n, start, step, name
+
+    @property
+    def output(self):
+        return [self.name]
+
+
 class InputBatches(SparkPlan):
     """Leaf: pre-materialized device batches (scan stand-in)."""
 
@@ -428,6 +440,27 @@ class GpuShuffledHashJoinExec(SparkPlan):
         yield ColumnarBatch(cols)
 
 
+class GpuRangeExec(SparkPlan):
+    """Replaces RangeExec: generates the id column directly in HBM (the
+    scan-feed adjacency, SURVEY §8(f).1 — no RowToColumnar CPU tax)."""
+
+    def __init__(self, n: int, start: int = 0, step: int = 1, name: str = "id"):
+        super().__init__()
+        self.n, self.start, self.step, self.name = n, start, step, name
+
+    @property
+    def output(self):
+        return [self.name]
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def execute_columnar(self):
+        from . import gpuq
+        yield ColumnarBatch({self.name: gpuq.range_i64(self.n, self.start, self.step)})
+
+
 class GpuFilterExec(SparkPlan):
     """Replaces FilterExec for col OP literal predicates:
     stable compaction on device, then payload gather by the passing-row
@@ -518,6 +551,8 @@ class GpuColumnarRule:
             return GpuFilterExec(plan.col, plan.op, plan.literal, *children)
         if isinstance(plan, ProjectExec):
             return GpuProjectExec(plan.projections, *children)
+        if isinstance(plan, RangeExec):
+            return GpuRangeExec(plan.n, plan.start, plan.step, plan.name)
         plan.children = children
         return plan
 

@@ -87,6 +87,8 @@ def lib() -> ctypes.CDLL:
         L.gpuq_project_binop.restype = i32
         L.gpuq_project_binop.argtypes = [vp, i64, _Col, vp, ctypes.c_double, i64,
                                          i32, vp]
+        L.gpuq_range_i64.restype = i32
+        L.gpuq_range_i64.argtypes = [vp, i64, i64, i64, vp]
         L.gpuq_cast_i64_f64.restype = i32
         L.gpuq_cast_i64_f64.argtypes = [vp, i64, vp, vp]
         L.gpuq_profiling.restype = None
@@ -302,4 +304,10 @@ def project_binop(a: torch.Tensor, op: str, b=None, literal=None):
 def cast_i64_f64(t: torch.Tensor) -> torch.Tensor:
     out = torch.empty(t.numel(), dtype=torch.float64, device=t.device)
     _check(lib().gpuq_cast_i64_f64(_stream(), t.numel(), t.data_ptr(), out.data_ptr()))
+    return out
+
+
+def range_i64(n: int, start: int = 0, step: int = 1, device="cuda") -> torch.Tensor:
+    out = torch.empty(n, dtype=torch.int64, device=device)
+    _check(lib().gpuq_range_i64(_stream(), n, start, step, out.data_ptr()))
     return out
