@@ -38,3 +38,20 @@ for bn in [1_000_000, 30_000_000, 500_000_000]:
     print(f"join bn={bn:>11,} table={cap*16/2**20:8.1f}MB probe={ms:7.2f}ms", flush=True)
     del bkeys, ws
     torch.cuda.empty_cache()
+
+# LDS pre-agg A/B at low cardinality (re-run section)
+if os.environ.get("DIAG_LDS_AGG"):
+    n = 1_000_000_000
+    for groups in [4, 1_000, 500_000]:
+        keys = gq.gen_i64(seed=1, n=n, range_=groups)
+        vals = gq.gen_f64_unit(seed=2, n=n)
+        cap = 1 << max(4, (groups * 2 - 1).bit_length())
+        ws = gq.agg_workspace(cap)
+        for _ in range(2):
+            gq.hash_agg(keys, vals, cap, workspace=ws, max_groups=groups + 2, ops=gq.AGG_SUM)
+        gq.kernel_stats_reset()
+        gq.hash_agg(keys, vals, cap, workspace=ws, max_groups=groups + 2, ops=gq.AGG_SUM)
+        ms, _ = gq.kernel_stats("agg_build")
+        print(f"LDS-agg groups={groups:>9,} cap={cap:>7} build={ms:7.2f}ms", flush=True)
+        del keys, vals, ws
+        torch.cuda.empty_cache()
