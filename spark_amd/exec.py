@@ -293,14 +293,12 @@ class GpuHashAggregateExec(SparkPlan):
             ops |= gpuq.AGG_COUNT
         for batch in self.children[0].execute_columnar():
             keys = batch.column(self.group_key)
-            if self.mode == "final":
-                # merge partial sums/counts: sum of sums; counts via SUM of
-                # partial counts cast to f64 is wrong — merge uses SUM sem.
-                # For round 1 final-merge sums only (counts merged as sums
-                # of int-valued f64 is exact below 2^53).
-                vals = batch.column(val_col).to(torch.float64)
-            else:
-                vals = batch.column(val_col)
+            vals = batch.column(val_col)
+            if vals.dtype == torch.int64:
+                # COUNT over any column / final-mode merge of partial counts:
+                # values ride as f64 (exact below 2^53); SUM(int64)'s int64
+                # result type is a round-2 item (Sum.scala resultType)
+                vals = gpuq.cast_i64_f64(vals)
             n = keys.numel()
             cap = self.capacity or (1 << max(10, int(n).bit_length()))
             out = gpuq.hash_agg(keys, vals, cap, ops=ops or gpuq.AGG_SUM,
