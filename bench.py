@@ -190,6 +190,50 @@ class JoinWorkload:
         del self.bkeys, self.bpay, self.pkeys, self.ppay, self.part_ws, self.join_ws, self.probe_ws
 
 
+class Q1Workload:
+    """Config-5 progress probe (single-GPU slice of a TPC-H Q1-shaped plan):
+    filter (shipdate <= cut) -> project (disc_price = price*(1-disc)) ->
+    GROUP BY returnflag (low cardinality => the LDS pre-agg path) with
+    SUM(disc_price) + COUNT. Driven through the exec-node layer (the same
+    operator objects a ColumnarRule would produce)."""
+
+    def __init__(self, gq, rows, rank):
+        from spark_amd import exec as gx
+        self.gx = gx
+        off = rank * 7
+        self.cols = dict(
+            retflag=gq.gen_i64(seed=71 + off, n=rows, range_=6),
+            qty=gq.gen_f64_unit(seed=72 + off, n=rows),
+            price=gq.gen_f64_unit(seed=73 + off, n=rows),
+            disc=gq.gen_f64_unit(seed=74 + off, n=rows),
+            shipdate=gq.gen_f64_unit(seed=75 + off, n=rows),
+        )
+        self.rows = rows
+        self.ngroups = 0
+
+    def step(self):
+        gx = self.gx
+        scan = gx.InputBatches([gx.ColumnarBatch(dict(self.cols))])
+        plan = gx.HashAggregateExec(
+            "retflag", [("sum", "disc_price"), ("count", "disc_price")],
+            "complete",
+            gx.ProjectExec(
+                ["retflag", ("disc_price", "price", "*", "one_minus_disc", None)],
+                gx.ProjectExec(
+                    ["retflag", "price", ("one_minus_disc", "neg_disc", "+", None, 1)],
+                    gx.ProjectExec(
+                        ["retflag", "price", ("neg_disc", "disc", "*", None, -1)],
+                        gx.FilterExec("shipdate", "<=", 0.98, scan)))),
+            capacity=64)
+        plan = gx.GpuColumnarRule().pre_columnar_transitions(plan)
+        out = next(plan.execute_columnar())
+        self.ngroups = out.num_rows()
+        out.close()
+
+    def free(self):
+        self.cols.clear()
+
+
 def time_workload(w, steps, warmup, world):
     for _ in range(warmup):
         w.step()
@@ -229,7 +273,7 @@ def main():
     p.add_argument("--quick", action="store_true", help="small sizes (CI/sanity)")
     p.add_argument("--no-cpu-baseline", action="store_true")
     p.add_argument("--cpu-sample-rows", type=int, default=40_000_000)
-    p.add_argument("--workloads", default="sort,agg,join")
+    p.add_argument("--workloads", default="sort,agg,join,q1")
     args = p.parse_args()
     if args.quick:
         args.rows, args.agg_rows, args.agg_groups, args.join_rows = \
@@ -288,6 +332,17 @@ def main():
             "matches_local": nm,
             "exchange": "rccl_all_to_all" if world > 1 else "none (single GPU)",
         }
+
+    if "q1" in wl:
+        w = Q1Workload(gq, args.join_rows, rank)
+        sec = time_workload(w, args.steps, args.warmup, world)
+        ng = w.ngroups
+        w.free()
+        torch.cuda.empty_cache()
+        results["q1"] = {"sec_per_step": sec,
+                         "rows_per_sec": args.join_rows * world / sec,
+                         "plan": "filter->project x3->groupby(sum,count)",
+                         "ngroups": ng}
 
     if rank != 0:
         return
