@@ -117,9 +117,11 @@ class SortExec(_CpuNode):
 
 
 class HashAggregateExec(_CpuNode):
-    def __init__(self, group_key: str, aggs: List[Tuple[str, str]], mode: str, child):
+    def __init__(self, group_key: str, aggs: List[Tuple[str, str]], mode: str, child,
+                 capacity: Optional[int] = None):
         super().__init__(child)
         self.group_key, self.aggs, self.mode = group_key, aggs, mode
+        self.capacity = capacity
 
     @property
     def output(self):
@@ -534,7 +536,7 @@ class GpuColumnarRule:
             return GpuSortExec(plan.sort_order, plan.global_sort, *children)
         if isinstance(plan, HashAggregateExec):
             return GpuHashAggregateExec(plan.group_key, plan.aggs, plan.mode,
-                                        *children)
+                                        *children, capacity=plan.capacity)
         if isinstance(plan, ShuffledHashJoinExec):
             return GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
                                            plan.build_side, *children)
