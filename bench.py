@@ -82,17 +82,20 @@ class SortWorkload:
         self.ws = gq.sort_workspace(rows)
         self.out1 = torch.empty(rows, dtype=torch.int64, device="cuda")
         self.out2 = torch.empty(rows, dtype=torch.float64, device="cuda")
+        self.perm = torch.empty(rows, dtype=torch.int32, device="cuda")
+        self.skeys = torch.empty(rows, dtype=torch.int64, device="cuda")
 
     def step(self):
         gq = self.gq
-        perm, skeys = gq.sort_perm(self.keys, workspace=self.ws)
+        perm, skeys = gq.sort_perm(self.keys, workspace=self.ws,
+                                   out_perm=self.perm, out_keys=self.skeys)
         gq.lib().gpuq_gather(gq._stream(), self.rows, gq._col(self.pay1),
                              perm.data_ptr(), self.out1.data_ptr())
         gq.lib().gpuq_gather(gq._stream(), self.rows, gq._col(self.pay2),
                              perm.data_ptr(), self.out2.data_ptr())
 
     def free(self):
-        del self.keys, self.pay1, self.pay2, self.ws, self.out1, self.out2
+        del self.keys, self.pay1, self.pay2, self.ws, self.out1, self.out2, self.perm, self.skeys
 
 
 class AggWorkload:

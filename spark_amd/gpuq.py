@@ -157,14 +157,20 @@ def sort_workspace(n: int, device="cuda") -> torch.Tensor:
 
 
 def sort_perm(keys: torch.Tensor, desc=False, nulls_first=None, workspace=None,
-              out_keys: bool = True, key_validity=None):
+              out_keys=True, key_validity=None, out_perm=None):
+    """out_keys: True (allocate), False/None (skip), or a preallocated tensor.
+    out_perm: optional preallocated int32 tensor (large callers reuse buffers
+    to avoid fresh multi-GB hipMallocs per call)."""
     n = keys.numel()
     if nulls_first is None:
         nulls_first = not desc
     if workspace is None:
         workspace = sort_workspace(n, keys.device)
-    perm = torch.empty(n, dtype=torch.int32, device=keys.device)  # u32 bits
-    ok = torch.empty(n, dtype=keys.dtype, device=keys.device) if out_keys else None
+    perm = out_perm if out_perm is not None else         torch.empty(n, dtype=torch.int32, device=keys.device)  # u32 bits
+    if isinstance(out_keys, torch.Tensor):
+        ok = out_keys
+    else:
+        ok = torch.empty(n, dtype=keys.dtype, device=keys.device) if out_keys else None
     _check(lib().gpuq_sort_perm(_stream(), n, _col(keys, key_validity), int(desc), int(nulls_first),
                                 perm.data_ptr(), _dp(ok),
                                 workspace.data_ptr(), workspace.numel()))
