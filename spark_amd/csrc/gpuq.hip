@@ -837,14 +837,20 @@ retry:
     HIP_TRY(hipMemsetAsync(w.err, 0, 8, s));
   }
 
+  /* last active pass writes row ids straight into out_perm (saves the
+   * 4 B/row device copy) */
+  int last_byte = -1;
+  for (int byte = 0; byte < 8; byte++)
+    if (((bits_changed >> (byte * 8)) & 0xff) != 0) last_byte = byte;
   uint64_t *kin = w.ka, *kout = w.kb;
   uint32_t *iin = w.ia, *iout = w.ib;
   for (int byte = 0; byte < 8; byte++) {
     if (((bits_changed >> (byte * 8)) & 0xff) == 0) continue;  /* RadixSort.java:126 skip */
     int shift = byte * 8;
+    uint32_t* iout_pass = (byte == last_byte) ? out_perm + sorted_at : iout;
     if (onesweep) {
       { hipEvent_t _pe = prof_begin(s);
-      launch_scatter<0, true>(s, geom, nb, n_sort, kin, iin, kout, iout, nullptr,
+      launch_scatter<0, true>(s, geom, nb, n_sort, kin, iin, kout, iout_pass, nullptr,
                               shift, 0, w.state, w.gbase + byte * 256, w.err,
                               (uint32_t)(byte + 1));
       prof_end("radix_scatter", s, _pe); }
@@ -857,13 +863,13 @@ retry:
       int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
       if (rc) return rc;
       { hipEvent_t _pe = prof_begin(s);
-      launch_scatter<0, false>(s, geom, nb, n_sort, kin, iin, kout, iout, w.hist_scan,
+      launch_scatter<0, false>(s, geom, nb, n_sort, kin, iin, kout, iout_pass, w.hist_scan,
                                shift, 0, nullptr, nullptr, nullptr, 0);
       prof_end("radix_scatter", s, _pe); }
       HIP_TRY(hipGetLastError());
     }
     uint64_t* tk = kin; kin = kout; kout = tk;
-    uint32_t* ti = iin; iin = iout; iout = ti;
+    uint32_t* ti = iin; iin = iout_pass; iout = ti;
   }
 
   if (onesweep) {
@@ -890,7 +896,8 @@ retry:
       goto retry;
     }
   }
-  HIP_TRY(hipMemcpyAsync(out_perm + sorted_at, iin, n_sort * 4, hipMemcpyDeviceToDevice, s));
+  if (last_byte < 0)  /* zero passes: identity permutation */
+    HIP_TRY(hipMemcpyAsync(out_perm + sorted_at, iin, n_sort * 4, hipMemcpyDeviceToDevice, s));
   if (out_keys) {
     if (key.validity) {
       /* null rows' data values are whatever the input held — gather the
