@@ -146,6 +146,24 @@ int gpuq_hash_agg_i64_f64(void* stream, int64_t nrows,
                           double* out_sums, uint8_t* out_sum_valid,
                           int64_t* out_counts, int64_t* out_ngroups);
 
+/* Multi-aggregate: one pass computing up to 6 accumulators per group
+ * (HashAggregateExec evaluates a list of aggregate expressions,
+ * HashAggregateExec.scala:68-76 — e.g. TPC-H Q1's 8 aggregates).
+ * spec_ops[j]: 0=SUM(float64 col), 1=COUNT(col), 2=COUNT(*);
+ * spec_cols[j] indexes vals[] (ignored for COUNT(*)). out_accs[j] is a
+ * device array per spec: f64 for SUM, i64 for COUNT. Small tables
+ * (cap*(1+nspecs)*8 <= 64 KB) aggregate per-block in LDS first.
+ * Round-1: SUM value columns must be non-null (pair with COUNT for NULL
+ * tracking). */
+int64_t gpuq_hash_agg_multi_workspace_bytes(int64_t capacity, int32_t nspecs);
+int gpuq_hash_agg_multi(void* stream, int64_t nrows, gpuq_col key,
+                        const gpuq_col* vals, const int32_t* spec_ops,
+                        const int32_t* spec_cols, int32_t nspecs,
+                        void* workspace, int64_t capacity,
+                        int32_t first_batch, int32_t finalize,
+                        int64_t* out_keys, uint8_t* out_key_valid,
+                        void* const* out_accs, int64_t* out_ngroups);
+
 /* ---------------------------------------------------------------- */
 /* PARTITION — replaces ShuffleExchangeExec's partition-id + write   */
 /* path (exchange/ShuffleExchangeExec.scala:357-470 +                */

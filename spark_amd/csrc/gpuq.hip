@@ -1676,6 +1676,275 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
   return GPUQ_OK;
 }
 
+/* ================= multi-aggregate ================= */
+/*
+ * One pass computing up to GPUQ_AGG_MAX_SPECS accumulators per group
+ * (HashAggregateExec evaluates a LIST of aggregate expressions —
+ * HashAggregateExec.scala:68-76; TPC-H Q1 has 8). Slot layout:
+ * [key][acc_0]..[acc_{A-1}], stride = 1+A u64 words. SUM accs are f64
+ * bits, COUNT accs u64. Small tables (cap*(1+A)*8 <= 64 KB) aggregate in
+ * a per-block LDS table first (contention wall removal, same as the
+ * single-value path). Round-1 constraint: SUM value columns must be
+ * non-null (per-acc NULL tracking pairs a SUM with a COUNT spec).
+ */
+
+#define AGG_MAX_SPECS 6
+
+struct agg_cols { const double* p[AGG_MAX_SPECS]; };
+struct agg_ops_spec { int op[AGG_MAX_SPECS]; };  /* 0=SUM 1=COUNT(col) 2=COUNT(*) */
+struct agg_valid { const uint8_t* v[AGG_MAX_SPECS]; };
+struct agg_outs { void* p[AGG_MAX_SPECS]; };
+
+struct agg_multi_special {
+  unsigned long long acc[2][AGG_MAX_SPECS];  /* [0]=key -1 group, [1]=NULL group */
+  unsigned long long seen[2];
+  unsigned long long out_cursor;
+  unsigned long long overflow;
+};
+
+struct agg_multi_ws {
+  unsigned long long* tab;
+  agg_multi_special* sp;
+};
+
+static void agg_multi_ws_layout(int64_t cap, int nspecs, agg_multi_ws* w,
+                                char* base, int64_t* total) {
+  int64_t off = 0;
+  auto take = [&](int64_t bytes) {
+    char* p = base ? base + off : nullptr;
+    off += (bytes + 255) & ~255LL;
+    return p;
+  };
+  w->tab = (unsigned long long*)take(cap * 8 * (1 + nspecs));
+  w->sp = (agg_multi_special*)take(sizeof(agg_multi_special));
+  *total = off;
+}
+
+extern "C" int64_t gpuq_hash_agg_multi_workspace_bytes(int64_t cap, int32_t nspecs) {
+  agg_multi_ws w; int64_t total;
+  agg_multi_ws_layout(cap, nspecs, &w, nullptr, &total);
+  return total;
+}
+
+__global__ void k_aggm_init(int64_t cap, int stride, unsigned long long* tab) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < cap; i += gs) {
+    tab[(int64_t)stride * i] = AGG_EMPTY;
+    for (int j = 1; j < stride; j++) tab[(int64_t)stride * i + j] = 0;
+  }
+}
+
+DEV void aggm_update(unsigned long long* acc, int nspecs, const agg_ops_spec ops,
+                     const agg_cols cols, const agg_valid av, int64_t i) {
+  for (int j = 0; j < nspecs; j++) {
+    int op = ops.op[j];
+    if (op == 0) {
+      atomicAdd((double*)&acc[j], cols.p[j][i]);
+    } else if (op == 1) {
+      if (bit_valid(av.v[j], i)) atomicAdd(&acc[j], 1ull);
+    } else {
+      atomicAdd(&acc[j], 1ull);
+    }
+  }
+}
+
+template <bool LDS>
+__global__ __launch_bounds__(256)
+void k_aggm_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
+                  agg_cols cols, agg_ops_spec ops, agg_valid av, int nspecs,
+                  unsigned long long* tab, agg_multi_special* sp,
+                  int64_t cap_mask) {
+  const int stride = 1 + nspecs;
+  extern __shared__ __attribute__((aligned(16))) unsigned long long lt[];
+  if (LDS) {
+    for (int j = threadIdx.x; j < (int)(cap_mask + 1) * stride; j += blockDim.x)
+      lt[j] = (j % stride == 0) ? AGG_EMPTY : 0;
+    __syncthreads();
+  }
+  unsigned long long* t = LDS ? lt : tab;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += gs) {
+    bool kv = bit_valid(kvalid, i);
+    int64_t k = kv ? keys[i] : 0;
+    if (!kv || (unsigned long long)k == AGG_EMPTY) {
+      int which = kv ? 0 : 1;
+      atomicMax(&sp->seen[which], 1ull);
+      aggm_update(sp->acc[which], nspecs, ops, cols, av, i);
+      continue;
+    }
+    uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
+    for (int probes = 0;; probes++) {
+      unsigned long long cur = LDS ? t[(int64_t)stride * slot]
+          : __hip_atomic_load(&t[(int64_t)stride * slot], __ATOMIC_RELAXED,
+                              __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == (unsigned long long)k) break;
+      if (cur == AGG_EMPTY) {
+        unsigned long long prev = atomicCAS(&t[(int64_t)stride * slot], AGG_EMPTY,
+                                            (unsigned long long)k);
+        if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
+      }
+      slot = (slot + 1) & (uint64_t)cap_mask;
+      if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
+    }
+    aggm_update(&t[(int64_t)stride * slot + 1], nspecs, ops, cols, av, i);
+  }
+  if (LDS) {
+    __syncthreads();
+    for (int sI = threadIdx.x; sI < (int)(cap_mask + 1); sI += blockDim.x) {
+      unsigned long long k = lt[(int64_t)stride * sI];
+      if (k == AGG_EMPTY) continue;
+      uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)k, 42)) & (uint64_t)cap_mask;
+      for (;;) {
+        unsigned long long cur = __hip_atomic_load(&tab[(int64_t)stride * slot],
+                                                   __ATOMIC_RELAXED,
+                                                   __HIP_MEMORY_SCOPE_AGENT);
+        if (cur == k) break;
+        if (cur == AGG_EMPTY) {
+          unsigned long long prev = atomicCAS(&tab[(int64_t)stride * slot], AGG_EMPTY, k);
+          if (prev == AGG_EMPTY || prev == k) break;
+        }
+        slot = (slot + 1) & (uint64_t)cap_mask;
+      }
+      for (int j = 0; j < nspecs; j++) {
+        unsigned long long v = lt[(int64_t)stride * sI + 1 + j];
+        if (ops.op[j] == 0)
+          atomicAdd((double*)&tab[(int64_t)stride * slot + 1 + j],
+                    __longlong_as_double((long long)v));
+        else
+          atomicAdd(&tab[(int64_t)stride * slot + 1 + j], v);
+      }
+    }
+  }
+}
+
+__global__ void k_aggm_compact(int64_t cap, const unsigned long long* tab,
+                               agg_multi_special* sp, agg_ops_spec ops, int nspecs,
+                               int64_t* out_keys, uint8_t* out_kvalid,
+                               agg_outs outs) {
+  constexpr int ROUNDS = AGGC_CHUNK / 256;
+  const int stride = 1 + nspecs;
+  __shared__ unsigned long long block_base;
+  __shared__ uint32_t wtot[4];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = threadIdx.x / WAVE;
+  const int64_t base = (int64_t)blockIdx.x * AGGC_CHUNK;
+  uint32_t occ_mask[ROUNDS / 32 + 1];
+  for (int m = 0; m < ROUNDS / 32 + 1; m++) occ_mask[m] = 0;
+  uint32_t lane_total = 0;
+  for (int r = 0; r < ROUNDS; r++) {
+    int64_t i = base + r * 256 + threadIdx.x;
+    bool occ = i < cap && tab[(int64_t)stride * i] != AGG_EMPTY;
+    if (occ) { occ_mask[r / 32] |= 1u << (r & 31); lane_total++; }
+  }
+  uint32_t incl = wave_inclusive_scan(lane_total);
+  if (lane == WAVE - 1) wtot[wave] = incl;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint32_t tot = 0;
+    for (int w = 0; w < 4; w++) { uint32_t t = wtot[w]; wtot[w] = tot; tot += t; }
+    block_base = tot ? atomicAdd(&sp->out_cursor, (unsigned long long)tot) : 0;
+  }
+  __syncthreads();
+  int64_t o = (int64_t)block_base + wtot[wave] + (incl - lane_total);
+  for (int r = 0; r < ROUNDS; r++) {
+    if (!((occ_mask[r / 32] >> (r & 31)) & 1)) continue;
+    int64_t i = base + r * 256 + threadIdx.x;
+    out_keys[o] = (int64_t)tab[(int64_t)stride * i];
+    out_kvalid[o] = 1;
+    for (int j = 0; j < nspecs; j++) {
+      unsigned long long v = tab[(int64_t)stride * i + 1 + j];
+      if (ops.op[j] == 0) ((double*)outs.p[j])[o] = __longlong_as_double((long long)v);
+      else ((int64_t*)outs.p[j])[o] = (int64_t)v;
+    }
+    o++;
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    for (int which = 0; which < 2; which++) {
+      if (!sp->seen[which]) continue;
+      int64_t q = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
+      out_keys[q] = which == 0 ? -1 : 0;
+      out_kvalid[q] = which == 0 ? 1 : 0;
+      for (int j = 0; j < nspecs; j++) {
+        unsigned long long v = sp->acc[which][j];
+        if (ops.op[j] == 0) ((double*)outs.p[j])[q] = __longlong_as_double((long long)v);
+        else ((int64_t*)outs.p[j])[q] = (int64_t)v;
+      }
+    }
+  }
+}
+
+extern "C" int gpuq_hash_agg_multi(void* stream, int64_t n, gpuq_col key,
+                                   const gpuq_col* vals, const int32_t* spec_ops,
+                                   const int32_t* spec_cols, int32_t nspecs,
+                                   void* workspace, int64_t cap,
+                                   int32_t first_batch, int32_t finalize,
+                                   int64_t* out_keys, uint8_t* out_key_valid,
+                                   void* const* out_accs, int64_t* out_ngroups) {
+  hipStream_t s = (hipStream_t)stream;
+  if (cap <= 0 || (cap & (cap - 1)))
+    FAIL(GPUQ_ERR_INVALID, "aggm: capacity %lld not a power of two", (long long)cap);
+  if (nspecs < 1 || nspecs > AGG_MAX_SPECS)
+    FAIL(GPUQ_ERR_INVALID, "aggm: nspecs %d not in [1,%d]", nspecs, AGG_MAX_SPECS);
+  if (key.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "aggm: key must be int64");
+  agg_cols cols = {}; agg_ops_spec ops = {}; agg_valid av = {};
+  for (int j = 0; j < nspecs; j++) {
+    ops.op[j] = spec_ops[j];
+    if (spec_ops[j] == 0) {
+      const gpuq_col& c = vals[spec_cols[j]];
+      if (c.dtype != GPUQ_FLOAT64) FAIL(GPUQ_ERR_INVALID, "aggm: SUM col must be float64");
+      if (c.validity) FAIL(GPUQ_ERR_INVALID, "aggm: SUM cols must be non-null "
+                           "(pair with a COUNT spec for NULL tracking)");
+      cols.p[j] = (const double*)c.data;
+    } else if (spec_ops[j] == 1) {
+      av.v[j] = vals[spec_cols[j]].validity;
+    } else if (spec_ops[j] != 2) {
+      FAIL(GPUQ_ERR_INVALID, "aggm: bad op %d", spec_ops[j]);
+    }
+  }
+  int stride = 1 + nspecs;
+  agg_multi_ws w; int64_t need;
+  agg_multi_ws_layout(cap, nspecs, &w, (char*)workspace, &need);
+  if (first_batch) {
+    k_aggm_init<<<grid1d(cap), 256, 0, s>>>(cap, stride, w.tab);
+    HIP_TRY(hipGetLastError());
+    HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_multi_special), s));
+  }
+  if (n > 0) {
+    int64_t lds_bytes = cap * stride * 8;
+    bool lds = lds_bytes <= (64 << 10) && getenv("GPUQ_NO_LDS_AGG") == nullptr;
+    { hipEvent_t _pe = prof_begin(s);
+    if (lds)
+      k_aggm_build<true><<<hash_grid(n), 256, (uint32_t)lds_bytes, s>>>(
+          n, (const int64_t*)key.data, key.validity, cols, ops, av, nspecs,
+          w.tab, w.sp, cap - 1);
+    else
+      k_aggm_build<false><<<hash_grid(n), 256, 0, s>>>(
+          n, (const int64_t*)key.data, key.validity, cols, ops, av, nspecs,
+          w.tab, w.sp, cap - 1);
+    prof_end("aggm_build", s, _pe); }
+    HIP_TRY(hipGetLastError());
+  }
+  if (finalize) {
+    agg_multi_special hsp;
+    HIP_TRY(hipMemcpyAsync(&hsp, w.sp, sizeof(hsp), hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "aggm: hash table overflow (capacity %lld)", (long long)cap);
+    agg_outs outs = {};
+    for (int j = 0; j < nspecs; j++) outs.p[j] = out_accs[j];
+    dim3 cgrid((uint32_t)((cap + AGGC_CHUNK - 1) / AGGC_CHUNK));
+    k_aggm_compact<<<cgrid, 256, 0, s>>>(cap, w.tab, w.sp, ops, nspecs,
+                                         out_keys, out_key_valid, outs);
+    HIP_TRY(hipGetLastError());
+    agg_multi_special hsp2;
+    HIP_TRY(hipMemcpyAsync(&hsp2, w.sp, sizeof(hsp2), hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    *out_ngroups = (int64_t)hsp2.out_cursor;
+  }
+  return GPUQ_OK;
+}
+
 /* ================= filter / project (SURVEY (f).2) ================= */
 /*
  * FILTER — replaces FilterExec for single-comparison predicates

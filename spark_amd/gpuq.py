@@ -66,6 +66,12 @@ def lib() -> ctypes.CDLL:
         L.gpuq_hash_agg_i64_f64.restype = i32
         L.gpuq_hash_agg_i64_f64.argtypes = [vp, i64, _Col, _Col, vp, i64, i32, i32, i32,
                                             vp, vp, vp, vp, vp, ctypes.POINTER(i64)]
+        L.gpuq_hash_agg_multi_workspace_bytes.restype = i64
+        L.gpuq_hash_agg_multi_workspace_bytes.argtypes = [i64, i32]
+        L.gpuq_hash_agg_multi.restype = i32
+        L.gpuq_hash_agg_multi.argtypes = [vp, i64, _Col, vp, vp, vp, i32,
+                                          vp, i64, i32, i32, vp, vp, vp,
+                                          ctypes.POINTER(i64)]
         L.gpuq_partition_workspace_bytes.restype = i64
         L.gpuq_partition_workspace_bytes.argtypes = [i64, i32]
         L.gpuq_partition_perm.restype = i32
@@ -317,3 +323,46 @@ def range_i64(n: int, start: int = 0, step: int = 1, device="cuda") -> torch.Ten
     out = torch.empty(n, dtype=torch.int64, device=device)
     _check(lib().gpuq_range_i64(_stream(), n, start, step, out.data_ptr()))
     return out
+
+
+def hash_agg_multi(keys: torch.Tensor, specs, capacity: int, key_validity=None,
+                   max_groups=None):
+    """specs: list of ("sum", col_tensor) / ("count", col_tensor, validity) /
+    ("count*", None). Returns (keys, key_valid, [acc_j...]) sliced to
+    ngroups; SUM accs f64, COUNT accs i64."""
+    n = keys.numel()
+    dev = keys.device
+    nspecs = len(specs)
+    cols, ops, colidx, outs = [], [], [], []
+    for s in specs:
+        kind = s[0]
+        if kind == "sum":
+            ops.append(0); colidx.append(len(cols)); cols.append(_col(s[1]))
+            outs.append(torch.empty(max_groups or capacity + 2,
+                                    dtype=torch.float64, device=dev))
+        elif kind == "count":
+            ops.append(1); colidx.append(len(cols))
+            cols.append(_col(s[1], s[2] if len(s) > 2 else None))
+            outs.append(torch.empty(max_groups or capacity + 2,
+                                    dtype=torch.int64, device=dev))
+        else:
+            ops.append(2); colidx.append(0)
+            outs.append(torch.empty(max_groups or capacity + 2,
+                                    dtype=torch.int64, device=dev))
+    if not cols:
+        cols = [_col(keys)]  # placeholder, unused
+    cols_arr = (_Col * len(cols))(*cols)
+    ops_arr = (ctypes.c_int32 * nspecs)(*ops)
+    cid_arr = (ctypes.c_int32 * nspecs)(*colidx)
+    outp_arr = (ctypes.c_void_p * nspecs)(*[t.data_ptr() for t in outs])
+    ws = torch.empty(lib().gpuq_hash_agg_multi_workspace_bytes(capacity, nspecs),
+                     dtype=torch.uint8, device=dev)
+    ok = torch.empty(max_groups or capacity + 2, dtype=torch.int64, device=dev)
+    okv = torch.empty(max_groups or capacity + 2, dtype=torch.uint8, device=dev)
+    ng = ctypes.c_int64(0)
+    _check(lib().gpuq_hash_agg_multi(
+        _stream(), n, _col(keys, key_validity), cols_arr, ops_arr, cid_arr,
+        nspecs, ws.data_ptr(), capacity, 1, 1, ok.data_ptr(), okv.data_ptr(),
+        outp_arr, ctypes.byref(ng)))
+    gn = ng.value
+    return ok[:gn], okv[:gn], [t[:gn] for t in outs]

@@ -282,3 +282,40 @@ def test_agg_multi_batch_accumulation(gq):
     go, oo = np.argsort(gk), np.argsort(ok)
     assert (gk[go] == ok[oo]).all() and (gc[go] == ocnt[oo]).all()
     np.testing.assert_allclose(gs[go], osum[oo], rtol=1e-6)
+
+
+def test_multi_aggregate(gq):
+    """One pass, 4 accumulators (the Q1 shape): SUM(a), SUM(b), COUNT(b),
+    COUNT(*) — vs oracle per-accumulator."""
+    n, ngroups = 400_000, 6
+    keys = oracle.gen_i64(seed=90, n=n, range_=ngroups)
+    a = oracle.gen_f64_unit(seed=91, n=n)
+    b = oracle.gen_f64_unit(seed=92, n=n)
+    da, db = to_dev(a), to_dev(b)
+    ok, okv, accs = gq.hash_agg_multi(
+        to_dev(keys), [("sum", da), ("sum", db), ("count", db), ("count*", None)],
+        capacity=64, max_groups=16)
+    oka, _, osuma, _, _ = oracle.hash_agg(keys, a)
+    okb, _, osumb, _, ocntb = oracle.hash_agg(keys, b)
+    g = np.argsort(ok.cpu().numpy())
+    oa = np.argsort(oka)
+    assert (ok.cpu().numpy()[g] == oka[oa]).all()
+    np.testing.assert_allclose(accs[0].cpu().numpy()[g], osuma[oa], rtol=1e-6)
+    np.testing.assert_allclose(accs[1].cpu().numpy()[g], osumb[oa], rtol=1e-6)
+    assert (accs[2].cpu().numpy()[g] == ocntb[oa]).all()
+    assert (accs[3].cpu().numpy()[g] == ocntb[oa]).all()  # no nulls: count*==count
+
+
+def test_multi_aggregate_large_table(gq):
+    # global (non-LDS) path: many groups
+    n, ngroups = 500_000, 100_000
+    keys = oracle.gen_i64(seed=93, n=n, range_=ngroups)
+    a = oracle.gen_f64_unit(seed=94, n=n)
+    ok, okv, accs = gq.hash_agg_multi(
+        to_dev(keys), [("sum", to_dev(a)), ("count*", None)],
+        capacity=1 << 18, max_groups=ngroups + 2)
+    oka, _, osuma, _, ocnt = oracle.hash_agg(keys, a)
+    g, oa = np.argsort(ok.cpu().numpy()), np.argsort(oka)
+    assert (ok.cpu().numpy()[g] == oka[oa]).all()
+    np.testing.assert_allclose(accs[0].cpu().numpy()[g], osuma[oa], rtol=1e-6)
+    assert (accs[1].cpu().numpy()[g] == ocnt[oa]).all()
