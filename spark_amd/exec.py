@@ -287,6 +287,10 @@ class GpuHashAggregateExec(SparkPlan):
         from . import gpuq
         fns = {fn for fn, _ in self.aggs}
         assert fns <= {"sum", "count", "avg"}, f"unsupported aggs {fns}"
+        cols_used = {col for _, col in self.aggs}
+        if len(cols_used) > 1 or len(self.aggs) > 2:
+            yield from self._execute_multi(gpuq)
+            return
         val_col = next(col for fn, col in self.aggs)
         ops = 0
         if fns & {"sum", "avg"} or self.mode == "final":
@@ -316,6 +320,41 @@ class GpuHashAggregateExec(SparkPlan):
                         osum, "/", b=gpuq.cast_i64_f64(ocnt))
                 else:
                     cols[f"{fn}({col})"] = osum if fn == "sum" else ocnt
+            batch.close()
+            yield ColumnarBatch(cols, validity={self.group_key: None})
+
+    def _execute_multi(self, gpuq):
+        """multi-accumulator path (one gpuq_hash_agg_multi pass): several
+        aggregate expressions over multiple value columns (the Q1 shape)."""
+        for batch in self.children[0].execute_columnar():
+            keys = batch.column(self.group_key)
+            specs = []
+            slots = []  # (fn, col, acc indices)
+            for fn, col in self.aggs:
+                t = batch.column(col)
+                if t.dtype == torch.int64:
+                    t = gpuq.cast_i64_f64(t)
+                if fn == "sum":
+                    specs.append(("sum", t)); slots.append((fn, col, [len(specs) - 1]))
+                elif fn == "count":
+                    specs.append(("count", t, batch.validity(col)))
+                    slots.append((fn, col, [len(specs) - 1]))
+                else:  # avg = sum + count
+                    specs.append(("sum", t))
+                    specs.append(("count", t, batch.validity(col)))
+                    slots.append((fn, col, [len(specs) - 2, len(specs) - 1]))
+            n = keys.numel()
+            cap = self.capacity or (1 << max(10, int(n).bit_length()))
+            ok, okv, accs = gpuq.hash_agg_multi(
+                keys, specs, cap, key_validity=batch.validity(self.group_key),
+                max_groups=min(n, cap) + 2)
+            cols = {self.group_key: ok}
+            for fn, col, idx in slots:
+                if fn == "avg":
+                    cols[f"avg({col})"] = gpuq.project_binop(
+                        accs[idx[0]], "/", b=gpuq.cast_i64_f64(accs[idx[1]]))
+                else:
+                    cols[f"{fn}({col})"] = accs[idx[0]]
             batch.close()
             yield ColumnarBatch(cols, validity={self.group_key: None})
 

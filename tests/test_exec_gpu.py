@@ -148,3 +148,24 @@ def test_range_scan_feed():
     g = np.argsort(got["g"])
     assert (got["g"][g] == uk).all()
     assert (got["count(g)"][g] == cnt).all()
+
+
+def test_q1_shape_multi_agg_plan():
+    """The full Q1-shaped aggregate through the exec layer: several SUMs over
+    different columns + AVG + COUNT in one pass."""
+    n = 150_000
+    keys = oracle.gen_i64(seed=50, n=n, range_=6)
+    qty = oracle.gen_f64_unit(seed=51, n=n)
+    price = oracle.gen_f64_unit(seed=52, n=n)
+    scan = gx.InputBatches([dev_batch(k=keys, qty=qty, price=price)])
+    got = run_plan(gx.HashAggregateExec(
+        "k", [("sum", "qty"), ("sum", "price"), ("avg", "qty"), ("count", "price")],
+        "complete", scan, capacity=64))
+    ok, _, osq, _, ocnt = oracle.hash_agg(keys, qty)
+    _, _, osp, _, _ = oracle.hash_agg(keys, price)
+    g, o = np.argsort(got["k"]), np.argsort(ok)
+    assert (got["k"][g] == ok[o]).all()
+    np.testing.assert_allclose(got["sum(qty)"][g], osq[o], rtol=1e-6)
+    np.testing.assert_allclose(got["sum(price)"][g], osp[o], rtol=1e-6)
+    np.testing.assert_allclose(got["avg(qty)"][g], osq[o] / ocnt[o], rtol=1e-6)
+    assert (got["count(price)"][g] == ocnt[o]).all()
