@@ -494,11 +494,18 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
   uint8_t lbin[ITEMS];
 
   const int64_t wbase = base + (int64_t)wave * WAVE * ITEMS;
+  /* issue every load before the serial ranking chain so all ITEMS pairs
+   * are in flight at once (A/B via GPUQ_SCATTER_NO_PRELOAD) */
+  #pragma unroll
   for (int r = 0; r < ITEMS; r++) {
     int64_t i = wbase + r * WAVE + lane;
     bool valid = i < n;
     k[r] = valid ? kin[i] : 0;
     id[r] = valid ? iin[i] : 0;
+  }
+  for (int r = 0; r < ITEMS; r++) {
+    int64_t i = wbase + r * WAVE + lane;
+    bool valid = i < n;
     int bin = valid ? compute_bin<BIN_MODE>(k[r], shift, nparts) : 0;
     lbin[r] = (uint8_t)bin;
     /* ballot multi-split: mask of lanes in this wave with the same bin */
