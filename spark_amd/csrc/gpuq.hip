@@ -503,32 +503,25 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
     k[r] = valid ? kin[i] : 0;
     id[r] = valid ? iin[i] : 0;
   }
-  /* ballot multi-split masks for every round are independent - compute
-   * them all first; only the per-bin LDS accumulation stays serial */
-  uint64_t same[ITEMS];
-  #pragma unroll
   for (int r = 0; r < ITEMS; r++) {
     int64_t i = wbase + r * WAVE + lane;
     bool valid = i < n;
     int bin = valid ? compute_bin<BIN_MODE>(k[r], shift, nparts) : 0;
     lbin[r] = (uint8_t)bin;
-    uint64_t m = __ballot(valid);
+    /* ballot multi-split: mask of lanes in this wave with the same bin */
+    uint64_t active = __ballot(valid);
+    uint64_t same = active;
     for (int b = 0; b < 8; b++) {
       uint64_t bl = __ballot((bin >> b) & 1);
-      m &= ((bin >> b) & 1) ? bl : ~bl;
+      same &= ((bin >> b) & 1) ? bl : ~bl;
     }
-    same[r] = m;
-  }
-  for (int r = 0; r < ITEMS; r++) {
-    int bin = lbin[r];
-    uint64_t below = same[r] & ((1ULL << lane) - 1);
+    uint64_t below = same & ((1ULL << lane) - 1);
     int rank = __popcll(below);
-    int leader = __ffsll((unsigned long long)same[r]) - 1;
+    int leader = __ffsll((unsigned long long)same) - 1;
     uint32_t basecnt = 0;
-    bool valid = (same[r] >> lane) & 1;
     if (valid && lane == leader) {
       basecnt = wave_hist[wave][bin];
-      wave_hist[wave][bin] = basecnt + __popcll(same[r]);
+      wave_hist[wave][bin] = basecnt + __popcll(same);
     }
     basecnt = __shfl(basecnt, leader);
     lrank[r] = (uint16_t)(basecnt + rank);
