@@ -62,3 +62,29 @@ def shuffle_exchange_gpu(key: torch.Tensor, payload: Dict[str, torch.Tensor],
     out, _ = exchange_columns(cols, in_splits, group=group)
     k = out.pop("__key__")
     return k, out
+
+
+def broadcast_gather(cols: Dict[str, torch.Tensor], group=None) -> Dict[str, torch.Tensor]:
+    """BroadcastExchangeExec analog (SURVEY §8(f).3,
+    joins/BroadcastHashJoinExec.scala:40): every rank contributes its slice
+    of the (small) build side and receives the whole relation — an RCCL
+    all-gather(v) over xGMI instead of Spark's driver-collect + Netty
+    torrent broadcast. Received rows are source-rank-major, source order
+    preserved (deterministic build order on every rank)."""
+    world = dist.get_world_size(group)
+    device = next(iter(cols.values())).device
+    n_local = torch.tensor([next(iter(cols.values())).numel()],
+                           dtype=torch.int64, device=device)
+    sizes = torch.empty(world, dtype=torch.int64, device=device)
+    dist.all_gather_into_tensor(sizes, n_local, group=group)
+    counts = sizes.cpu().tolist()
+    mx = max(counts)
+    out = {}
+    for name, t in cols.items():
+        # all_gather needs equal sizes: pad to the max, slice after
+        padded = torch.empty(mx, dtype=t.dtype, device=device)
+        padded[:t.numel()] = t
+        parts = [torch.empty(mx, dtype=t.dtype, device=device) for _ in counts]
+        dist.all_gather(parts, padded, group=group)
+        out[name] = torch.cat([p[:c] for p, c in zip(parts, counts)])
+    return out

@@ -200,6 +200,30 @@ class RangeExec(_CpuNode):
         return [self.name]
 
 
+class BroadcastHashJoinExec(_CpuNode):
+    """CPU placeholder (joins/BroadcastHashJoinExec.scala:40): inner
+    equi-join with a broadcast (small) build side."""
+
+    def __init__(self, left_key: str, right_key: str, build_side: str,
+                 left, right):
+        super().__init__(left, right)
+        assert build_side in ("left", "right")
+        self.left_key, self.right_key, self.build_side = left_key, right_key, build_side
+
+    @property
+    def output(self):
+        return self.children[0].output + self.children[1].output
+
+
+class BroadcastExchangeExec(_CpuNode):
+    def __init__(self, child):
+        super().__init__(child)
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+
 class InputBatches(SparkPlan):
     """Leaf: pre-materialized device batches (scan stand-in)."""
 
@@ -562,6 +586,47 @@ class GpuProjectExec(SparkPlan):
             yield ColumnarBatch(cols)
 
 
+class GpuBroadcastExchangeExec(SparkPlan):
+    """Replaces BroadcastExchangeExec: RCCL all-gather of the build-side
+    batch across ranks (single-rank: identity). Unlike the shuffled
+    exchange there is no partitioning — every rank gets the whole
+    relation."""
+
+    def __init__(self, child):
+        super().__init__(child)
+
+    @property
+    def output(self):
+        return self.children[0].output
+
+    @property
+    def supports_columnar(self):
+        return True
+
+    def execute_columnar(self):
+        import torch.distributed as dist
+        from .exchange import broadcast_gather
+        for batch in self.children[0].execute_columnar():
+            if dist.is_initialized() and dist.get_world_size() > 1:
+                cols = broadcast_gather(batch.columns())
+                batch.close()
+                yield ColumnarBatch(cols)
+            else:
+                yield batch
+
+
+class GpuBroadcastHashJoinExec(GpuShuffledHashJoinExec):
+    """Replaces BroadcastHashJoinExec: identical device-side build/probe
+    kernels; the build child is expected to be a GpuBroadcastExchangeExec,
+    and the PROBE side needs no distribution at all (BroadcastDistribution,
+    joins/BroadcastHashJoinExec.scala:60-66)."""
+
+    def required_child_distribution(self):
+        dists = [Distribution("unspecified"), Distribution("unspecified")]
+        dists[0 if self.build_side == "left" else 1] = Distribution("broadcast")
+        return dists
+
+
 class GpuColumnarRule:
     """The injected rule (ColumnarRule, Columnar.scala:36-50; injection via
     SparkSessionExtensions.injectColumnar:168; applied at
@@ -592,6 +657,11 @@ class GpuColumnarRule:
             return GpuProjectExec(plan.projections, *children)
         if isinstance(plan, RangeExec):
             return GpuRangeExec(plan.n, plan.start, plan.step, plan.name)
+        if isinstance(plan, BroadcastExchangeExec):
+            return GpuBroadcastExchangeExec(*children)
+        if isinstance(plan, BroadcastHashJoinExec):
+            return GpuBroadcastHashJoinExec(plan.left_key, plan.right_key,
+                                            plan.build_side, *children)
         plan.children = children
         return plan
 

@@ -100,3 +100,35 @@ def test_exchange_empty_partitions():
     mp.start_processes(_worker_empty_split, args=(29532, q), nprocs=WORLD,
                        join=True, start_method="spawn")
     assert q.empty(), q.get()
+
+
+def _worker_broadcast(rank, port, fail_q):
+    try:
+        import torch
+        import torch.distributed as dist
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                          RANK=str(rank), WORLD_SIZE=str(WORLD))
+        dist.init_process_group("gloo")
+        from spark_amd.exchange import broadcast_gather
+        # unequal slice sizes per rank
+        n = 5 + rank * 3
+        cols = {"k": torch.arange(n, dtype=torch.int64) + rank * 100,
+                "p": torch.arange(n, dtype=torch.float64) * (rank + 1)}
+        out = broadcast_gather(cols)
+        exp_k = torch.cat([torch.arange(5 + r * 3, dtype=torch.int64) + r * 100
+                           for r in range(WORLD)])
+        exp_p = torch.cat([torch.arange(5 + r * 3, dtype=torch.float64) * (r + 1)
+                           for r in range(WORLD)])
+        assert (out["k"] == exp_k).all() and (out["p"] == exp_p).all()
+        dist.destroy_process_group()
+    except Exception as e:
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}")
+        raise
+
+
+def test_broadcast_gather_gloo_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.start_processes(_worker_broadcast, args=(29533, q), nprocs=WORLD,
+                       join=True, start_method="spawn")
+    assert q.empty(), q.get()

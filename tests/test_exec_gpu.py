@@ -169,3 +169,19 @@ def test_q1_shape_multi_agg_plan():
     np.testing.assert_allclose(got["sum(price)"][g], osp[o], rtol=1e-6)
     np.testing.assert_allclose(got["avg(qty)"][g], osq[o] / ocnt[o], rtol=1e-6)
     assert (got["count(price)"][g] == ocnt[o]).all()
+
+
+def test_broadcast_hash_join_single_rank():
+    # single rank: broadcast exchange is identity; join kernels identical
+    bn, pn = 10_000, 80_000
+    bkeys = oracle.gen_i64(seed=60, n=bn, range_=8_000)
+    pkeys = oracle.gen_i64(seed=61, n=pn, range_=8_000)
+    left = gx.InputBatches([dev_batch(lk=pkeys)])
+    right = gx.BroadcastExchangeExec(gx.InputBatches([dev_batch(rk=bkeys)]))
+    got = run_plan(gx.BroadcastHashJoinExec("lk", "rk", "right", left, right))
+    op, ob = oracle.join_inner(bkeys, pkeys)
+    assert len(got["lk"]) == len(op)
+    g = np.lexsort((got["rk"], got["lk"]))
+    o = np.lexsort((bkeys[ob], pkeys[op]))
+    assert (got["lk"][g] == pkeys[op][o]).all()
+    assert (got["rk"][g] == bkeys[ob][o]).all()

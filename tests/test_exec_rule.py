@@ -71,3 +71,16 @@ def test_sort_merge_join_replaced_by_gpu_hash_join():
     plan = gx.GpuColumnarRule().pre_columnar_transitions(smj)
     assert isinstance(plan, gx.GpuShuffledHashJoinExec)
     assert plan.left_key == "a" and plan.right_key == "b"
+
+
+def test_broadcast_join_rule():
+    scan = gx.InputBatches.__new__(gx.InputBatches)
+    gx.SparkPlan.__init__(scan)
+    scan._batches = []
+    bx = gx.BroadcastExchangeExec(scan)
+    bhj = gx.BroadcastHashJoinExec("a", "b", "right", scan, bx)
+    plan = gx.GpuColumnarRule().pre_columnar_transitions(bhj)
+    assert isinstance(plan, gx.GpuBroadcastHashJoinExec)
+    assert isinstance(plan.children[1], gx.GpuBroadcastExchangeExec)
+    dists = plan.required_child_distribution()
+    assert dists[1].kind == "broadcast" and dists[0].kind == "unspecified"
