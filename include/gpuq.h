@@ -180,7 +180,7 @@ int64_t gpuq_partition_workspace_bytes(int64_t nrows, int32_t num_parts);
 /* Computes the stable permutation that groups rows by partition id and the
  * per-partition row counts. out_perm[nrows] (uint32), out_counts[num_parts]
  * (int64, device). Gather columns with gpuq_gather afterwards.
- * num_parts <= 65536. */
+ * num_parts <= 65536 (two radix passes above 256). */
 int gpuq_partition_perm(void* stream, int64_t nrows, gpuq_col key,
                         int32_t num_parts, uint32_t* out_perm,
                         int64_t* out_counts,

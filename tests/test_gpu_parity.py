@@ -151,7 +151,7 @@ def test_agg_overflow_detected(gq):
 
 # ---------- partition ----------
 
-@pytest.mark.parametrize("nparts", [1, 8, 200, 256])
+@pytest.mark.parametrize("nparts", [1, 8, 200, 256, 2000, 65536])
 def test_partition_parity(gq, nparts):
     n = 500_000
     keys = oracle.gen_i64(seed=77, n=n)
