@@ -938,7 +938,9 @@ __global__ void k_partition_pids(int64_t n, const int64_t* keys, const uint8_t* 
                                  int32_t nparts, uint64_t* pid_as_key, uint32_t* idx,
                                  unsigned long long* counts /* [nparts] */) {
   __shared__ uint32_t h[256];
-  for (int b = threadIdx.x; b < nparts; b += blockDim.x) h[b] = 0;
+  bool lds_counts = nparts <= 256;
+  if (lds_counts)
+    for (int b = threadIdx.x; b < nparts; b += blockDim.x) h[b] = 0;
   __syncthreads();
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -948,11 +950,13 @@ __global__ void k_partition_pids(int64_t n, const int64_t* keys, const uint8_t* 
     int pid = spark_pmod(hsh, nparts);
     pid_as_key[i] = (uint64_t)pid;
     idx[i] = (uint32_t)i;
-    atomicAdd(&h[pid], 1u);
+    if (lds_counts) atomicAdd(&h[pid], 1u);
+    else atomicAdd(&counts[pid], 1ull);
   }
   __syncthreads();
-  for (int b = threadIdx.x; b < nparts; b += blockDim.x)
-    if (h[b]) atomicAdd(&counts[b], (unsigned long long)h[b]);
+  if (lds_counts)
+    for (int b = threadIdx.x; b < nparts; b += blockDim.x)
+      if (h[b]) atomicAdd(&counts[b], (unsigned long long)h[b]);
 }
 
 extern "C" int64_t gpuq_partition_workspace_bytes(int64_t n, int32_t nparts) {
@@ -968,8 +972,8 @@ extern "C" int gpuq_partition_perm(void* stream, int64_t n, gpuq_col key,
                                    void* workspace, int64_t workspace_bytes) {
   hipStream_t s = (hipStream_t)stream;
   if (n > 0xFFFFFFFFLL) FAIL(GPUQ_ERR_INVALID, "partition: nrows %lld > 2^32", (long long)n);
-  if (nparts < 1 || nparts > 256)
-    FAIL(GPUQ_ERR_INVALID, "partition: num_parts %d not in [1,256] (round 1)", nparts);
+  if (nparts < 1 || nparts > 65536)
+    FAIL(GPUQ_ERR_INVALID, "partition: num_parts %d not in [1,65536]", nparts);
   if (key.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "partition: key must be int64");
   sort_ws w; int64_t need;
   sort_ws_layout(n, 256, &w, (char*)workspace, &need);
@@ -986,18 +990,26 @@ extern "C" int gpuq_partition_perm(void* stream, int64_t n, gpuq_col key,
   scatter_geom geom = get_sort_geom();
   int tile = geom.block * geom.items;
   int64_t nb = sort_nblocks(n, tile);
-  { hipEvent_t _pe = prof_begin(s);
-    k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, w.ka, 0, w.hist, (int)nb, tile);
+  /* stable LSB radix over the pid: one 8-bit pass, two when nparts > 256 */
+  int passes = nparts > 256 ? 2 : 1;
+  uint64_t *kin = w.ka, *kout = w.kb;
+  uint32_t *iin = w.ia, *iout = w.ib;
+  for (int p = 0; p < passes; p++) {
+    { hipEvent_t _pe = prof_begin(s);
+    k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, p * 8, w.hist, (int)nb, tile);
     prof_end("radix_hist", s, _pe); }
-  HIP_TRY(hipGetLastError());
-  int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
-  if (rc) return rc;
-  { hipEvent_t _pe = prof_begin(s);
-    launch_scatter<0, false>(s, geom, nb, n, w.ka, w.ia, w.kb, w.ib, w.hist_scan,
-                             0, 0, nullptr, nullptr, nullptr, 0);
+    HIP_TRY(hipGetLastError());
+    int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
+    if (rc) return rc;
+    uint32_t* iout_pass = (p == passes - 1) ? out_perm : iout;
+    { hipEvent_t _pe = prof_begin(s);
+    launch_scatter<0, false>(s, geom, nb, n, kin, iin, kout, iout_pass, w.hist_scan,
+                             p * 8, 0, nullptr, nullptr, nullptr, 0);
     prof_end("radix_scatter", s, _pe); }
-  HIP_TRY(hipGetLastError());
-  HIP_TRY(hipMemcpyAsync(out_perm, w.ib, n * 4, hipMemcpyDeviceToDevice, s));
+    HIP_TRY(hipGetLastError());
+    uint64_t* tk = kin; kin = kout; kout = tk;
+    uint32_t* ti = iin; iin = iout_pass; iout = ti;
+  }
   return GPUQ_OK;
 }
 
