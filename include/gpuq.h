@@ -186,6 +186,18 @@ int gpuq_partition_perm(void* stream, int64_t nrows, gpuq_col key,
                         int64_t* out_counts,
                         void* workspace, int64_t workspace_bytes);
 
+/* Range partition (global ORDER BY across GPUs): bin = first bound >=
+ * key in the sort order (RangePartitioning, ShuffleExchangeExec.scala:
+ * 379-400 — the host layer samples keys and picks quantile bounds);
+ * NULL keys go to the first/last partition per nulls_first. bounds:
+ * device array of nbounds entries, same dtype as the key; out_counts:
+ * nbounds+1 entries. Stable, like the hash partition. */
+int gpuq_range_partition_perm(void* stream, int64_t nrows, gpuq_col key,
+                              int32_t desc, int32_t nulls_first,
+                              const void* bounds, int32_t nbounds,
+                              uint32_t* out_perm, int64_t* out_counts,
+                              void* workspace, int64_t workspace_bytes);
+
 /* ---------------------------------------------------------------- */
 /* HASH JOIN — replaces ShuffledHashJoinExec inner join              */
 /* (joins/ShuffledHashJoinExec.scala:103-132; build side analog of   */

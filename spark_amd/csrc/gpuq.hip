@@ -1013,6 +1013,92 @@ extern "C" int gpuq_partition_perm(void* stream, int64_t n, gpuq_col key,
   return GPUQ_OK;
 }
 
+/* ---- range partition: bin = first bound >= key (RangePartitioning,
+ * exchange/ShuffleExchangeExec.scala:379-400: sampled bounds, keys <= 
+ * bounds[j] stay in partition j; partition order follows the sort order,
+ * so a per-rank sort after the exchange yields a globally ordered
+ * rank-major result). Keys and bounds are radix-encoded with the SAME
+ * encoding as the sort (desc = complement), so ascending logic covers
+ * both directions. NULL keys go to the first (nulls_first) or last
+ * partition, matching SortOrder null placement. nbounds <= 2047. */
+__global__ void k_range_pids(int64_t n, const void* keys, const uint8_t* validity,
+                             int dtype, int desc, int nulls_first,
+                             const void* bounds, int nbounds,
+                             uint64_t* pid_as_key, uint32_t* idx,
+                             unsigned long long* counts /* [nbounds+1] */) {
+  extern __shared__ __attribute__((aligned(16))) unsigned long long eb[];
+  for (int j = threadIdx.x; j < nbounds; j += blockDim.x) {
+    uint64_t e = (dtype == GPUQ_FLOAT64) ? encode_f64(((const double*)bounds)[j])
+                                         : encode_i64(((const int64_t*)bounds)[j]);
+    eb[j] = desc ? ~e : e;
+  }
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int pid;
+    if (!bit_valid(validity, i)) {
+      pid = nulls_first ? 0 : nbounds;
+    } else {
+      uint64_t e = (dtype == GPUQ_FLOAT64) ? encode_f64(((const double*)keys)[i])
+                                           : encode_i64(((const int64_t*)keys)[i]);
+      if (desc) e = ~e;
+      int lo = 0, hi = nbounds;          /* first j with e <= eb[j] */
+      while (lo < hi) {
+        int mid = (lo + hi) >> 1;
+        if (e <= eb[mid]) hi = mid; else lo = mid + 1;
+      }
+      pid = lo;
+    }
+    pid_as_key[i] = (uint64_t)pid;
+    idx[i] = (uint32_t)i;
+    atomicAdd(&counts[pid], 1ull);
+  }
+}
+
+extern "C" int gpuq_range_partition_perm(void* stream, int64_t n, gpuq_col key,
+                                         int32_t desc, int32_t nulls_first,
+                                         const void* bounds, int32_t nbounds,
+                                         uint32_t* out_perm, int64_t* out_counts,
+                                         void* workspace, int64_t workspace_bytes) {
+  hipStream_t s = (hipStream_t)stream;
+  if (n > 0xFFFFFFFFLL) FAIL(GPUQ_ERR_INVALID, "range_partition: nrows %lld > 2^32", (long long)n);
+  if (nbounds < 0 || nbounds > 2047)
+    FAIL(GPUQ_ERR_INVALID, "range_partition: nbounds %d not in [0,2047]", nbounds);
+  if (key.dtype != GPUQ_INT64 && key.dtype != GPUQ_FLOAT64)
+    FAIL(GPUQ_ERR_INVALID, "range_partition: unsupported dtype %d", key.dtype);
+  sort_ws w; int64_t need;
+  sort_ws_layout(n, 256, &w, (char*)workspace, &need);
+  if (workspace_bytes < need)
+    FAIL(GPUQ_ERR_INVALID, "range_partition: workspace %lld < %lld",
+         (long long)workspace_bytes, (long long)need);
+  HIP_TRY(hipMemsetAsync(out_counts, 0, (size_t)(nbounds + 1) * 8, s));
+  if (n == 0) return GPUQ_OK;
+  k_range_pids<<<grid1d(n), 256, (uint32_t)(nbounds * 8), s>>>(
+      n, key.data, key.validity, key.dtype, desc, nulls_first, bounds, nbounds,
+      w.ka, w.ia, (unsigned long long*)out_counts);
+  HIP_TRY(hipGetLastError());
+  scatter_geom geom = get_sort_geom();
+  int tile = geom.block * geom.items;
+  int64_t nb = sort_nblocks(n, tile);
+  int passes = (nbounds + 1) > 256 ? 2 : 1;
+  uint64_t *kin = w.ka, *kout = w.kb;
+  uint32_t *iin = w.ia, *iout = w.ib;
+  for (int p = 0; p < passes; p++) {
+    k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, p * 8, w.hist, (int)nb, tile);
+    HIP_TRY(hipGetLastError());
+    int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
+    if (rc) return rc;
+    uint32_t* iout_pass = (p == passes - 1) ? out_perm : iout;
+    launch_scatter<0, false>(s, geom, nb, n, kin, iin, kout, iout_pass, w.hist_scan,
+                             p * 8, 0, nullptr, nullptr, nullptr, 0);
+    HIP_TRY(hipGetLastError());
+    uint64_t* tk = kin; kin = kout; kout = tk;
+    uint32_t* ti = iin; iin = iout_pass; iout = ti;
+  }
+  return GPUQ_OK;
+}
+
 /* ================= hash aggregate ================= */
 /*
  * Open-address table, EMPTY key sentinel = -1 (memset 0xFF); rows whose key

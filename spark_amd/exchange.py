@@ -88,3 +88,31 @@ def broadcast_gather(cols: Dict[str, torch.Tensor], group=None) -> Dict[str, tor
         dist.all_gather(parts, padded, group=group)
         out[name] = torch.cat([p[:c] for p, c in zip(parts, counts)])
     return out
+
+
+def range_exchange(key: torch.Tensor, payload: Dict[str, torch.Tensor],
+                   desc=False, samples_per_rank: int = 4096, group=None):
+    """Global ORDER BY exchange (the RangePartitioning path,
+    ShuffleExchangeExec.scala:379-400): sample keys, all-gather the samples,
+    take quantile bounds, range-partition on device, all-to-all. After a
+    local sort on each rank, rank-major order is the global sort order.
+    Returns (key, payload) holding this rank's range."""
+    from . import gpuq
+    world = dist.get_world_size(group)
+    n = key.numel()
+    # evenly-strided local sample, sorted on device with the GPU sort
+    stride = max(1, n // samples_per_rank)
+    idx = torch.arange(0, n, stride, dtype=torch.int32, device=key.device)
+    local_sample = gpuq.gather(key, idx)
+    gathered = broadcast_gather({"s": local_sample}, group=group)["s"]
+    _, sorted_samples = gpuq.sort_perm(gathered, desc=desc)
+    m = sorted_samples.numel()
+    bidx = torch.arange(1, world, dtype=torch.int32, device=key.device) * (m // world)
+    bounds = gpuq.gather(sorted_samples, bidx.clamp(max=m - 1))
+    perm, counts = gpuq.range_partition_perm(key, bounds, desc=desc)
+    cols = {"__key__": gpuq.gather(key, perm)}
+    for name, t in payload.items():
+        cols[name] = gpuq.gather(t, perm)
+    out, _ = exchange_columns(cols, counts.cpu().tolist(), group=group)
+    k = out.pop("__key__")
+    return k, out

@@ -76,6 +76,9 @@ def lib() -> ctypes.CDLL:
         L.gpuq_partition_workspace_bytes.argtypes = [i64, i32]
         L.gpuq_partition_perm.restype = i32
         L.gpuq_partition_perm.argtypes = [vp, i64, _Col, i32, vp, vp, vp, i64]
+        L.gpuq_range_partition_perm.restype = i32
+        L.gpuq_range_partition_perm.argtypes = [vp, i64, _Col, i32, i32, vp, i32,
+                                                vp, vp, vp, i64]
         L.gpuq_join_build_workspace_bytes.restype = i64
         L.gpuq_join_build_workspace_bytes.argtypes = [i64, i64]
         L.gpuq_join_probe_workspace_bytes.restype = i64
@@ -366,3 +369,23 @@ def hash_agg_multi(keys: torch.Tensor, specs, capacity: int, key_validity=None,
         outp_arr, ctypes.byref(ng)))
     gn = ng.value
     return ok[:gn], okv[:gn], [t[:gn] for t in outs]
+
+
+def range_partition_perm(keys: torch.Tensor, bounds: torch.Tensor, desc=False,
+                         nulls_first=None, workspace=None, key_validity=None):
+    """Stable range partition (RangePartitioning analog): returns
+    (perm, counts[len(bounds)+1])."""
+    n = keys.numel()
+    dev = keys.device
+    if nulls_first is None:
+        nulls_first = not desc
+    nparts = bounds.numel() + 1
+    if workspace is None:
+        workspace = partition_workspace(n, nparts, dev)
+    perm = torch.empty(n, dtype=torch.int32, device=dev)
+    counts = torch.empty(nparts, dtype=torch.int64, device=dev)
+    _check(lib().gpuq_range_partition_perm(
+        _stream(), n, _col(keys, key_validity), int(desc), int(nulls_first),
+        bounds.data_ptr(), bounds.numel(), perm.data_ptr(), counts.data_ptr(),
+        workspace.data_ptr(), workspace.numel()))
+    return perm, counts

@@ -132,3 +132,55 @@ def test_broadcast_gather_gloo_world2():
     mp.start_processes(_worker_broadcast, args=(29533, q), nprocs=WORLD,
                        join=True, start_method="spawn")
     assert q.empty(), q.get()
+
+
+def _worker_range(rank, port, fail_q):
+    try:
+        import numpy as np
+        import torch
+        import torch.distributed as dist
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                          RANK=str(rank), WORLD_SIZE=str(WORLD))
+        dist.init_process_group("gloo")
+        import oracle
+        from spark_amd.exchange import exchange_columns, broadcast_gather
+        # host-logic shape of range_exchange with oracle/numpy as the test
+        # driver (the device kernels are covered by -m gpu parity tests):
+        n = 20_000
+        keys = oracle.gen_i64(seed=300 + rank, n=n)
+        sample = keys[::max(1, n // 1024)]
+        gathered = broadcast_gather({"s": torch.from_numpy(sample.copy())})["s"].numpy()
+        ss = np.sort(gathered)
+        m = len(ss)
+        bounds = ss[[(j + 1) * (m // WORLD) for j in range(WORLD - 1)]]
+        pids = np.searchsorted(bounds, keys, side="left")
+        perm = np.argsort(pids, kind="stable")
+        counts = np.bincount(pids, minlength=WORLD).tolist()
+        out, _ = exchange_columns({"k": torch.from_numpy(keys[perm])}, counts)
+        got = np.sort(out["k"].numpy())
+        # ranges are disjoint and ordered across ranks
+        lo = -2**63 if rank == 0 else int(bounds[rank - 1])
+        hi = 2**63 - 1 if rank == WORLD - 1 else int(bounds[rank])
+        assert (got > lo).all() if rank > 0 else True
+        assert (got <= hi).all()
+        # global order: concatenating rank 0's sorted range then rank 1's is sorted
+        mx = torch.tensor([int(got.max()) if len(got) else -2**63])
+        mn = torch.tensor([int(got.min()) if len(got) else 2**63 - 1])
+        maxs = [torch.zeros(1, dtype=torch.int64) for _ in range(WORLD)]
+        mins = [torch.zeros(1, dtype=torch.int64) for _ in range(WORLD)]
+        dist.all_gather(maxs, mx)
+        dist.all_gather(mins, mn)
+        for r in range(WORLD - 1):
+            assert int(maxs[r]) <= int(mins[r + 1])
+        dist.destroy_process_group()
+    except Exception as e:
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}")
+        raise
+
+
+def test_range_exchange_logic_gloo_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.start_processes(_worker_range, args=(29534, q), nprocs=WORLD,
+                       join=True, start_method="spawn")
+    assert q.empty(), q.get()

@@ -319,3 +319,29 @@ def test_multi_aggregate_large_table(gq):
     assert (ok.cpu().numpy()[g] == oka[oa]).all()
     np.testing.assert_allclose(accs[0].cpu().numpy()[g], osuma[oa], rtol=1e-6)
     assert (accs[1].cpu().numpy()[g] == ocnt[oa]).all()
+
+
+@pytest.mark.parametrize("desc", [False, True])
+@pytest.mark.parametrize("dtype", ["i64", "f64"])
+def test_range_partition_parity(gq, desc, dtype):
+    n = 300_000
+    if dtype == "i64":
+        keys = oracle.gen_i64(seed=70, n=n)
+        bounds = np.sort(oracle.gen_i64(seed=71, n=7))
+    else:
+        keys = oracle.gen_f64_unit(seed=70, n=n)
+        bounds = np.sort(oracle.gen_f64_unit(seed=71, n=7))
+    if desc:
+        bounds = bounds[::-1].copy()  # bounds follow the sort order
+    perm, counts = gq.range_partition_perm(to_dev(keys), to_dev(bounds), desc=desc)
+    # expected pid = first bound >= key in the sort order:
+    #   asc:  #{j : bounds[j] < key}   (bounds ascending)
+    #   desc: #{j : bounds[j] > key}   (bounds descending)
+    if desc:
+        pids = (bounds[None, :] > keys[:, None]).sum(axis=1)
+    else:
+        pids = np.searchsorted(bounds, keys, side="left")
+    exp_counts = np.bincount(pids, minlength=len(bounds) + 1)
+    assert (counts.cpu().numpy() == exp_counts).all()
+    exp_perm = np.argsort(pids, kind="stable")
+    assert (perm.cpu().numpy().astype(np.uint32) == exp_perm.astype(np.uint32)).all()
