@@ -267,9 +267,24 @@ class GpuSortExec(SparkPlan):
         return [Distribution("unspecified")]
 
     def execute_columnar(self):
+        import torch.distributed as dist
         from . import gpuq
         o = self.sort_order
         for batch in self.children[0].execute_columnar():
+            if (self.global_sort and dist.is_initialized()
+                    and dist.get_world_size() > 1):
+                # global ORDER BY: range exchange first (the EnsureRequirements
+                # RangePartitioning insertion, exchange/EnsureRequirements.scala:296);
+                # after the local sort below, rank-major order is global order
+                from .exchange import range_exchange
+                payload = {n_: t for n_, t in batch.columns().items()
+                           if n_ != o.key}
+                k, payload = range_exchange(batch.column(o.key), payload,
+                                            desc=o.descending)
+                cols = {o.key: k}
+                cols.update(payload)
+                batch.close()
+                batch = ColumnarBatch(cols)
             keys = batch.column(o.key)
             perm, skeys = gpuq.sort_perm(keys, desc=o.descending,
                                          nulls_first=o.nulls_first,
