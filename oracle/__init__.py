@@ -65,6 +65,10 @@ def lib() -> ctypes.CDLL:
         L.oracle_join_inner_i64.argtypes = [ctypes.c_void_p, ctypes.c_void_p, i64,
                                             ctypes.c_void_p, ctypes.c_void_p, i64,
                                             ctypes.c_void_p, ctypes.c_void_p, i64]
+        L.oracle_sort_perm_i64_mt.restype = None
+        L.oracle_sort_perm_i64_mt.argtypes = [ctypes.c_void_p, i64, ctypes.c_void_p, i32]
+        L.oracle_gather_i64_mt.restype = None
+        L.oracle_gather_i64_mt.argtypes = [ctypes.c_void_p, ctypes.c_void_p, i64, ctypes.c_void_p]
         L.xorshift_state_init.restype = i64
         L.xorshift_state_init.argtypes = [i64]
         L.xorshift_state_next_long.restype = i64
@@ -211,4 +215,21 @@ def gen_i64(seed: int, n: int, range_: int = 0, start: int = 0) -> np.ndarray:
 def gen_f64_unit(seed: int, n: int, start: int = 0) -> np.ndarray:
     out = np.empty(n, dtype=np.float64)
     lib().gen_fill_f64_unit(seed, start, n, _ptr(out))
+    return out
+
+
+def sort_perm_mt(keys: np.ndarray, nthreads: int = 0) -> np.ndarray:
+    """OpenMP-parallel sort permutation (the multithreaded CPU-baseline leg;
+    int64 asc non-null — same semantics as sort_perm)."""
+    keys = np.ascontiguousarray(keys, dtype=np.int64)
+    out = np.empty(len(keys), dtype=np.int64)
+    lib().oracle_sort_perm_i64_mt(_ptr(keys), len(keys), _ptr(out), nthreads)
+    return out
+
+
+def gather_i64_mt(arr: np.ndarray, perm: np.ndarray) -> np.ndarray:
+    out = np.empty(len(perm), dtype=np.int64)
+    lib().oracle_gather_i64_mt(_ptr(np.ascontiguousarray(arr, dtype=np.int64)),
+                               _ptr(np.ascontiguousarray(perm, dtype=np.int64)),
+                               len(perm), _ptr(out))
     return out

@@ -170,3 +170,11 @@ def test_xorshift_restatement_lock():
         fix["seed123_masked_ff_first32"]
     r = oracle.XorShiftRandom(1)
     assert [r.next_int(5) for _ in range(16)] == fix["seed1_nextInt5_first16"]
+
+
+def test_mt_sort_matches_single_thread():
+    # the OpenMP baseline leg must produce the identical stable permutation
+    keys = oracle.gen_i64(9, 500_000)
+    assert (oracle.sort_perm_mt(keys, 4) == oracle.sort_perm(keys)).all()
+    low = oracle.gen_i64(10, 100_000, range_=50)  # heavy ties: stability
+    assert (oracle.sort_perm_mt(low, 8) == oracle.sort_perm(low)).all()
