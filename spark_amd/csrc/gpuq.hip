@@ -467,7 +467,10 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
                      int shift, int nblocks, int nparts,
                      unsigned long long* state /* [nblocks][256] */,
                      const uint32_t* gbase /* [256] pass bin bases */,
-                     unsigned long long* err_flag, uint32_t epoch) {
+                     unsigned long long* err_flag, uint32_t epoch,
+                     void* decode_out /* non-null on the final pass: write
+                       decoded keys here instead of encoded keys to kout */,
+                     int decode_mode /* 1=i64 2=f64, +4 = desc */) {
   constexpr int WAVES = BLOCK / WAVE;
   constexpr int TILE = BLOCK * ITEMS;
   __shared__ uint32_t wave_hist[WAVES][256];
@@ -636,7 +639,19 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
       uint64_t kk = stage_k[j];
       int bin = compute_bin<BIN_MODE>(kk, shift, nparts);
       uint32_t dst = bin_gbase[bin] + (uint32_t)j;
-      kout[dst] = kk;
+      if (decode_mode) {
+        /* final pass: emit the DECODED key column directly (the separate
+           decode kernel and this pass's encoded-key write both disappear) */
+        uint64_t e = (decode_mode & 4) ? ~kk : kk;
+        if (decode_mode & 2) {
+          uint64_t mask = ((e >> 63) ? 0 : ~0ULL) | SIGNBIT;
+          ((uint64_t*)decode_out)[dst] = e ^ mask;
+        } else {
+          ((int64_t*)decode_out)[dst] = (int64_t)(e ^ SIGNBIT);
+        }
+      } else {
+        kout[dst] = kk;
+      }
       iout[dst] = stage_i[j];
     }
   }
@@ -660,11 +675,12 @@ static void launch_scatter(hipStream_t s, scatter_geom g, int64_t nb,
                            uint64_t* kout, uint32_t* iout,
                            const uint32_t* scanned, int shift, int nparts,
                            unsigned long long* state, const uint32_t* gbase,
-                           unsigned long long* err_flag, uint32_t epoch) {
+                           unsigned long long* err_flag, uint32_t epoch,
+                           void* decode_out = nullptr, int decode_mode = 0) {
   dim3 grid((uint32_t)nb);
 #define LS(B, I) k_radix_scatter<BIN_MODE, B, I, LOOKBACK><<<grid, B, 0, s>>>( \
       n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase, \
-      err_flag, epoch)
+      err_flag, epoch, decode_out, decode_mode)
   if (g.block == 256 && g.items == 16) LS(256, 16);
   else if (g.block == 512 && g.items == 8) LS(512, 8);
   else if (g.block == 512 && g.items == 16) LS(512, 16);
@@ -855,11 +871,19 @@ retry:
     if (((bits_changed >> (byte * 8)) & 0xff) == 0) continue;  /* RadixSort.java:126 skip */
     int shift = byte * 8;
     uint32_t* iout_pass = (byte == last_byte) ? out_perm + sorted_at : iout;
+    /* fuse the key decode into the final pass when the caller wants keys
+     * and the null path is not rerouting them through a gather */
+    void* dec_out = nullptr;
+    int dec_mode = 0;
+    if (byte == last_byte && out_keys && !key.validity) {
+      dec_out = (char*)out_keys;  /* sorted_at == 0 without validity */
+      dec_mode = (key.dtype == GPUQ_FLOAT64 ? 2 : 1) | (desc ? 4 : 0);
+    }
     if (onesweep) {
       { hipEvent_t _pe = prof_begin(s);
       launch_scatter<0, true>(s, geom, nb, n_sort, kin, iin, kout, iout_pass, nullptr,
                               shift, 0, w.state, w.gbase + byte * 256, w.err,
-                              (uint32_t)(byte + 1));
+                              (uint32_t)(byte + 1), dec_out, dec_mode);
       prof_end("radix_scatter", s, _pe); }
       HIP_TRY(hipGetLastError());
     } else {
@@ -871,7 +895,7 @@ retry:
       if (rc) return rc;
       { hipEvent_t _pe = prof_begin(s);
       launch_scatter<0, false>(s, geom, nb, n_sort, kin, iin, kout, iout_pass, w.hist_scan,
-                               shift, 0, nullptr, nullptr, nullptr, 0);
+                               shift, 0, nullptr, nullptr, nullptr, 0, dec_out, dec_mode);
       prof_end("radix_scatter", s, _pe); }
       HIP_TRY(hipGetLastError());
     }
@@ -912,6 +936,8 @@ retry:
       gpuq_col kc = key; kc.validity = nullptr;
       int rc = gpuq_gather(stream, n, kc, out_perm, out_keys);
       if (rc) return rc;
+    } else if (last_byte >= 0) {
+      /* keys were decoded by the fused final pass */
     } else {
       if (key.dtype == GPUQ_FLOAT64) {
         if (desc) k_decode<GPUQ_FLOAT64, true><<<grid1d(n), 256, 0, s>>>(n, kin, out_keys);
