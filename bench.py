@@ -89,10 +89,10 @@ class SortWorkload:
         gq = self.gq
         perm, skeys = gq.sort_perm(self.keys, workspace=self.ws,
                                    out_perm=self.perm, out_keys=self.skeys)
-        gq.lib().gpuq_gather(gq._stream(), self.rows, gq._col(self.pay1),
-                             perm.data_ptr(), self.out1.data_ptr())
-        gq.lib().gpuq_gather(gq._stream(), self.rows, gq._col(self.pay2),
-                             perm.data_ptr(), self.out2.data_ptr())
+        gq.lib().gpuq_gather2_i64(gq._stream(), self.rows,
+                                  self.pay1.data_ptr(), self.pay2.data_ptr(),
+                                  perm.data_ptr(), self.out1.data_ptr(),
+                                  self.out2.data_ptr())
 
     def free(self):
         del self.keys, self.pay1, self.pay2, self.ws, self.out1, self.out2, self.perm, self.skeys

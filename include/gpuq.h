@@ -110,6 +110,10 @@ int gpuq_sort_perm(void* stream, int64_t nrows, gpuq_col key,
 int gpuq_gather(void* stream, int64_t nrows, gpuq_col col,
                 const uint32_t* perm, void* out);
 
+/* two 8-byte columns through one permutation in one kernel */
+int gpuq_gather2_i64(void* stream, int64_t nrows, const void* a, const void* b,
+                     const uint32_t* perm, void* out_a, void* out_b);
+
 /* ---------------------------------------------------------------- */
 /* HASH AGGREGATE — replaces HashAggregateExec                       */
 /* (execution/aggregate/HashAggregateExec.scala:99-151) for          */

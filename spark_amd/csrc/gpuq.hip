@@ -223,9 +223,48 @@ extern "C" int gpuq_gen_f64_unit(void* stream, uint64_t seed, uint64_t start,
 
 template <typename T>
 __global__ void k_gather(int64_t n, const T* in, const uint32_t* perm, T* out) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  /* 4-way unrolled so four independent random loads are in flight per lane
+   * (random gathers are latency-bound; same lesson as the scatter preload) */
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i + 3 * stride < n; i += 4 * stride) {
+    uint32_t p0 = perm[i], p1 = perm[i + stride], p2 = perm[i + 2 * stride],
+             p3 = perm[i + 3 * stride];
+    T v0 = in[p0], v1 = in[p1], v2 = in[p2], v3 = in[p3];
+    out[i] = v0; out[i + stride] = v1;
+    out[i + 2 * stride] = v2; out[i + 3 * stride] = v3;
+  }
   for (; i < n; i += stride) out[i] = in[perm[i]];
+}
+
+/* two columns through one permutation (both payload gathers of a sort step
+ * in one kernel: one perm read, two independent random loads in flight) */
+__global__ void k_gather2(int64_t n, const uint64_t* a, const uint64_t* b,
+                          const uint32_t* perm, uint64_t* oa, uint64_t* ob) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i + stride < n; i += 2 * stride) {
+    uint32_t p0 = perm[i], p1 = perm[i + stride];
+    uint64_t a0 = a[p0], b0 = b[p0], a1 = a[p1], b1 = b[p1];
+    oa[i] = a0; ob[i] = b0;
+    oa[i + stride] = a1; ob[i + stride] = b1;
+  }
+  for (; i < n; i += stride) {
+    uint32_t p = perm[i];
+    oa[i] = a[p]; ob[i] = b[p];
+  }
+}
+
+extern "C" int gpuq_gather2_i64(void* stream, int64_t n, const void* a,
+                                const void* b, const uint32_t* perm,
+                                void* oa, void* ob) {
+  { hipEvent_t _pe = prof_begin((hipStream_t)stream);
+  k_gather2<<<grid1d(n), 256, 0, (hipStream_t)stream>>>(
+      n, (const uint64_t*)a, (const uint64_t*)b, perm,
+      (uint64_t*)oa, (uint64_t*)ob);
+  prof_end("gather2", (hipStream_t)stream, _pe); }
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
 }
 
 extern "C" int gpuq_gather(void* stream, int64_t n, gpuq_col col,

@@ -345,3 +345,20 @@ def test_range_partition_parity(gq, desc, dtype):
     assert (counts.cpu().numpy() == exp_counts).all()
     exp_perm = np.argsort(pids, kind="stable")
     assert (perm.cpu().numpy().astype(np.uint32) == exp_perm.astype(np.uint32)).all()
+
+
+def test_gather2_parity(gq):
+    import torch as _t
+    n = 200_000
+    a = oracle.gen_i64(seed=95, n=n)
+    b = oracle.gen_f64_unit(seed=96, n=n)
+    keys = oracle.gen_i64(seed=97, n=n, range_=999)
+    perm, _ = gq.sort_perm(to_dev(keys))
+    da, db = to_dev(a), to_dev(b.view(np.int64))
+    oa = _t.empty(n, dtype=_t.int64, device="cuda")
+    ob = _t.empty(n, dtype=_t.int64, device="cuda")
+    gq._check(gq.lib().gpuq_gather2_i64(gq._stream(), n, da.data_ptr(), db.data_ptr(),
+                                        perm.data_ptr(), oa.data_ptr(), ob.data_ptr()))
+    exp = oracle.sort_perm(keys)
+    assert (oa.cpu().numpy() == a[exp]).all()
+    assert (ob.cpu().numpy() == b.view(np.int64)[exp]).all()
