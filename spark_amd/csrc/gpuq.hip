@@ -703,7 +703,7 @@ static scatter_geom get_sort_geom(void) {
     const char* e = getenv("GPUQ_SORT_GEOM");
     int b = 0, it = 0;
     if (e && sscanf(e, "%dx%d", &b, &it) == 2) { g.block = b; g.items = it; }
-    else { g.block = 1024; g.items = 8; }
+    else { g.block = 512; g.items = 10; }
   }
   return g;
 }
