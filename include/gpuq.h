@@ -153,7 +153,9 @@ int gpuq_hash_agg_i64_f64(void* stream, int64_t nrows,
 /* Multi-aggregate: one pass computing up to 6 accumulators per group
  * (HashAggregateExec evaluates a list of aggregate expressions,
  * HashAggregateExec.scala:68-76 — e.g. TPC-H Q1's 8 aggregates).
- * spec_ops[j]: 0=SUM(float64 col), 1=COUNT(col), 2=COUNT(*);
+ * spec_ops[j]: 0=SUM(float64 col), 1=COUNT(col), 2=COUNT(*),
+ * 3=SUM(int64 col) -> int64 (Sum.scala LongType result; non-ansi
+ * overflow wraps);
  * spec_cols[j] indexes vals[] (ignored for COUNT(*)). out_accs[j] is a
  * device array per spec: f64 for SUM, i64 for COUNT. Small tables
  * (cap*(1+nspecs)*8 <= 64 KB) aggregate per-block in LDS first.

@@ -1925,6 +1925,10 @@ DEV void aggm_update(unsigned long long* acc, int nspecs, const agg_ops_spec ops
       atomicAdd((double*)&acc[j], cols.p[j][i]);
     } else if (op == 1) {
       if (bit_valid(av.v[j], i)) atomicAdd(&acc[j], 1ull);
+    } else if (op == 3) {
+      /* SUM(int64) -> int64 (Sum.scala resultType LongType; non-ansi
+       * overflow wraps = two's-complement u64 add, bit-exact) */
+      atomicAdd(&acc[j], (unsigned long long)((const int64_t*)cols.p[j])[i]);
     } else {
       atomicAdd(&acc[j], 1ull);
     }
@@ -1995,7 +1999,7 @@ void k_aggm_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
           atomicAdd((double*)&tab[(int64_t)stride * slot + 1 + j],
                     __longlong_as_double((long long)v));
         else
-          atomicAdd(&tab[(int64_t)stride * slot + 1 + j], v);
+          atomicAdd(&tab[(int64_t)stride * slot + 1 + j], v);  /* COUNT + SUM_I64 */
       }
     }
   }
@@ -2073,9 +2077,10 @@ extern "C" int gpuq_hash_agg_multi(void* stream, int64_t n, gpuq_col key,
   agg_cols cols = {}; agg_ops_spec ops = {}; agg_valid av = {};
   for (int j = 0; j < nspecs; j++) {
     ops.op[j] = spec_ops[j];
-    if (spec_ops[j] == 0) {
+    if (spec_ops[j] == 0 || spec_ops[j] == 3) {
       const gpuq_col& c = vals[spec_cols[j]];
-      if (c.dtype != GPUQ_FLOAT64) FAIL(GPUQ_ERR_INVALID, "aggm: SUM col must be float64");
+      int want = spec_ops[j] == 0 ? GPUQ_FLOAT64 : GPUQ_INT64;
+      if (c.dtype != want) FAIL(GPUQ_ERR_INVALID, "aggm: SUM col dtype mismatch");
       if (c.validity) FAIL(GPUQ_ERR_INVALID, "aggm: SUM cols must be non-null "
                            "(pair with a COUNT spec for NULL tracking)");
       cols.p[j] = (const double*)c.data;

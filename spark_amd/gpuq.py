@@ -342,9 +342,16 @@ def hash_agg_multi(keys: torch.Tensor, specs, capacity: int, key_validity=None,
     for s in specs:
         kind = s[0]
         if kind == "sum":
-            ops.append(0); colidx.append(len(cols)); cols.append(_col(s[1]))
-            outs.append(torch.empty(max_groups or capacity + 2,
-                                    dtype=torch.float64, device=dev))
+            t = s[1]
+            if t.dtype == torch.int64:
+                ops.append(3)  # SUM(int64) -> int64, wrapping (Sum.scala)
+                outs.append(torch.empty(max_groups or capacity + 2,
+                                        dtype=torch.int64, device=dev))
+            else:
+                ops.append(0)
+                outs.append(torch.empty(max_groups or capacity + 2,
+                                        dtype=torch.float64, device=dev))
+            colidx.append(len(cols)); cols.append(_col(t))
         elif kind == "count":
             ops.append(1); colidx.append(len(cols))
             cols.append(_col(s[1], s[2] if len(s) > 2 else None))

@@ -362,3 +362,20 @@ def test_gather2_parity(gq):
     exp = oracle.sort_perm(keys)
     assert (oa.cpu().numpy() == a[exp]).all()
     assert (ob.cpu().numpy() == b.view(np.int64)[exp]).all()
+
+
+def test_multi_aggregate_sum_int64(gq):
+    # SUM over an int64 column returns exact int64 (wrapping on overflow,
+    # Sum.scala resultType LongType)
+    n, ngroups = 200_000, 50
+    keys = oracle.gen_i64(seed=98, n=n, range_=ngroups)
+    vals = oracle.gen_i64(seed=99, n=n, range_=1_000_000)
+    ok, okv, accs = gq.hash_agg_multi(
+        to_dev(keys), [("sum", to_dev(vals)), ("count*", None)],
+        capacity=256, max_groups=64)
+    g = np.argsort(ok.cpu().numpy())
+    uk = np.unique(keys)
+    exp = np.zeros(ngroups, dtype=np.int64)
+    np.add.at(exp, keys, vals)
+    assert (ok.cpu().numpy()[g] == uk).all()
+    assert (accs[0].cpu().numpy()[g] == exp).all()  # bit-exact int64 sums
