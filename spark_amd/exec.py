@@ -371,8 +371,8 @@ class GpuHashAggregateExec(SparkPlan):
             slots = []  # (fn, col, acc indices)
             for fn, col in self.aggs:
                 t = batch.column(col)
-                if t.dtype == torch.int64:
-                    t = gpuq.cast_i64_f64(t)
+                if fn == "avg" and t.dtype == torch.int64:
+                    t = gpuq.cast_i64_f64(t)   # AVG divides f64 sums
                 if fn == "sum":
                     specs.append(("sum", t)); slots.append((fn, col, [len(specs) - 1]))
                 elif fn == "count":
