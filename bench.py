@@ -117,10 +117,20 @@ class AggWorkload:
         import ctypes
         gq = self.gq
         ng = ctypes.c_int64(0)
-        rc = gq.lib().gpuq_hash_agg_i64_f64(
-            gq._stream(), self.rows, gq._col(self.keys), gq._col(self.vals),
-            self.ws.data_ptr(), self.cap, 1, 1, 1,  # ops=SUM (config 3 shape)
-            *[t.data_ptr() for t in self.outs], ctypes.byref(ng))
+        if os.environ.get("GPUQ_PART_AGG"):
+            if not hasattr(self, "pws"):
+                self.pws = torch.empty(
+                    gq.lib().gpuq_hash_agg_part_workspace_bytes(self.rows, self.cap),
+                    dtype=torch.uint8, device="cuda")
+            rc = gq.lib().gpuq_hash_agg_partitioned(
+                gq._stream(), self.rows, gq._col(self.keys), gq._col(self.vals),
+                self.pws.data_ptr(), self.cap, 1,
+                *[t.data_ptr() for t in self.outs], ctypes.byref(ng))
+        else:
+            rc = gq.lib().gpuq_hash_agg_i64_f64(
+                gq._stream(), self.rows, gq._col(self.keys), gq._col(self.vals),
+                self.ws.data_ptr(), self.cap, 1, 1, 1,  # ops=SUM (config 3 shape)
+                *[t.data_ptr() for t in self.outs], ctypes.byref(ng))
         gq._check(rc)
         self.ngroups = ng.value
 

@@ -150,6 +150,21 @@ int gpuq_hash_agg_i64_f64(void* stream, int64_t nrows,
                           double* out_sums, uint8_t* out_sum_valid,
                           int64_t* out_counts, int64_t* out_ngroups);
 
+/* Partitioned aggregation: same results contract as gpuq_hash_agg_i64_f64
+ * for a single batch with NON-NULL values — rows are first ordered by a
+ * 16-bit Murmur bucket (two stable ranked-scatter passes over (key,val)
+ * pairs), then each chunk aggregates in an LDS table and merges once per
+ * distinct key; sidesteps the global-atomic throughput wall at mid/high
+ * cardinality. GPUQ_ERR_OVERFLOW with *out_ngroups = -1 => a chunk held
+ * more distinct keys than the LDS table; fall back to the direct path. */
+int64_t gpuq_hash_agg_part_workspace_bytes(int64_t nrows, int64_t capacity);
+int gpuq_hash_agg_partitioned(void* stream, int64_t nrows,
+                              gpuq_col key, gpuq_col val,
+                              void* workspace, int64_t capacity, int32_t ops,
+                              int64_t* out_keys, uint8_t* out_key_valid,
+                              double* out_sums, uint8_t* out_sum_valid,
+                              int64_t* out_counts, int64_t* out_ngroups);
+
 /* Multi-aggregate: one pass computing up to 6 accumulators per group
  * (HashAggregateExec evaluates a list of aggregate expressions,
  * HashAggregateExec.scala:68-76 — e.g. TPC-H Q1's 8 aggregates).

@@ -354,9 +354,13 @@ template <int BIN_MODE>
 DEV int compute_bin(uint64_t key, int shift, int nparts) {
   (void)nparts;
   if (BIN_MODE == 0) return (int)((key >> shift) & 0xff);
-  /* BIN_MODE 2: top 8 bits of Murmur3(key,42) — hash-order bucketing so a
-   * bucketed stream sweeps a hash-ordered table monotonically (join) */
-  return (int)(((uint32_t)mm3_hash_long((int64_t)key, 42)) >> 24);
+  if (BIN_MODE == 2)
+    /* top 8 bits of Murmur3(key,42) — hash-order bucketing so a bucketed
+     * stream sweeps a hash-ordered table monotonically (join) */
+    return (int)(((uint32_t)mm3_hash_long((int64_t)key, 42)) >> 24);
+  /* BIN_MODE 3: byte `shift/8` of the TOP-16 murmur bits (two stable
+   * passes order rows by a 16-bit hash bucket — partitioned aggregation) */
+  return (int)((((uint32_t)mm3_hash_long((int64_t)key, 42)) >> (16 + shift)) & 0xff);
 }
 
 /* validity split for sort-with-nulls: pair key = 0 for the group that
@@ -498,10 +502,10 @@ static int exclusive_scan_u32(hipStream_t s, int64_t n, const uint32_t* in,
 #define OSW_EPOCH(x) ((x) >> 56)
 #define OSW_SPIN_LIMIT (1u << 22)
 
-template <int BIN_MODE, int BLOCK, int ITEMS, bool LOOKBACK>
+template <int BIN_MODE, int BLOCK, int ITEMS, bool LOOKBACK, typename PayT = uint32_t>
 __global__ __launch_bounds__(BLOCK)
-void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
-                     uint64_t* kout, uint32_t* iout,
+void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
+                     uint64_t* kout, PayT* iout,
                      const uint32_t* scanned /* [256][nblocks] exclusive */,
                      int shift, int nblocks, int nparts,
                      unsigned long long* state /* [nblocks][256] */,
@@ -517,7 +521,7 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
   __shared__ uint32_t bin_gbase[256];     /* global dest minus local start    */
   __shared__ uint32_t wtot[WAVES <= 4 ? 4 : WAVES];
   __shared__ uint64_t stage_k[TILE];
-  __shared__ uint32_t stage_i[TILE];
+  __shared__ PayT stage_i[TILE];
 
   const int tid = threadIdx.x;
   const int wave = tid / WAVE, lane = tid & (WAVE - 1);
@@ -531,7 +535,7 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const uint32_t* iin,
   /* wave w owns the contiguous sub-tile [w*WAVE*ITEMS, ...): element order
    * within the block = (wave, round, lane) = linear tile order. */
   uint64_t k[ITEMS];
-  uint32_t id[ITEMS];
+  PayT id[ITEMS];
   uint16_t lrank[ITEMS];
   uint8_t lbin[ITEMS];
 
@@ -1858,7 +1862,235 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
   return GPUQ_OK;
 }
 
-/* ================= multi-aggregate ================= */
+/* ---- partitioned aggregation (mid/high cardinality) ----
+ * The direct table is atomic-throughput-bound (~22 G f64-adds/s at 10M
+ * groups, tools/diag_hash.py). Here rows are first ordered by a 16-bit
+ * Murmur bucket (two stable ranked-scatter passes carrying (key, val)
+ * pairs), so any contiguous chunk holds few DISTINCT keys; each block then
+ * aggregates its chunk in a small LDS table and merges once per distinct
+ * key into the global table (a group spans at most a few chunks). */
+
+#define PAGG_CHUNK 16384
+#define PAGG_LDS_SLOTS 2048
+
+template <int OPS, int SLOT>
+__global__ __launch_bounds__(256)
+void k_agg_part(int64_t n, const uint64_t* keys, const uint64_t* vals,
+                unsigned long long* tab, agg_special* sp, int64_t cap_mask) {
+  __shared__ unsigned long long lt[PAGG_LDS_SLOTS * 3];
+  for (int j = threadIdx.x; j < PAGG_LDS_SLOTS * 3; j += blockDim.x)
+    lt[j] = (j % 3 == 0) ? AGG_EMPTY : 0;
+  __syncthreads();
+  const int64_t base = (int64_t)blockIdx.x * PAGG_CHUNK;
+  bool overflow = false;
+  for (int r = 0; r < PAGG_CHUNK / 256 && !overflow; r++) {
+    int64_t i = base + r * 256 + threadIdx.x;
+    if (i >= n) break;
+    if (keys[i] == AGG_EMPTY) continue;   /* special rows, handled upstream */
+    int64_t k = (int64_t)keys[i];
+    double v = __longlong_as_double((long long)vals[i]);
+    uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (PAGG_LDS_SLOTS - 1);
+    for (int probes = 0;; probes++) {
+      unsigned long long cur = lt[3 * slot];
+      if (cur == (unsigned long long)k) break;
+      if (cur == AGG_EMPTY) {
+        unsigned long long prev = atomicCAS(&lt[3 * slot], AGG_EMPTY,
+                                            (unsigned long long)k);
+        if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
+      }
+      slot = (slot + 1) & (PAGG_LDS_SLOTS - 1);
+      if (probes >= PAGG_LDS_SLOTS) { overflow = true; break; }
+    }
+    if (overflow) { atomicMax(&sp->overflow, 2ull); break; }
+    if (OPS & AGG_OP_SUM) atomicAdd((double*)&lt[3 * slot + 1], v);
+    if (OPS & AGG_OP_COUNT) atomicAdd(&lt[3 * slot + 2], 1ull);
+  }
+  __syncthreads();
+  /* merge occupied LDS slots into the global table */
+  for (int j = threadIdx.x; j < PAGG_LDS_SLOTS; j += blockDim.x) {
+    unsigned long long k = lt[3 * j];
+    if (k == AGG_EMPTY) continue;
+    uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)k, 42)) & (uint64_t)cap_mask;
+    for (int probes = 0;; probes++) {
+      unsigned long long cur = __hip_atomic_load(&tab[SLOT * slot], __ATOMIC_RELAXED,
+                                                 __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == k) break;
+      if (cur == AGG_EMPTY) {
+        unsigned long long prev = atomicCAS(&tab[SLOT * slot], AGG_EMPTY, k);
+        if (prev == AGG_EMPTY || prev == k) break;
+      }
+      slot = (slot + 1) & (uint64_t)cap_mask;
+      if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
+    }
+    if (OPS & AGG_OP_SUM)
+      atomicAdd((double*)&tab[SLOT * slot + 1],
+                __longlong_as_double((long long)lt[3 * j + 1]));
+    if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], lt[3 * j + 2]);
+  }
+}
+
+/* pack (key, f64-val-bits) pairs for the bucket ordering, position-aligned
+ * with the input. NULL-key and real -1-key rows accumulate into the special
+ * slots here and become EMPTY pairs (skipped by the chunk aggregator).
+ * Requires non-null values (the partitioned path is gated on that). */
+template <int OPS>
+__global__ void k_agg_part_pairs(int64_t n, const int64_t* keys, const uint8_t* kvalid,
+                                 const double* vals,
+                                 uint64_t* out_k, uint64_t* out_v, agg_special* sp) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    bool kv = bit_valid(kvalid, i);
+    int64_t k = kv ? keys[i] : 0;
+    double v = vals[i];
+    if (!kv || (unsigned long long)k == AGG_EMPTY) {
+      unsigned long long* pseen = kv ? &sp->m1_seen : &sp->nul_seen;
+      double* psum = kv ? &sp->m1_sum : &sp->nul_sum;
+      unsigned long long* pcnt = kv ? &sp->m1_cnt : &sp->nul_cnt;
+      atomicMax(pseen, 1ull);
+      if (OPS & AGG_OP_SUM) atomicAdd(psum, v);
+      if (OPS & AGG_OP_COUNT) atomicAdd(pcnt, 1ull);
+      out_k[i] = AGG_EMPTY;
+      out_v[i] = 0;
+      continue;
+    }
+    out_k[i] = (uint64_t)k;
+    out_v[i] = (unsigned long long)__double_as_longlong(v);
+  }
+}
+
+/* u64-payload scatter launcher (fixed 512x8 geometry: 16 B/element staging
+ * keeps two blocks per CU) */
+template <int BIN_MODE>
+static void launch_scatter64(hipStream_t s, int64_t nb, int64_t n,
+                             const uint64_t* kin, const uint64_t* iin,
+                             uint64_t* kout, uint64_t* iout,
+                             const uint32_t* scanned, int shift) {
+  k_radix_scatter<BIN_MODE, 512, 8, false, uint64_t><<<dim3((uint32_t)nb), 512, 0, s>>>(
+      n, kin, iin, kout, iout, scanned, shift, (int)nb, 0,
+      nullptr, nullptr, nullptr, 0, nullptr, 0);
+}
+
+struct pagg_ws {
+  uint64_t *pk_a, *pv_a, *pk_b, *pv_b;
+  uint32_t *hist, *hist_scan, *block_sums;
+  unsigned long long* tab;
+  agg_special* sp;
+};
+
+static void pagg_ws_layout(int64_t n, int64_t cap, pagg_ws* w, char* base, int64_t* total) {
+  const int tile = 512 * 8;
+  int64_t nb = sort_nblocks(n, tile);
+  int64_t hist_n = 256 * nb;
+  int64_t scan_blocks = (hist_n + SCAN_TILE - 1) / SCAN_TILE + 1;
+  int64_t off = 0;
+  auto take = [&](int64_t bytes) {
+    char* p = base ? base + off : nullptr;
+    off += (bytes + 255) & ~255LL;
+    return p;
+  };
+  w->pk_a = (uint64_t*)take(n * 8);
+  w->pv_a = (uint64_t*)take(n * 8);
+  w->pk_b = (uint64_t*)take(n * 8);
+  w->pv_b = (uint64_t*)take(n * 8);
+  w->hist = (uint32_t*)take(hist_n * 4);
+  w->hist_scan = (uint32_t*)take(hist_n * 4);
+  w->block_sums = (uint32_t*)take(scan_blocks * 4);
+  w->tab = (unsigned long long*)take(cap * 24);
+  w->sp = (agg_special*)take(sizeof(agg_special));
+  *total = off;
+}
+
+extern "C" int64_t gpuq_hash_agg_part_workspace_bytes(int64_t n, int64_t cap) {
+  pagg_ws w; int64_t total;
+  pagg_ws_layout(n, cap, &w, nullptr, &total);
+  return total;
+}
+
+/* Partitioned aggregation: same results contract as gpuq_hash_agg_i64_f64
+ * (single-shot; requires non-null values). Returns GPUQ_ERR_OVERFLOW with
+ * *out_ngroups = -1 when a chunk exceeded the LDS table (caller falls back
+ * to the direct path). */
+extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
+                                         gpuq_col key, gpuq_col val,
+                                         void* workspace, int64_t cap, int32_t ops,
+                                         int64_t* out_keys, uint8_t* out_key_valid,
+                                         double* out_sums, uint8_t* out_sum_valid,
+                                         int64_t* out_counts, int64_t* out_ngroups) {
+  hipStream_t s = (hipStream_t)stream;
+  if (cap <= 0 || (cap & (cap - 1)))
+    FAIL(GPUQ_ERR_INVALID, "pagg: capacity %lld not a power of two", (long long)cap);
+  if (key.dtype != GPUQ_INT64 || val.dtype != GPUQ_FLOAT64)
+    FAIL(GPUQ_ERR_INVALID, "pagg: expected int64 key + float64 val");
+  if (val.validity) FAIL(GPUQ_ERR_INVALID, "pagg: values must be non-null");
+  pagg_ws w; int64_t need;
+  pagg_ws_layout(n, cap, &w, (char*)workspace, &need);
+  k_agg_init<3><<<grid1d(cap), 256, 0, s>>>(cap, w.tab);
+  HIP_TRY(hipGetLastError());
+  HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_special), s));
+  if (n > 0) {
+    { hipEvent_t _pe = prof_begin(s);
+    if (ops == AGG_OP_SUM)
+      k_agg_part_pairs<AGG_OP_SUM><<<grid1d(n), 256, 0, s>>>(
+          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
+          w.pk_a, w.pv_a, w.sp);
+    else
+      k_agg_part_pairs<3><<<grid1d(n), 256, 0, s>>>(
+          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
+          w.pk_a, w.pv_a, w.sp);
+    prof_end("pagg_pairs", s, _pe); }
+    HIP_TRY(hipGetLastError());
+    const int tile = 512 * 8;
+    int64_t nb = sort_nblocks(n, tile);
+    uint64_t *kin = w.pk_a, *vin = w.pv_a, *kout = w.pk_b, *vout = w.pv_b;
+    for (int p = 0; p < 2; p++) {
+      { hipEvent_t _pe = prof_begin(s);
+      k_radix_hist<3><<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, p * 8, w.hist, (int)nb, tile);
+      prof_end("pagg_hist", s, _pe); }
+      HIP_TRY(hipGetLastError());
+      int rc = exclusive_scan_u32(s, 256 * nb, w.hist, w.hist_scan, w.block_sums);
+      if (rc) return rc;
+      { hipEvent_t _pe = prof_begin(s);
+      launch_scatter64<3>(s, nb, n, kin, vin, kout, vout, w.hist_scan, p * 8);
+      prof_end("pagg_scatter", s, _pe); }
+      HIP_TRY(hipGetLastError());
+      uint64_t* t;
+      t = kin; kin = kout; kout = t;
+      t = vin; vin = vout; vout = t;
+    }
+    { hipEvent_t _pe = prof_begin(s);
+    int64_t nchunks = (n + PAGG_CHUNK - 1) / PAGG_CHUNK;
+    if (ops == AGG_OP_SUM)
+      k_agg_part<AGG_OP_SUM, 3><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
+          n, kin, vin, w.tab, w.sp, cap - 1);
+    else
+      k_agg_part<3, 3><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
+          n, kin, vin, w.tab, w.sp, cap - 1);
+    prof_end("pagg_chunks", s, _pe); }
+    HIP_TRY(hipGetLastError());
+  }
+  agg_special hsp;
+  HIP_TRY(hipMemcpyAsync(&hsp, w.sp, sizeof(hsp), hipMemcpyDeviceToHost, s));
+  HIP_TRY(hipStreamSynchronize(s));
+  if (hsp.overflow == 2ull) { *out_ngroups = -1;
+    FAIL(GPUQ_ERR_OVERFLOW, "pagg: chunk exceeded the LDS table (fall back to direct)"); }
+  if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "pagg: table overflow (capacity %lld)", (long long)cap);
+  dim3 cgrid((uint32_t)((cap + AGGC_CHUNK - 1) / AGGC_CHUNK));
+  if (ops == AGG_OP_SUM)
+    k_agg_compact<AGG_OP_SUM, 3><<<cgrid, 256, 0, s>>>(
+        cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid, out_counts);
+  else
+    k_agg_compact<3, 3><<<cgrid, 256, 0, s>>>(
+        cap, w.tab, w.sp, out_keys, out_key_valid, out_sums, out_sum_valid, out_counts);
+  HIP_TRY(hipGetLastError());
+  agg_special hsp2;
+  HIP_TRY(hipMemcpyAsync(&hsp2, w.sp, sizeof(hsp2), hipMemcpyDeviceToHost, s));
+  HIP_TRY(hipStreamSynchronize(s));
+  *out_ngroups = (int64_t)hsp2.out_cursor;
+  return GPUQ_OK;
+}
+
+/* ================= multi-aggregate ================= *//* ================= multi-aggregate ================= */
 /*
  * One pass computing up to GPUQ_AGG_MAX_SPECS accumulators per group
  * (HashAggregateExec evaluates a LIST of aggregate expressions —

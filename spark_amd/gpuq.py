@@ -68,6 +68,12 @@ def lib() -> ctypes.CDLL:
         L.gpuq_hash_agg_i64_f64.restype = i32
         L.gpuq_hash_agg_i64_f64.argtypes = [vp, i64, _Col, _Col, vp, i64, i32, i32, i32,
                                             vp, vp, vp, vp, vp, ctypes.POINTER(i64)]
+        L.gpuq_hash_agg_part_workspace_bytes.restype = i64
+        L.gpuq_hash_agg_part_workspace_bytes.argtypes = [i64, i64]
+        L.gpuq_hash_agg_partitioned.restype = i32
+        L.gpuq_hash_agg_partitioned.argtypes = [vp, i64, _Col, _Col, vp, i64, i32,
+                                                vp, vp, vp, vp, vp,
+                                                ctypes.POINTER(i64)]
         L.gpuq_hash_agg_multi_workspace_bytes.restype = i64
         L.gpuq_hash_agg_multi_workspace_bytes.argtypes = [i64, i32]
         L.gpuq_hash_agg_multi.restype = i32
@@ -398,3 +404,29 @@ def range_partition_perm(keys: torch.Tensor, bounds: torch.Tensor, desc=False,
         bounds.data_ptr(), bounds.numel(), perm.data_ptr(), counts.data_ptr(),
         workspace.data_ptr(), workspace.numel()))
     return perm, counts
+
+
+def hash_agg_partitioned(keys: torch.Tensor, vals: torch.Tensor, capacity: int,
+                         workspace=None, max_groups=None, key_validity=None,
+                         ops=AGG_SUM | AGG_COUNT):
+    """Partitioned aggregation (bucket-ordered + LDS chunk tables); same
+    results contract as hash_agg for non-null values."""
+    n = keys.numel()
+    dev = keys.device
+    if workspace is None:
+        workspace = torch.empty(lib().gpuq_hash_agg_part_workspace_bytes(n, capacity),
+                                dtype=torch.uint8, device=dev)
+    mg = max_groups if max_groups is not None else min(n + 2, capacity + 2)
+    ok = torch.empty(mg, dtype=torch.int64, device=dev)
+    okv = torch.empty(mg, dtype=torch.uint8, device=dev)
+    osum = torch.empty(mg, dtype=torch.float64, device=dev)
+    osv = torch.empty(mg, dtype=torch.uint8, device=dev)
+    ocnt = torch.empty(mg, dtype=torch.int64, device=dev)
+    ng = ctypes.c_int64(0)
+    _check(lib().gpuq_hash_agg_partitioned(
+        _stream(), n, _col(keys, key_validity), _col(vals),
+        workspace.data_ptr(), capacity, ops,
+        ok.data_ptr(), okv.data_ptr(), osum.data_ptr(), osv.data_ptr(),
+        ocnt.data_ptr(), ctypes.byref(ng)))
+    gn = ng.value
+    return ok[:gn], okv[:gn], osum[:gn], osv[:gn], ocnt[:gn]

@@ -379,3 +379,17 @@ def test_multi_aggregate_sum_int64(gq):
     np.add.at(exp, keys, vals)
     assert (ok.cpu().numpy()[g] == uk).all()
     assert (accs[0].cpu().numpy()[g] == exp).all()  # bit-exact int64 sums
+
+
+@pytest.mark.parametrize("n,ngroups", [(2_000_000, 100_000), (500_000, 37),
+                                       (100_000, 90_000)])
+def test_partitioned_agg_parity(gq, n, ngroups):
+    keys = oracle.gen_i64(seed=n + 7, n=n, range_=ngroups)
+    # sprinkle -1 keys (special path) on top
+    keys[::1000] = -1
+    vals = oracle.gen_f64_unit(seed=n + 8, n=n)
+    cap = 1 << max(6, (ngroups * 2 - 1).bit_length())
+    gk, gkv, gs, gsv, gc = (t.cpu().numpy() for t in
+                            gq.hash_agg_partitioned(to_dev(keys), to_dev(vals), cap))
+    ok, okv, osum, osv, ocnt = oracle.hash_agg(keys, vals)
+    agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt)
