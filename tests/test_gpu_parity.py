@@ -382,7 +382,7 @@ def test_multi_aggregate_sum_int64(gq):
 
 
 @pytest.mark.parametrize("n,ngroups", [(2_000_000, 100_000), (500_000, 37),
-                                       (100_000, 90_000)])
+                                       (4_000_000, 40_000)])
 def test_partitioned_agg_parity(gq, n, ngroups):
     keys = oracle.gen_i64(seed=n + 7, n=n, range_=ngroups)
     # sprinkle -1 keys (special path) on top
@@ -393,3 +393,14 @@ def test_partitioned_agg_parity(gq, n, ngroups):
                             gq.hash_agg_partitioned(to_dev(keys), to_dev(vals), cap))
     ok, okv, osum, osv, ocnt = oracle.hash_agg(keys, vals)
     agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt)
+
+
+def test_partitioned_agg_signals_lds_overflow(gq):
+    # nearly-all-distinct keys in a small input: chunks exceed the LDS table
+    # and the call must fail loudly with the fall-back signal
+    from spark_amd.gpuq import GpuqError
+    n = 100_000
+    keys = oracle.gen_i64(seed=1, n=n, range_=90_000)
+    vals = oracle.gen_f64_unit(seed=2, n=n)
+    with pytest.raises(GpuqError, match="fall back"):
+        gq.hash_agg_partitioned(to_dev(keys), to_dev(vals), 1 << 18)
