@@ -729,6 +729,14 @@ static void launch_scatter(hipStream_t s, scatter_geom g, int64_t nb,
   else if (g.block == 512 && g.items == 16) LS(512, 16);
   else if (g.block == 1024 && g.items == 8) LS(1024, 8);
   else if (g.block == 512 && g.items == 4) LS(512, 4);
+  else if (g.block == 512 && g.items == 6)
+    k_radix_scatter<BIN_MODE, 512, 6, LOOKBACK><<<grid, 512, 0, s>>>(
+      n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+      err_flag, epoch, decode_out, decode_mode);
+  else if (g.block == 256 && g.items == 12)
+    k_radix_scatter<BIN_MODE, 256, 12, LOOKBACK><<<grid, 256, 0, s>>>(
+      n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+      err_flag, epoch, decode_out, decode_mode);
   else if (g.block == 512 && g.items == 12)
     k_radix_scatter<BIN_MODE, 512, 12, LOOKBACK><<<grid, 512, 0, s>>>(
       n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
