@@ -277,6 +277,9 @@ class GpuSortExec(SparkPlan):
                 # RangePartitioning insertion, exchange/EnsureRequirements.scala:296);
                 # after the local sort below, rank-major order is global order
                 from .exchange import range_exchange
+                assert batch.validity(o.key) is None, (
+                    "global ORDER BY with NULL keys across ranks: validity "
+                    "does not travel through the exchange yet (round-2)")
                 payload = {n_: t for n_, t in batch.columns().items()
                            if n_ != o.key}
                 k, payload = range_exchange(batch.column(o.key), payload,

@@ -74,6 +74,7 @@ def test_sort_all_equal_keys(gq):
     keys = np.full(n, 7, dtype=np.int64)
     perm, skeys = gq.sort_perm(to_dev(keys))
     assert (perm.cpu().numpy() == np.arange(n)).all()  # zero passes, identity
+    assert (skeys.cpu().numpy() == 7).all()            # zero-pass decode path
 
 
 def test_sort_gather_payload(gq):
