@@ -98,3 +98,22 @@ def test_join_duplicates_both_sides():
     assert len(op) == 6
     assert sorted(zip(op.tolist(), ob.tolist())) == [
         (0, 0), (0, 1), (0, 2), (1, 0), (1, 1), (1, 2)]
+
+
+def test_baseline_config1_groupby_sum_1m():
+    """BASELINE configs[0]: local[2] df.groupBy(key).agg(sum(val)) on a
+    1M-row 2-col DataFrame — the CPU-runnable case, checked through the
+    oracle restatement with the checkAnswer-style order-insensitive compare."""
+    n = 1_000_000
+    keys = oracle.gen_i64(1042, n, range_=10_000)
+    vals = oracle.gen_f64_unit(1043, n)
+    ok, okv, osum, osv, ocnt = oracle.hash_agg(keys, vals)
+    assert okv.all() and osv.all()
+    # independent check: numpy groupby
+    uk = np.unique(keys)
+    exp = np.zeros(len(uk))
+    np.add.at(exp, np.searchsorted(uk, keys), vals)
+    order = np.argsort(ok)
+    assert (ok[order] == uk).all()
+    np.testing.assert_allclose(osum[order], exp, rtol=1e-9)
+    assert int(ocnt.sum()) == n
