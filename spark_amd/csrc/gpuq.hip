@@ -1256,83 +1256,45 @@ __global__ void k_agg_init(int64_t cap, unsigned long long* tab) {
 #define AGG_OP_SUM 1
 #define AGG_OP_COUNT 2
 
-/* slow path for one row (specials, collisions, validity) */
-template <int OPS, int SLOT>
-DEV bool agg_build_row(int64_t i, int64_t k, bool kv, bool vv, double v,
-                       unsigned long long first_probe, bool have_first,
-                       unsigned long long* tab, agg_special* sp,
-                       int64_t cap_mask) {
-  if (!kv || (unsigned long long)k == AGG_EMPTY) {
-    double* psum = kv ? &sp->m1_sum : &sp->nul_sum;
-    unsigned long long* pcnt = kv ? &sp->m1_cnt : &sp->nul_cnt;
-    unsigned long long* pseen = kv ? &sp->m1_seen : &sp->nul_seen;
-    atomicMax(pseen, 1ull);
-    if (vv) {
-      if (OPS & AGG_OP_SUM) atomicAdd(psum, v);
-      if (OPS & AGG_OP_COUNT) atomicAdd(pcnt, 1ull);
-    }
-    return true;
-  }
-  uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
-  for (int probes = 0;; probes++) {
-    unsigned long long cur = (probes == 0 && have_first) ? first_probe
-        : __hip_atomic_load(&tab[SLOT * slot], __ATOMIC_RELAXED,
-                            __HIP_MEMORY_SCOPE_AGENT);
-    if (cur == (unsigned long long)k) break;
-    if (cur == AGG_EMPTY) {
-      unsigned long long prev = atomicCAS(&tab[SLOT * slot], AGG_EMPTY, (unsigned long long)k);
-      if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
-    }
-    slot = (slot + 1) & (uint64_t)cap_mask;
-    if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return false; }
-  }
-  if (vv) {
-    if (OPS & AGG_OP_SUM) atomicAdd((double*)&tab[SLOT * slot + 1], v);
-    if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], 1ull);
-  }
-  return true;
-}
-
 template <int OPS, int SLOT>
 __global__ void k_agg_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                             const double* vals, const uint8_t* vvalid,
                             unsigned long long* tab, agg_special* sp,
                             int64_t cap_mask) {
-  /* waves were ~75% parked on the dependent probe load (SQ counters):
-   * batch four rows so four first-probe loads are in flight at once */
-  constexpr int B = 4;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  for (; i + (B - 1) * stride < n; i += B * stride) {
-    int64_t kk[B]; double vv_[B]; bool kv[B], vv[B];
-    uint64_t slot[B];
-    unsigned long long first[B];
-    #pragma unroll
-    for (int r = 0; r < B; r++) {
-      int64_t j = i + r * stride;
-      kv[r] = bit_valid(kvalid, j);
-      vv[r] = bit_valid(vvalid, j);
-      kk[r] = kv[r] ? keys[j] : 0;
-      vv_[r] = vv[r] ? vals[j] : 0.0;
-      slot[r] = ((uint32_t)mm3_hash_long(kk[r], 42)) & (uint64_t)cap_mask;
-    }
-    #pragma unroll
-    for (int r = 0; r < B; r++)
-      first[r] = __hip_atomic_load(&tab[SLOT * slot[r]], __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_AGENT);
-    #pragma unroll
-    for (int r = 0; r < B; r++)
-      if (!agg_build_row<OPS, SLOT>(i + r * stride, kk[r], kv[r], vv[r], vv_[r],
-                                    first[r], true, tab, sp, cap_mask))
-        return;
-  }
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
     bool kv = bit_valid(kvalid, i);
     bool vv = bit_valid(vvalid, i);
-    int64_t k = kv ? keys[i] : 0;
     double v = vv ? vals[i] : 0.0;
-    if (!agg_build_row<OPS, SLOT>(i, k, kv, vv, v, 0, false, tab, sp, cap_mask))
-      return;
+    if (!kv || (unsigned long long)keys[i] == AGG_EMPTY) {
+      double* psum = kv ? &sp->m1_sum : &sp->nul_sum;
+      unsigned long long* pcnt = kv ? &sp->m1_cnt : &sp->nul_cnt;
+      unsigned long long* pseen = kv ? &sp->m1_seen : &sp->nul_seen;
+      atomicMax(pseen, 1ull);
+      if (vv) {
+        if (OPS & AGG_OP_SUM) atomicAdd(psum, v);
+        if (OPS & AGG_OP_COUNT) atomicAdd(pcnt, 1ull);
+      }
+      continue;
+    }
+    int64_t k = keys[i];
+    uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (uint64_t)cap_mask;
+    for (int probes = 0;; probes++) {
+      unsigned long long cur = __hip_atomic_load(&tab[SLOT * slot], __ATOMIC_RELAXED,
+                                                 __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == (unsigned long long)k) break;
+      if (cur == AGG_EMPTY) {
+        unsigned long long prev = atomicCAS(&tab[SLOT * slot], AGG_EMPTY, (unsigned long long)k);
+        if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
+      }
+      slot = (slot + 1) & (uint64_t)cap_mask;
+      if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
+    }
+    if (vv) {
+      if (OPS & AGG_OP_SUM) atomicAdd((double*)&tab[SLOT * slot + 1], v);
+      if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], 1ull);
+    }
   }
 }
 
