@@ -94,3 +94,20 @@ def test_join_halfbillion_properties(gq):
     hit = bs[idx_c] == ps
     expected = int((bk_c[bo][idx_c][hit] * pk_c[po][hit]).sum())
     assert nm == expected
+
+
+def test_sort_past_int32_rows(gq):
+    """Row ids are u32: exercise the path past 2^31 rows (config-2 headroom;
+    2.2B rows x 12B pairs + workspace ~ 160 GB of the 288 GB HBM)."""
+    n = 2_200_000_000
+    if ROWS < 1_000_000_000:
+        pytest.skip("scaled-down run")
+    keys = gq.gen_i64(seed=2, n=n)
+    ws = gq.sort_workspace(n)
+    perm, skeys = gq.sort_perm(keys, workspace=ws)
+    del ws
+    assert bool((skeys[1:] >= skeys[:-1]).all())
+    pu = perm.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    assert bool((skeys == keys[pu]).all())
+    del keys, skeys, perm, pu
+    torch.cuda.empty_cache()
