@@ -834,7 +834,10 @@ void k_scatter_ablate(int64_t n, const uint64_t* kin, const uint32_t* iin,
     for (int w = 0; w < wave; w++) woff += wtot[w];
     uint32_t excl = bin_start[bin] + woff;
     bin_start[bin] = excl;
-    bin_gbase[bin] = gbase[bin] + (uint32_t)(base >> 2) - excl; /* fake but data-dependent */
+    /* realistic layout: uniform data contributes ~TILE/256 rows per bin per
+     * block, so block b's bin-k run starts near gbase[k] + b*TILE/256 —
+     * the same write pattern the real scanned offsets produce */
+    bin_gbase[bin] = gbase[bin] + (uint32_t)(base >> 8) - excl;
   }
   __syncthreads();
   if (MODE == 1) {

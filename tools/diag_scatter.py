@@ -11,7 +11,7 @@ keys = gq.gen_i64(seed=1, n=n)           # raw bits as u64 keys
 idx = torch.arange(n, dtype=torch.int32, device="cuda")
 kout = torch.empty(n, dtype=torch.int64, device="cuda")
 iout = torch.empty(n, dtype=torch.int32, device="cuda")
-gbase = torch.zeros(256, dtype=torch.int32, device="cuda")
+gbase = (torch.arange(256, dtype=torch.int64, device="cuda") * (n // 256)).to(torch.int32)
 sink = torch.zeros(1, dtype=torch.int64, device="cuda")
 L = gq.lib()
 L.gpuq_scatter_ablate.restype = ctypes.c_int32
