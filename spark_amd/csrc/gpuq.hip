@@ -682,20 +682,37 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
       uint64_t kk = stage_k[j];
       int bin = compute_bin<BIN_MODE>(kk, shift, nparts);
       uint32_t dst = bin_gbase[bin] + (uint32_t)j;
+      uint64_t kv_out;
+      void* kdst;
       if (decode_mode) {
         /* final pass: emit the DECODED key column directly (the separate
            decode kernel and this pass's encoded-key write both disappear) */
         uint64_t e = (decode_mode & 4) ? ~kk : kk;
         if (decode_mode & 2) {
           uint64_t mask = ((e >> 63) ? 0 : ~0ULL) | SIGNBIT;
-          ((uint64_t*)decode_out)[dst] = e ^ mask;
+          kv_out = e ^ mask;
         } else {
-          ((int64_t*)decode_out)[dst] = (int64_t)(e ^ SIGNBIT);
+          kv_out = e ^ SIGNBIT;
         }
+        kdst = &((uint64_t*)decode_out)[dst];
       } else {
-        kout[dst] = kk;
+        kv_out = kk;
+        kdst = &kout[dst];
       }
+#ifdef GPUQ_DRAIN_SC1
+      /* write-through: scattered partial-line stores skip the L2
+       * write-allocate read-for-ownership (guide: publish-large) */
+      __hip_atomic_store((uint64_t*)kdst, kv_out, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store((PayT*)&iout[dst], stage_i[j], __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+#elif defined(GPUQ_DRAIN_NT)
+      __builtin_nontemporal_store(kv_out, (uint64_t*)kdst);
+      __builtin_nontemporal_store(stage_i[j], &iout[dst]);
+#else
+      *(uint64_t*)kdst = kv_out;
       iout[dst] = stage_i[j];
+#endif
     }
   }
 }
