@@ -889,6 +889,127 @@ void k_scatter_ablate(int64_t n, const uint64_t* kin, const uint32_t* iin,
   }
 }
 
+/* MODE 4/5: 512x8 tile-4096 variant comparing drain store shapes:
+ * 4 = two stores per element (8B key + 4B idx, separate arrays, as shipped)
+ * 5 = one 16B record store per element (key, idx, pad) */
+template <int MODE>
+__global__ __launch_bounds__(512)
+void k_scatter_ablate2(int64_t n, const uint64_t* kin, const uint32_t* iin,
+                       uint64_t* kout, uint32_t* iout, uint32_t* rout,
+                       const uint32_t* gbase, int shift) {
+  constexpr int BLOCK = 512, ITEMS = 8;
+  constexpr int WAVES = BLOCK / WAVE;
+  constexpr int TILE = BLOCK * ITEMS;
+  __shared__ uint32_t wave_hist[WAVES][256];
+  __shared__ uint32_t bin_start[256];
+  __shared__ uint32_t bin_gbase[256];
+  __shared__ uint32_t wtot[WAVES];
+  __shared__ uint64_t stage_k[TILE];
+  __shared__ uint32_t stage_i[TILE];
+  const int tid = threadIdx.x;
+  const int wave = tid / WAVE, lane = tid & (WAVE - 1);
+  const int64_t base = (int64_t)blockIdx.x * TILE;
+  const int tile_n = (int)min((int64_t)TILE, n - base);
+  for (int b = tid; b < WAVES * 256; b += BLOCK) ((uint32_t*)wave_hist)[b] = 0;
+  __syncthreads();
+  uint64_t k[ITEMS]; uint32_t id[ITEMS]; uint16_t lrank[ITEMS]; uint8_t lbin[ITEMS];
+  const int64_t wbase = base + (int64_t)wave * WAVE * ITEMS;
+  #pragma unroll
+  for (int r = 0; r < ITEMS; r++) {
+    int64_t i = wbase + r * WAVE + lane;
+    bool valid = i < n;
+    k[r] = valid ? kin[i] : 0;
+    id[r] = valid ? iin[i] : 0;
+  }
+  for (int r = 0; r < ITEMS; r++) {
+    int64_t i = wbase + r * WAVE + lane;
+    bool valid = i < n;
+    int bin = valid ? (int)((k[r] >> shift) & 0xff) : 0;
+    lbin[r] = (uint8_t)bin;
+    uint64_t m = __ballot(valid);
+    for (int b = 0; b < 8; b++) {
+      uint64_t bl = __ballot((bin >> b) & 1);
+      m &= ((bin >> b) & 1) ? bl : ~bl;
+    }
+    uint64_t below = m & ((1ULL << lane) - 1);
+    int rank = __popcll(below);
+    int leader = __ffsll((unsigned long long)m) - 1;
+    uint32_t basecnt = 0;
+    if (valid && lane == leader) {
+      basecnt = wave_hist[wave][bin];
+      wave_hist[wave][bin] = basecnt + __popcll(m);
+    }
+    basecnt = __shfl(basecnt, leader);
+    lrank[r] = (uint16_t)(basecnt + rank);
+    __builtin_amdgcn_wave_barrier();
+  }
+  __syncthreads();
+  if (tid < 256) {
+    int bin = tid;
+    uint32_t acc = 0;
+    for (int w = 0; w < WAVES; w++) {
+      uint32_t t = wave_hist[w][bin];
+      wave_hist[w][bin] = acc; acc += t;
+    }
+    uint32_t inc = wave_inclusive_scan(acc);
+    if (lane == WAVE - 1) wtot[wave] = inc;
+    bin_start[bin] = inc - acc;
+  }
+  __syncthreads();
+  if (tid < 256) {
+    int bin = tid;
+    uint32_t woff = 0;
+    for (int w = 0; w < wave; w++) woff += wtot[w];
+    uint32_t excl = bin_start[bin] + woff;
+    bin_start[bin] = excl;
+    bin_gbase[bin] = gbase[bin] + (uint32_t)(base >> 8) - excl;
+  }
+  __syncthreads();
+  for (int r = 0; r < ITEMS; r++) {
+    int64_t i = wbase + r * WAVE + lane;
+    if (i < n) {
+      uint32_t pos = bin_start[lbin[r]] + wave_hist[wave][lbin[r]] + lrank[r];
+      stage_k[pos] = k[r];
+      stage_i[pos] = id[r];
+    }
+  }
+  __syncthreads();
+  for (int r = 0; r < ITEMS; r++) {
+    int j = r * BLOCK + tid;
+    if (j < tile_n) {
+      uint64_t kk = stage_k[j];
+      int bin = (int)((kk >> shift) & 0xff);
+      uint32_t dst = (bin_gbase[bin] + (uint32_t)j) % (uint32_t)n;
+      if (MODE == 4) {
+        kout[dst] = kk;
+        iout[dst] = stage_i[j];
+      } else {
+        uint4 rec;
+        rec.x = (uint32_t)kk; rec.y = (uint32_t)(kk >> 32);
+        rec.z = stage_i[j]; rec.w = 0;
+        ((uint4*)rout)[dst] = rec;
+      }
+    }
+  }
+}
+
+extern "C" int gpuq_scatter_ablate2(void* stream, int64_t n, const void* kin,
+                                    const void* iin, void* kout, void* iout,
+                                    void* rout, const void* gbase, int32_t mode) {
+  hipStream_t s = (hipStream_t)stream;
+  int64_t nb = (n + 4095) / 4096;
+  if (mode == 4)
+    k_scatter_ablate2<4><<<dim3((uint32_t)nb), 512, 0, s>>>(
+        n, (const uint64_t*)kin, (const uint32_t*)iin, (uint64_t*)kout,
+        (uint32_t*)iout, (uint32_t*)rout, (const uint32_t*)gbase, 0);
+  else
+    k_scatter_ablate2<5><<<dim3((uint32_t)nb), 512, 0, s>>>(
+        n, (const uint64_t*)kin, (const uint32_t*)iin, (uint64_t*)kout,
+        (uint32_t*)iout, (uint32_t*)rout, (const uint32_t*)gbase, 0);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
 extern "C" int gpuq_scatter_ablate(void* stream, int64_t n, const void* kin,
                                    const void* iin, void* kout, void* iout,
                                    const void* gbase, void* sink, int32_t mode) {
