@@ -29,3 +29,20 @@ for mode, name in [(0, "loads"), (1, "+rank"), (2, "+stage"), (3, "+drain(full)"
         run()
     torch.cuda.synchronize()
     print(f"mode {mode} {name:14s} {(time.perf_counter()-t0)/5*1e3:7.2f} ms", flush=True)
+
+# drain store shape A/B (512x8 tile 4096): two stores (12B) vs one 16B record
+rout = torch.empty(n * 2, dtype=torch.int64, device="cuda")  # 16B records
+L.gpuq_scatter_ablate2.restype = ctypes.c_int32
+L.gpuq_scatter_ablate2.argtypes = [ctypes.c_void_p, ctypes.c_int64] + [ctypes.c_void_p] * 6 + [ctypes.c_int32]
+for mode, name in [(4, "two-stores-12B"), (5, "one-16B-record")]:
+    def run2():
+        rc = L.gpuq_scatter_ablate2(gq._stream(), n, keys.data_ptr(), idx.data_ptr(),
+                                    kout.data_ptr(), iout.data_ptr(), rout.data_ptr(),
+                                    gbase.data_ptr(), mode)
+        assert rc == 0
+    run2(); torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(5):
+        run2()
+    torch.cuda.synchronize()
+    print(f"mode {mode} {name:15s} {(time.perf_counter()-t0)/5*1e3:7.2f} ms", flush=True)
