@@ -35,8 +35,8 @@ def lib() -> ctypes.CDLL:
         build()
         _lib = ctypes.CDLL(_SO)
         L = _lib
-        i8, i32, i64, u64, f64, b = (
-            ctypes.c_uint8, ctypes.c_int32, ctypes.c_int64, ctypes.c_uint64,
+        i32, i64, u64, f64, b = (
+            ctypes.c_int32, ctypes.c_int64, ctypes.c_uint64,
             ctypes.c_double, ctypes.c_bool)
         P = ctypes.POINTER
         L.mm3_hash_int.restype = i32

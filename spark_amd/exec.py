@@ -13,7 +13,7 @@ layer would; it is what tests and the TPC-H-style pipelines drive.
 No CPU fallback anywhere: executing these nodes without the HIP engine (or
 without a GPU) raises.
 """
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 import torch
