@@ -502,7 +502,8 @@ static int exclusive_scan_u32(hipStream_t s, int64_t n, const uint32_t* in,
 #define OSW_EPOCH(x) ((x) >> 56)
 #define OSW_SPIN_LIMIT (1u << 22)
 
-template <int BIN_MODE, int BLOCK, int ITEMS, bool LOOKBACK, typename PayT = uint32_t>
+template <int BIN_MODE, int BLOCK, int ITEMS, bool LOOKBACK, typename PayT = uint32_t,
+          int RADIX_BITS = 8>
 __global__ __launch_bounds__(BLOCK)
 void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
                      uint64_t* kout, PayT* iout,
@@ -516,9 +517,10 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
                      int decode_mode /* 1=i64 2=f64, +4 = desc */) {
   constexpr int WAVES = BLOCK / WAVE;
   constexpr int TILE = BLOCK * ITEMS;
-  __shared__ uint32_t wave_hist[WAVES][256];
-  __shared__ uint32_t bin_start[256];     /* in-block exclusive start per bin */
-  __shared__ uint32_t bin_gbase[256];     /* global dest minus local start    */
+  constexpr int BINS = 1 << RADIX_BITS;
+  __shared__ uint32_t wave_hist[WAVES][BINS];
+  __shared__ uint32_t bin_start[BINS];    /* in-block exclusive start per bin */
+  __shared__ uint32_t bin_gbase[BINS];    /* global dest minus local start    */
   __shared__ uint32_t wtot[WAVES <= 4 ? 4 : WAVES];
   __shared__ uint64_t stage_k[TILE];
   __shared__ PayT stage_i[TILE];
@@ -528,7 +530,7 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
   const int64_t base = (int64_t)blockIdx.x * TILE;
   const int tile_n = (int)min((int64_t)TILE, n - base);
 
-  for (int b = tid; b < WAVES * 256; b += BLOCK)
+  for (int b = tid; b < WAVES * BINS; b += BLOCK)
     ((uint32_t*)wave_hist)[b] = 0;
   __syncthreads();
 
@@ -553,11 +555,12 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
     int64_t i = wbase + r * WAVE + lane;
     bool valid = i < n;
     int bin = valid ? compute_bin<BIN_MODE>(k[r], shift, nparts) : 0;
+    if (RADIX_BITS < 8) bin &= (1 << RADIX_BITS) - 1;
     lbin[r] = (uint8_t)bin;
     /* ballot multi-split: mask of lanes in this wave with the same bin */
     uint64_t active = __ballot(valid);
     uint64_t same = active;
-    for (int b = 0; b < 8; b++) {
+    for (int b = 0; b < RADIX_BITS; b++) {
       uint64_t bl = __ballot((bin >> b) & 1);
       same &= ((bin >> b) & 1) ? bl : ~bl;
     }
@@ -582,7 +585,7 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
   /* cross-wave exclusive prefix per bin + block-wide exclusive scan over
    * bins. The first 256 threads each own one bin. */
   uint32_t acc = 0;
-  if (tid < 256) {
+  if (tid < BINS) {
     int bin = tid;
     for (int w = 0; w < WAVES; w++) {
       uint32_t t = wave_hist[w][bin];
@@ -600,7 +603,7 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
     bin_start[bin] = inc - acc;  /* provisional; add wave offsets after sync */
   }
   __syncthreads();
-  if (tid < 256) {
+  if (tid < BINS) {
     int bin = tid;
     uint32_t woff = 0;
     for (int w = 0; w < wave; w++) woff += wtot[w];
@@ -681,6 +684,7 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
     if (j < tile_n) {
       uint64_t kk = stage_k[j];
       int bin = compute_bin<BIN_MODE>(kk, shift, nparts);
+      if (RADIX_BITS < 8) bin &= (1 << RADIX_BITS) - 1;
       uint32_t dst = bin_gbase[bin] + (uint32_t)j;
       uint64_t kv_out;
       void* kdst;
@@ -1026,6 +1030,25 @@ extern "C" int gpuq_scatter_ablate(void* stream, int64_t n, const void* kin,
   return GPUQ_OK;
 }
 
+template <int BIN_MODE, bool LOOKBACK>
+static void launch_scatter4(hipStream_t s, scatter_geom g, int64_t nb,
+                            int64_t n, const uint64_t* kin, const uint32_t* iin,
+                            uint64_t* kout, uint32_t* iout,
+                            const uint32_t* scanned, int shift, int nparts,
+                            unsigned long long* state, const uint32_t* gbase,
+                            unsigned long long* err_flag, uint32_t epoch,
+                            void* decode_out = nullptr, int decode_mode = 0) {
+  dim3 grid((uint32_t)nb);
+  if (g.block == 1024 && g.items == 8)
+    k_radix_scatter<BIN_MODE, 1024, 8, LOOKBACK, uint32_t, 4><<<grid, 1024, 0, s>>>(
+        n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+        err_flag, epoch, decode_out, decode_mode);
+  else
+    k_radix_scatter<BIN_MODE, 512, 10, LOOKBACK, uint32_t, 4><<<grid, 512, 0, s>>>(
+        n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+        err_flag, epoch, decode_out, decode_mode);
+}
+
 /* decode sorted keys back to the output dtype */
 template <int DTYPE, bool DESC>
 __global__ void k_decode(int64_t n, const uint64_t* ek, void* out) {
@@ -1176,9 +1199,11 @@ extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
   HIP_TRY(hipStreamSynchronize(s));
   uint64_t bits_changed = hb[0] ^ hb[1];
 
+  bool nibble_mode = onesweep && getenv("GPUQ_SORT_BITS") != nullptr &&
+                     atoi(getenv("GPUQ_SORT_BITS")) == 4;
   int retries = 0;
 retry:
-  if (onesweep) {
+  if (onesweep && !nibble_mode) {
     /* per-pass exclusive bin bases from the one-read global histograms
      * (replaces the per-pass hist kernel + device scan) */
     uint32_t hgbase[8 * 256];
@@ -1192,18 +1217,39 @@ retry:
     HIP_TRY(hipMemcpyAsync(w.gbase, hgbase, sizeof(hgbase), hipMemcpyHostToDevice, s));
     HIP_TRY(hipMemsetAsync(w.state, 0, nb * 256 * 8, s));
     HIP_TRY(hipMemsetAsync(w.err, 0, 8, s));
+  } else if (nibble_mode) {
+    /* 16 levels of 4-bit digits (16-bin drains write 16x-longer runs);
+     * level bases derived from the same byte histograms */
+    uint32_t hgbase[16 * 16];
+    for (int lvl = 0; lvl < 16; lvl++) {
+      int byte = lvl / 2;
+      bool high = lvl & 1;
+      uint32_t cnt[16] = {0};
+      for (int v = 0; v < 256; v++) {
+        int nib = high ? (v >> 4) : (v & 15);
+        cnt[nib] += hghist[byte * 256 + v];
+      }
+      uint32_t run = 0;
+      for (int v = 0; v < 16; v++) { hgbase[lvl * 16 + v] = run; run += cnt[v]; }
+    }
+    HIP_TRY(hipMemcpyAsync(w.gbase, hgbase, sizeof(hgbase), hipMemcpyHostToDevice, s));
+    HIP_TRY(hipMemsetAsync(w.state, 0, nb * 256 * 8, s));
+    HIP_TRY(hipMemsetAsync(w.err, 0, 8, s));
   }
 
   /* last active pass writes row ids straight into out_perm (saves the
    * 4 B/row device copy) */
+  int nlevels = nibble_mode ? 16 : 8;
+  int lvl_bits = nibble_mode ? 4 : 8;
+  uint64_t lvl_mask = nibble_mode ? 0xfULL : 0xffULL;
   int last_byte = -1;
-  for (int byte = 0; byte < 8; byte++)
-    if (((bits_changed >> (byte * 8)) & 0xff) != 0) last_byte = byte;
+  for (int lvl = 0; lvl < nlevels; lvl++)
+    if (((bits_changed >> (lvl * lvl_bits)) & lvl_mask) != 0) last_byte = lvl;
   uint64_t *kin = w.ka, *kout = w.kb;
   uint32_t *iin = w.ia, *iout = w.ib;
-  for (int byte = 0; byte < 8; byte++) {
-    if (((bits_changed >> (byte * 8)) & 0xff) == 0) continue;  /* RadixSort.java:126 skip */
-    int shift = byte * 8;
+  for (int byte = 0; byte < nlevels; byte++) {
+    if (((bits_changed >> (byte * lvl_bits)) & lvl_mask) == 0) continue;  /* RadixSort.java:126 skip */
+    int shift = byte * lvl_bits;
     uint32_t* iout_pass = (byte == last_byte) ? out_perm + sorted_at : iout;
     /* fuse the key decode into the final pass when the caller wants keys
      * and the null path is not rerouting them through a gather */
@@ -1213,7 +1259,14 @@ retry:
       dec_out = (char*)out_keys;  /* sorted_at == 0 without validity */
       dec_mode = (key.dtype == GPUQ_FLOAT64 ? 2 : 1) | (desc ? 4 : 0);
     }
-    if (onesweep) {
+    if (nibble_mode) {
+      { hipEvent_t _pe = prof_begin(s);
+      launch_scatter4<0, true>(s, geom, nb, n_sort, kin, iin, kout, iout_pass, nullptr,
+                               shift, 0, w.state, w.gbase + byte * 16, w.err,
+                               (uint32_t)(byte + 1), dec_out, dec_mode);
+      prof_end("radix_scatter", s, _pe); }
+      HIP_TRY(hipGetLastError());
+    } else if (onesweep) {
       { hipEvent_t _pe = prof_begin(s);
       launch_scatter<0, true>(s, geom, nb, n_sort, kin, iin, kout, iout_pass, nullptr,
                               shift, 0, w.state, w.gbase + byte * 256, w.err,
