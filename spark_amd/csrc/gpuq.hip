@@ -630,9 +630,8 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
     int bin = tid;
     uint32_t excl = bin_gbase[bin];
     uint32_t own = acc;   /* this thread's phase-2 block count for its bin */
-    /* chunked decoupled lookback: LB predecessor words in flight per step
-     * (with few bins there are few walker lanes, so go wider) */
-    constexpr int LB = (BINS <= 16) ? 16 : 4;
+    /* chunked decoupled lookback: LB predecessor words in flight per step */
+    constexpr int LB = 4;
     unsigned long long pred = 0;
     uint32_t spins = 0;
     int p = (int)blockIdx.x - 1;
