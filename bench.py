@@ -41,6 +41,8 @@ def env_rank():
 
 def dist_setup():
     rank, world, local = env_rank()
+    if os.environ.get("GPUQ_FORCE_DEV0"):   # single-GPU multi-rank debug
+        local = 0
     torch.cuda.set_device(local if world > 1 else 0)
     if world > 1:
         import torch.distributed as dist
