@@ -51,6 +51,9 @@ def lib() -> ctypes.CDLL:
         L.prefix_double.argtypes = [f64]
         L.oracle_partition_ids_i64.restype = None
         L.oracle_partition_ids_i64.argtypes = [ctypes.c_void_p, ctypes.c_void_p, i64, i32, ctypes.c_void_p]
+        L.oracle_partition_ids_i64_multi.restype = None
+        L.oracle_partition_ids_i64_multi.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                                     i32, i64, i32, ctypes.c_void_p]
         L.oracle_radix_sort_longs.restype = i64
         L.oracle_radix_sort_longs.argtypes = [ctypes.c_void_p, i64, i32, i32, b, b]
         L.oracle_radix_sort_key_prefix.restype = i64
@@ -117,6 +120,19 @@ def partition_ids(keys: np.ndarray, num_parts: int, validity=None) -> np.ndarray
     keys = np.ascontiguousarray(keys, dtype=np.int64)
     out = np.empty(len(keys), dtype=np.int32)
     lib().oracle_partition_ids_i64(_ptr(keys), _ptr(validity), len(keys), num_parts, _ptr(out))
+    return out
+
+
+def partition_ids_multi(key_cols, num_parts: int, validity=None) -> np.ndarray:
+    """Partition ids for multi-column keys (hash chains column-wise,
+    hash.scala:849-860). key_cols: list of int64 arrays; validity: optional
+    stacked bitmaps [ncols][ceil(n/8)]."""
+    cols = np.ascontiguousarray(np.stack([np.asarray(c, dtype=np.int64)
+                                          for c in key_cols]))
+    n = cols.shape[1]
+    out = np.empty(n, dtype=np.int32)
+    lib().oracle_partition_ids_i64_multi(_ptr(cols), _ptr(validity), cols.shape[0],
+                                         n, num_parts, _ptr(out))
     return out
 
 

@@ -131,6 +131,25 @@ EXPORT void oracle_partition_ids_i64(const int64_t* keys, const uint8_t* validit
   }
 }
 
+/* multi-column partition ids: the running hash chains column-wise as the
+ * next column's seed (hash.scala:849-860 HashExpression.eval — Murmur3Hash
+ * over k columns: h = 42; for each col: h = hashLong(col_i, h), NULL
+ * columns leave h unchanged). keys: column-major [ncols][n]. */
+EXPORT void oracle_partition_ids_i64_multi(const int64_t* keys, const uint8_t* validity,
+                                           int32_t ncols, int64_t n,
+                                           int32_t num_parts, int32_t* out_pids) {
+  for (int64_t i = 0; i < n; i++) {
+    int32_t h = 42;
+    for (int32_t c = 0; c < ncols; c++) {
+      const uint8_t* v = validity ? validity + (size_t)c * ((n + 7) / 8) : NULL;
+      if (!v || (v[i >> 3] >> (i & 7)) & 1) {
+        h = mm3_hash_long(keys[(size_t)c * n + i], h);
+      }
+    }
+    out_pids[i] = spark_pmod(h, num_parts);
+  }
+}
+
 /* ------------------------------------------------------------------ */
 /* Sort-prefix encodings — restates                                    */
 /* core/.../unsafe/sort/PrefixComparators.java:66-83 (double) and the  */

@@ -55,3 +55,23 @@ def test_partition_ids_null_passes_seed_through():
     pids = oracle.partition_ids(keys, 8, validity=validity)
     assert pids[0] == oracle.pmod(42, 8)
     assert pids[1] == oracle.pmod(oracle.hash_long(7, 42), 8)
+
+
+def test_partition_ids_multi_column_chaining():
+    # Murmur3Hash(c1, c2, seed 42) chains: h = hashLong(c2, hashLong(c1, 42))
+    # (hash.scala:849-860); NULL columns leave the running hash unchanged.
+    c1 = np.array([1, 2, 3], dtype=np.int64)
+    c2 = np.array([10, 20, 30], dtype=np.int64)
+    pids = oracle.partition_ids_multi([c1, c2], 8)
+    for i in range(3):
+        h = oracle.hash_long(int(c2[i]), oracle.hash_long(int(c1[i]), 42))
+        assert pids[i] == oracle.pmod(h, 8)
+    # single column must agree with the single-column path
+    assert (oracle.partition_ids_multi([c1], 8) == oracle.partition_ids(c1, 8)).all()
+    # a NULL in column 1 skips that link in the chain
+    n = 3
+    vb = np.zeros((2, (n + 7) // 8), dtype=np.uint8)
+    vb[0] = np.packbits([0, 1, 1], bitorder="little")   # c1 row0 NULL
+    vb[1] = np.packbits([1, 1, 1], bitorder="little")
+    pids = oracle.partition_ids_multi([c1, c2], 8, validity=np.ascontiguousarray(vb))
+    assert pids[0] == oracle.pmod(oracle.hash_long(10, 42), 8)
