@@ -170,12 +170,16 @@ int gpuq_hash_agg_partitioned(void* stream, int64_t nrows,
  * HashAggregateExec.scala:68-76 — e.g. TPC-H Q1's 8 aggregates).
  * spec_ops[j]: 0=SUM(float64 col), 1=COUNT(col), 2=COUNT(*),
  * 3=SUM(int64 col) -> int64 (Sum.scala LongType result; non-ansi
- * overflow wraps);
+ * overflow wraps), 4=MIN(int64), 5=MAX(int64), 6=MIN(float64),
+ * 7=MAX(float64) (Min/Max.scala; float ordering = Java Double.compare:
+ * NaN greatest, -0.0 < 0.0);
  * spec_cols[j] indexes vals[] (ignored for COUNT(*)). out_accs[j] is a
- * device array per spec: f64 for SUM, i64 for COUNT. Small tables
- * (cap*(1+nspecs)*8 <= 64 KB) aggregate per-block in LDS first.
- * Round-1: SUM value columns must be non-null (pair with COUNT for NULL
- * tracking). */
+ * device array per spec: f64 for SUM_F64/MIN_F64/MAX_F64, i64 otherwise.
+ * Small tables (cap*(1+nspecs)*8 <= 64 KB) aggregate per-block in LDS
+ * first. Value ops skip NULL rows; an all-NULL group emits the op's
+ * neutral init (0 for SUM/COUNT, extremes for MIN/MAX) — pair with a
+ * COUNT spec to realize SQL NULL results (Sum.scala: NULL iff no
+ * non-null input). */
 int64_t gpuq_hash_agg_multi_workspace_bytes(int64_t capacity, int32_t nspecs);
 int gpuq_hash_agg_multi(void* stream, int64_t nrows, gpuq_col key,
                         const gpuq_col* vals, const int32_t* spec_ops,
@@ -184,6 +188,24 @@ int gpuq_hash_agg_multi(void* stream, int64_t nrows, gpuq_col key,
                         int32_t first_batch, int32_t finalize,
                         int64_t* out_keys, uint8_t* out_key_valid,
                         void* const* out_accs, int64_t* out_ngroups);
+
+/* Composite-key aggregate: GROUP BY (k1..kK), K <= 4 int64 columns with
+ * independent NULLability (the reference groups by an UnsafeRow key tuple,
+ * UnsafeFixedWidthAggregationMap.java:39). Slots claim by CAS + publish;
+ * tuple equality is verified against stored keys, so the full domain is
+ * exact. out_keys[c]: device i64 array per key column; out_kmask: one byte
+ * per group, bit c = key column c non-NULL. Same spec_ops as
+ * gpuq_hash_agg_multi. */
+int64_t gpuq_hash_agg_keys_workspace_bytes(int64_t capacity, int32_t nkeys,
+                                           int32_t nspecs);
+int gpuq_hash_agg_keys(void* stream, int64_t nrows,
+                       const gpuq_col* key_cols, int32_t nkeys,
+                       const gpuq_col* vals, const int32_t* spec_ops,
+                       const int32_t* spec_cols, int32_t nspecs,
+                       void* workspace, int64_t capacity,
+                       int32_t first_batch, int32_t finalize,
+                       void* const* out_keys, uint8_t* out_kmask,
+                       void* const* out_accs, int64_t* out_ngroups);
 
 /* ---------------------------------------------------------------- */
 /* PARTITION — replaces ShuffleExchangeExec's partition-id + write   */
@@ -206,6 +228,16 @@ int gpuq_partition_perm(void* stream, int64_t nrows, gpuq_col key,
                         int32_t num_parts, uint32_t* out_perm,
                         int64_t* out_counts,
                         void* workspace, int64_t workspace_bytes);
+
+/* Multi-column partition keys: pid = Pmod(Murmur3Hash(k1..kK, 42), n)
+ * with the hash seed-chained column-wise and NULL columns passing the
+ * running seed through (hash.scala:849-860 HashExpression.eval;
+ * partitioning.scala:328). nkeys <= 4, each int64. */
+int gpuq_partition_perm_multi(void* stream, int64_t nrows,
+                              const gpuq_col* key_cols, int32_t nkeys,
+                              int32_t num_parts, uint32_t* out_perm,
+                              int64_t* out_counts,
+                              void* workspace, int64_t workspace_bytes);
 
 /* Range partition (global ORDER BY across GPUs): bin = first bound >=
  * key in the sort order (RangePartitioning, ShuffleExchangeExec.scala:
@@ -282,6 +314,50 @@ int gpuq_project_binop(void* stream, int64_t nrows, gpuq_col a,
 
 /* int64 -> float64 cast (AVG evaluation: sum / cast(count)) */
 int gpuq_cast_i64_f64(void* stream, int64_t nrows, const int64_t* in, double* out);
+
+/* ---------------------------------------------------------------- */
+/* VALIDITY BITMAP UTILITIES — move Arrow validity bitmaps           */
+/* (LSB-first, ColumnVector null contract, ColumnVector.java:58-366) */
+/* through permutations and across the RCCL exchange (bitmaps travel */
+/* as u8 columns: all-to-all row splits are not byte-aligned).       */
+/* ---------------------------------------------------------------- */
+
+/* out bit i = src bit perm[i] (permute a validity bitmap alongside its
+ * data column) */
+int gpuq_gather_bits(void* stream, int64_t nrows, const uint8_t* src_bits,
+                     const uint32_t* perm, uint8_t* out_bits);
+
+/* bitmap <-> one-byte-per-row (exchange transport form) */
+int gpuq_bits_to_u8(void* stream, int64_t nrows, const uint8_t* bits,
+                    uint8_t* out);
+int gpuq_u8_to_bits(void* stream, int64_t nrows, const uint8_t* u8,
+                    uint8_t* out_bits);
+
+/* validity bitmap for key column `bit` of a composite-key result:
+ * out bit i = (mask[i] >> bit) & 1 (see gpuq_hash_agg_keys out_kmask) */
+int gpuq_maskbit_to_bits(void* stream, int64_t nrows, const uint8_t* mask,
+                         int32_t bit, uint8_t* out_bits);
+
+/* bit i = (in[i] != 0): merged-COUNT column -> NULL-ness of merged
+ * SUM/MIN/MAX results (final-mode aggregation, Sum.scala merge) */
+int gpuq_nonzero_to_bits(void* stream, int64_t nrows, const int64_t* in,
+                         uint8_t* out_bits);
+
+/* min/max/valid-count reduction of an int64 column. out_dev: device u64[3]
+ * = {min encoded (encode_i64), max encoded, count of valid rows}. */
+int gpuq_minmax_i64(void* stream, int64_t nrows, gpuq_col col,
+                    unsigned long long* out_dev);
+
+/* Pack two narrow int64 key columns into one: out = (a - a_bias) << shift
+ * | (b - b_bias). Caller proves the ranges via gpuq_minmax_i64 first.
+ * Feeds composite GROUP BY keys into the single-key fast path (per-block
+ * LDS tables at low cardinality). */
+int gpuq_pack2_i64(void* stream, int64_t nrows, const int64_t* a,
+                   const int64_t* b, int64_t a_bias, int64_t b_bias,
+                   int32_t shift, int64_t* out);
+int gpuq_unpack2_i64(void* stream, int64_t nrows, const int64_t* in,
+                     int64_t a_bias, int64_t b_bias, int32_t shift,
+                     int64_t* out_a, int64_t* out_b);
 
 #ifdef __cplusplus
 }

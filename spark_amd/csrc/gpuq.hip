@@ -158,6 +158,14 @@ DEV uint64_t encode_f64(double v) {
 /* SignedPrefixComparator order == unsigned order with sign flipped */
 DEV uint64_t encode_i64(int64_t v) { return (uint64_t)v ^ SIGNBIT; }
 
+/* inverses of the monotone encodings (also used by MIN/MAX accumulators,
+ * which hold encoded u64 so atomicMin/atomicMax give the right order) */
+DEV int64_t decode_i64(uint64_t e) { return (int64_t)(e ^ SIGNBIT); }
+DEV double decode_f64(uint64_t e) {
+  uint64_t mask = ((e >> 63) ? 0 : ~0ULL) | SIGNBIT;
+  return __longlong_as_double((long long)(e ^ mask));
+}
+
 DEV bool bit_valid(const uint8_t* validity, int64_t i) {
   return !validity || ((validity[i >> 3] >> (i & 7)) & 1);
 }
@@ -2437,9 +2445,21 @@ extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
 #define AGG_MAX_SPECS 6
 
 struct agg_cols { const double* p[AGG_MAX_SPECS]; };
-struct agg_ops_spec { int op[AGG_MAX_SPECS]; };  /* 0=SUM 1=COUNT(col) 2=COUNT(*) */
+/* ops (Sum.scala:113-180, Count.scala, Min/Max.scala null-skipping update
+ * expressions): 0=SUM_F64 1=COUNT(col) 2=COUNT(*) 3=SUM_I64(wrapping)
+ * 4=MIN_I64 5=MAX_I64 6=MIN_F64 7=MAX_F64. MIN/MAX accumulators hold the
+ * monotone u64 encoding (encode_i64/encode_f64) so u64 atomicMin/atomicMax
+ * realize the reference ordering incl. NaN-greatest and -0.0 < 0.0; the
+ * compact pass decodes. All value ops skip NULL rows; an all-NULL group's
+ * acc stays at its init value — callers discriminate via a paired COUNT
+ * (Sum/Min/Max evaluate to NULL iff no non-null input). */
+struct agg_ops_spec { int op[AGG_MAX_SPECS]; };
 struct agg_valid { const uint8_t* v[AGG_MAX_SPECS]; };
 struct agg_outs { void* p[AGG_MAX_SPECS]; };
+
+DEV unsigned long long aggm_acc_init(int op) {
+  return (op == 4 || op == 6) ? ~0ULL : 0ULL;  /* MIN: u64 max; else 0 */
+}
 
 struct agg_multi_special {
   unsigned long long acc[2][AGG_MAX_SPECS];  /* [0]=key -1 group, [1]=NULL group */
@@ -2472,12 +2492,21 @@ extern "C" int64_t gpuq_hash_agg_multi_workspace_bytes(int64_t cap, int32_t nspe
   return total;
 }
 
-__global__ void k_aggm_init(int64_t cap, int stride, unsigned long long* tab) {
+__global__ void k_aggm_init(int64_t cap, int stride, unsigned long long* tab,
+                            agg_ops_spec ops, agg_multi_special* sp) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t gs = (int64_t)gridDim.x * blockDim.x;
   for (; i < cap; i += gs) {
     tab[(int64_t)stride * i] = AGG_EMPTY;
-    for (int j = 1; j < stride; j++) tab[(int64_t)stride * i + j] = 0;
+    for (int j = 1; j < stride; j++)
+      tab[(int64_t)stride * i + j] = aggm_acc_init(ops.op[j - 1]);
+  }
+  if (sp && blockIdx.x == 0 && threadIdx.x == 0) {
+    sp->seen[0] = sp->seen[1] = 0;
+    sp->out_cursor = 0;
+    sp->overflow = 0;
+    for (int w = 0; w < 2; w++)
+      for (int j = 0; j < stride - 1; j++) sp->acc[w][j] = aggm_acc_init(ops.op[j]);
   }
 }
 
@@ -2485,18 +2514,33 @@ DEV void aggm_update(unsigned long long* acc, int nspecs, const agg_ops_spec ops
                      const agg_cols cols, const agg_valid av, int64_t i) {
   for (int j = 0; j < nspecs; j++) {
     int op = ops.op[j];
+    if (op == 2) { atomicAdd(&acc[j], 1ull); continue; }  /* COUNT(*) */
+    if (!bit_valid(av.v[j], i)) continue;  /* every other op skips NULLs */
     if (op == 0) {
       atomicAdd((double*)&acc[j], cols.p[j][i]);
     } else if (op == 1) {
-      if (bit_valid(av.v[j], i)) atomicAdd(&acc[j], 1ull);
+      atomicAdd(&acc[j], 1ull);
     } else if (op == 3) {
       /* SUM(int64) -> int64 (Sum.scala resultType LongType; non-ansi
        * overflow wraps = two's-complement u64 add, bit-exact) */
       atomicAdd(&acc[j], (unsigned long long)((const int64_t*)cols.p[j])[i]);
-    } else {
-      atomicAdd(&acc[j], 1ull);
+    } else if (op == 4 || op == 5) {
+      unsigned long long e = encode_i64(((const int64_t*)cols.p[j])[i]);
+      if (op == 4) atomicMin(&acc[j], e); else atomicMax(&acc[j], e);
+    } else {  /* 6/7: MIN/MAX f64 via the monotone encoding */
+      unsigned long long e = encode_f64(cols.p[j][i]);
+      if (op == 6) atomicMin(&acc[j], e); else atomicMax(&acc[j], e);
     }
   }
+}
+
+/* merge one accumulator word from a sub-table into the global table */
+DEV void aggm_merge_acc(unsigned long long* dst, int op, unsigned long long v) {
+  if (op == 0)
+    atomicAdd((double*)dst, __longlong_as_double((long long)v));
+  else if (op == 4 || op == 6) atomicMin(dst, v);
+  else if (op == 5 || op == 7) atomicMax(dst, v);
+  else atomicAdd(dst, v);  /* COUNT + SUM_I64 */
 }
 
 template <bool LDS>
@@ -2508,8 +2552,10 @@ void k_aggm_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
   const int stride = 1 + nspecs;
   extern __shared__ __attribute__((aligned(16))) unsigned long long lt[];
   if (LDS) {
-    for (int j = threadIdx.x; j < (int)(cap_mask + 1) * stride; j += blockDim.x)
-      lt[j] = (j % stride == 0) ? AGG_EMPTY : 0;
+    for (int j = threadIdx.x; j < (int)(cap_mask + 1) * stride; j += blockDim.x) {
+      int r = j % stride;
+      lt[j] = (r == 0) ? AGG_EMPTY : aggm_acc_init(ops.op[r - 1]);
+    }
     __syncthreads();
   }
   unsigned long long* t = LDS ? lt : tab;
@@ -2557,16 +2603,18 @@ void k_aggm_build(int64_t n, const int64_t* keys, const uint8_t* kvalid,
         }
         slot = (slot + 1) & (uint64_t)cap_mask;
       }
-      for (int j = 0; j < nspecs; j++) {
-        unsigned long long v = lt[(int64_t)stride * sI + 1 + j];
-        if (ops.op[j] == 0)
-          atomicAdd((double*)&tab[(int64_t)stride * slot + 1 + j],
-                    __longlong_as_double((long long)v));
-        else
-          atomicAdd(&tab[(int64_t)stride * slot + 1 + j], v);  /* COUNT + SUM_I64 */
-      }
+      for (int j = 0; j < nspecs; j++)
+        aggm_merge_acc(&tab[(int64_t)stride * slot + 1 + j], ops.op[j],
+                       lt[(int64_t)stride * sI + 1 + j]);
     }
   }
+}
+
+DEV void aggm_emit(void* out, int64_t o, int op, unsigned long long v) {
+  if (op == 0) ((double*)out)[o] = __longlong_as_double((long long)v);
+  else if (op == 4 || op == 5) ((int64_t*)out)[o] = decode_i64(v);
+  else if (op == 6 || op == 7) ((double*)out)[o] = decode_f64(v);
+  else ((int64_t*)out)[o] = (int64_t)v;
 }
 
 __global__ void k_aggm_compact(int64_t cap, const unsigned long long* tab,
@@ -2603,11 +2651,8 @@ __global__ void k_aggm_compact(int64_t cap, const unsigned long long* tab,
     int64_t i = base + r * 256 + threadIdx.x;
     out_keys[o] = (int64_t)tab[(int64_t)stride * i];
     out_kvalid[o] = 1;
-    for (int j = 0; j < nspecs; j++) {
-      unsigned long long v = tab[(int64_t)stride * i + 1 + j];
-      if (ops.op[j] == 0) ((double*)outs.p[j])[o] = __longlong_as_double((long long)v);
-      else ((int64_t*)outs.p[j])[o] = (int64_t)v;
-    }
+    for (int j = 0; j < nspecs; j++)
+      aggm_emit(outs.p[j], o, ops.op[j], tab[(int64_t)stride * i + 1 + j]);
     o++;
   }
   if (blockIdx.x == 0 && threadIdx.x == 0) {
@@ -2616,11 +2661,8 @@ __global__ void k_aggm_compact(int64_t cap, const unsigned long long* tab,
       int64_t q = (int64_t)atomicAdd(&sp->out_cursor, 1ull);
       out_keys[q] = which == 0 ? -1 : 0;
       out_kvalid[q] = which == 0 ? 1 : 0;
-      for (int j = 0; j < nspecs; j++) {
-        unsigned long long v = sp->acc[which][j];
-        if (ops.op[j] == 0) ((double*)outs.p[j])[q] = __longlong_as_double((long long)v);
-        else ((int64_t*)outs.p[j])[q] = (int64_t)v;
-      }
+      for (int j = 0; j < nspecs; j++)
+        aggm_emit(outs.p[j], q, ops.op[j], sp->acc[which][j]);
     }
   }
 }
@@ -2640,27 +2682,28 @@ extern "C" int gpuq_hash_agg_multi(void* stream, int64_t n, gpuq_col key,
   if (key.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "aggm: key must be int64");
   agg_cols cols = {}; agg_ops_spec ops = {}; agg_valid av = {};
   for (int j = 0; j < nspecs; j++) {
-    ops.op[j] = spec_ops[j];
-    if (spec_ops[j] == 0 || spec_ops[j] == 3) {
+    int op = spec_ops[j];
+    ops.op[j] = op;
+    if (op == 0 || op == 3 || (op >= 4 && op <= 7)) {
       const gpuq_col& c = vals[spec_cols[j]];
-      int want = spec_ops[j] == 0 ? GPUQ_FLOAT64 : GPUQ_INT64;
-      if (c.dtype != want) FAIL(GPUQ_ERR_INVALID, "aggm: SUM col dtype mismatch");
-      if (c.validity) FAIL(GPUQ_ERR_INVALID, "aggm: SUM cols must be non-null "
-                           "(pair with a COUNT spec for NULL tracking)");
+      int want = (op == 0 || op == 6 || op == 7) ? GPUQ_FLOAT64 : GPUQ_INT64;
+      if (c.dtype != want) FAIL(GPUQ_ERR_INVALID, "aggm: value col dtype mismatch op %d", op);
       cols.p[j] = (const double*)c.data;
-    } else if (spec_ops[j] == 1) {
+      av.v[j] = c.validity;  /* ops skip NULL rows; all-NULL groups stay at
+                              * the init acc — callers pair a COUNT spec to
+                              * emit SQL NULL (Sum/Min/Max.scala) */
+    } else if (op == 1) {
       av.v[j] = vals[spec_cols[j]].validity;
-    } else if (spec_ops[j] != 2) {
-      FAIL(GPUQ_ERR_INVALID, "aggm: bad op %d", spec_ops[j]);
+    } else if (op != 2) {
+      FAIL(GPUQ_ERR_INVALID, "aggm: bad op %d", op);
     }
   }
   int stride = 1 + nspecs;
   agg_multi_ws w; int64_t need;
   agg_multi_ws_layout(cap, nspecs, &w, (char*)workspace, &need);
   if (first_batch) {
-    k_aggm_init<<<grid1d(cap), 256, 0, s>>>(cap, stride, w.tab);
+    k_aggm_init<<<grid1d(cap), 256, 0, s>>>(cap, stride, w.tab, ops, w.sp);
     HIP_TRY(hipGetLastError());
-    HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_multi_special), s));
   }
   if (n > 0) {
     int64_t lds_bytes = cap * stride * 8;
@@ -2693,6 +2736,560 @@ extern "C" int gpuq_hash_agg_multi(void* stream, int64_t n, gpuq_col key,
     HIP_TRY(hipStreamSynchronize(s));
     *out_ngroups = (int64_t)hsp2.out_cursor;
   }
+  return GPUQ_OK;
+}
+
+/* ============ composite-key hash aggregate (multi-column GROUP BY) ======== */
+/*
+ * GROUP BY (k1..kK), K <= 4, int64 columns with independent NULLability —
+ * the reference groups by an UnsafeRow of the key tuple
+ * (UnsafeFixedWidthAggregationMap.java:39, TungstenAggregationIterator.scala:206).
+ * Table layout: separate arrays — sig[cap] (u64 claim word), keys[K][cap],
+ * kmask[cap] (bit c = key c non-NULL), acc[cap*nspecs]. A slot is claimed by
+ * CAS(sig: EMPTY->PENDING), the tuple is published (keys + mask), then the
+ * 64-bit signature is released into sig; probers spinning on PENDING retry
+ * from the loop head (single attempt per iteration — wave64 lockstep-safe,
+ * the publisher's branch always executes between iterations). Tuple equality
+ * is verified against the published keys, so signature collisions only cost
+ * extra probes, never correctness. No special slots needed: EMPTY/PENDING
+ * are reserved signature values no tuple maps to (remapped), so the full
+ * int64^K x NULL domain is exact.
+ */
+
+#define AGG_MAX_KEYS 4
+#define AGGK_EMPTY 0ULL
+#define AGGK_PENDING 1ULL
+
+struct agg_keycols { const int64_t* k[AGG_MAX_KEYS]; const uint8_t* v[AGG_MAX_KEYS]; };
+struct gpuq_outcols { void* p[AGG_MAX_KEYS]; };
+
+struct aggk_ws {
+  unsigned long long* sig;   /* [cap] */
+  int64_t* keys;             /* [nkeys][cap] */
+  uint8_t* kmask;            /* [cap] */
+  unsigned long long* acc;   /* [cap][nspecs] */
+  agg_multi_special* sp;     /* out_cursor/overflow only */
+};
+
+static void aggk_ws_layout(int64_t cap, int nkeys, int nspecs, aggk_ws* w,
+                           char* base, int64_t* total) {
+  int64_t off = 0;
+  auto take = [&](int64_t bytes) {
+    char* p = base ? base + off : nullptr;
+    off += (bytes + 255) & ~255LL;
+    return p;
+  };
+  w->sig = (unsigned long long*)take(cap * 8);
+  w->keys = (int64_t*)take(cap * 8 * nkeys);
+  w->kmask = (uint8_t*)take(cap);
+  w->acc = (unsigned long long*)take(cap * 8 * nspecs);
+  w->sp = (agg_multi_special*)take(sizeof(agg_multi_special));
+  *total = off;
+}
+
+extern "C" int64_t gpuq_hash_agg_keys_workspace_bytes(int64_t cap, int32_t nkeys,
+                                                      int32_t nspecs) {
+  aggk_ws w; int64_t total;
+  aggk_ws_layout(cap, nkeys, nspecs, &w, nullptr, &total);
+  return total;
+}
+
+__global__ void k_aggk_init(int64_t cap, int nspecs, agg_ops_spec ops,
+                            unsigned long long* sig, unsigned long long* acc,
+                            agg_multi_special* sp) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < cap; i += gs) {
+    sig[i] = AGGK_EMPTY;
+    for (int j = 0; j < nspecs; j++)
+      acc[(int64_t)nspecs * i + j] = aggm_acc_init(ops.op[j]);
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    sp->out_cursor = 0;
+    sp->overflow = 0;
+  }
+}
+
+/* slot index: the reference's own seed-chained Murmur3 over the key tuple
+ * (hash.scala:849-860 HashExpression.eval: NULL leaves the running hash
+ * unchanged); signature: splitmix64 chain over (value|NULL-token) pairs. */
+DEV void aggk_hash(const agg_keycols kc, int nkeys, int64_t i,
+                   uint32_t* slot_hash, unsigned long long* sig, uint8_t* mask) {
+  uint32_t h = 42;
+  uint64_t s = 0x243F6A8885A308D3ULL;
+  uint8_t m = 0;
+  for (int c = 0; c < nkeys; c++) {
+    bool kv = bit_valid(kc.v[c], i);
+    int64_t k = kv ? kc.k[c][i] : 0;
+    if (kv) { h = (uint32_t)mm3_hash_long(k, (int32_t)h); m |= (uint8_t)(1 << c); }
+    s = splitmix64((s ^ (kv ? (uint64_t)k : 0xD1B54A32D192ED03ULL)) + (uint64_t)c);
+  }
+  if (s == AGGK_EMPTY || s == AGGK_PENDING) s = 2;
+  *slot_hash = h; *sig = s; *mask = m;
+}
+
+__global__ __launch_bounds__(256)
+void k_aggk_build(int64_t n, agg_keycols kc, int nkeys,
+                  agg_cols cols, agg_ops_spec ops, agg_valid av, int nspecs,
+                  unsigned long long* sig, int64_t* tkeys, uint8_t* tkmask,
+                  unsigned long long* acc, agg_multi_special* sp,
+                  int64_t cap, int64_t cap_mask) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += gs) {
+    uint32_t h; unsigned long long mysig; uint8_t m;
+    aggk_hash(kc, nkeys, i, &h, &mysig, &m);
+    uint64_t slot = h & (uint64_t)cap_mask;
+    int64_t probes = 0;
+    for (;;) {
+      unsigned long long cur = __hip_atomic_load(&sig[slot], __ATOMIC_ACQUIRE,
+                                                 __HIP_MEMORY_SCOPE_AGENT);
+      if (cur == AGGK_EMPTY) {
+        unsigned long long prev = atomicCAS(&sig[slot], AGGK_EMPTY, AGGK_PENDING);
+        if (prev == AGGK_EMPTY) {
+          for (int c = 0; c < nkeys; c++)
+            tkeys[(int64_t)c * cap + slot] =
+                bit_valid(kc.v[c], i) ? kc.k[c][i] : 0;
+          tkmask[slot] = m;
+          __hip_atomic_store(&sig[slot], mysig, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_AGENT);
+          break;
+        }
+        continue;  /* lost the claim: reload (sees PENDING or a sig) */
+      }
+      if (cur == AGGK_PENDING) continue;  /* publisher runs between iterations */
+      if (cur == mysig) {
+        bool eq = tkmask[slot] == m;
+        for (int c = 0; eq && c < nkeys; c++)
+          if ((m >> c) & 1)
+            eq = tkeys[(int64_t)c * cap + slot] == kc.k[c][i];
+        if (eq) break;
+      }
+      slot = (slot + 1) & (uint64_t)cap_mask;
+      if (++probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
+    }
+    aggm_update(&acc[(int64_t)nspecs * slot], nspecs, ops, cols, av, i);
+  }
+}
+
+__global__ void k_aggk_compact(int64_t cap, int nkeys, int nspecs,
+                               const unsigned long long* sig, const int64_t* tkeys,
+                               const uint8_t* tkmask, const unsigned long long* acc,
+                               agg_multi_special* sp, agg_ops_spec ops,
+                               gpuq_outcols okeys, uint8_t* out_kmask,
+                               agg_outs outs) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  const int lane = threadIdx.x & (WAVE - 1);
+  for (; i - lane < cap; i += gs) {   /* whole waves iterate together */
+    bool occ = i < cap && sig[i] > AGGK_PENDING;
+    uint64_t mask = __ballot(occ);
+    if (!mask) continue;
+    int cnt = __popcll(mask);
+    unsigned long long base = 0;
+    if (lane == __ffsll((unsigned long long)mask) - 1)
+      base = atomicAdd(&sp->out_cursor, (unsigned long long)cnt);
+    base = __shfl(base, __ffsll((unsigned long long)mask) - 1);
+    if (!occ) continue;
+    int64_t o = (int64_t)base + __popcll(mask & ((1ULL << lane) - 1));
+    for (int c = 0; c < nkeys; c++)
+      ((int64_t*)okeys.p[c])[o] = tkeys[(int64_t)c * cap + i];
+    out_kmask[o] = tkmask[i];
+    for (int j = 0; j < nspecs; j++)
+      aggm_emit(outs.p[j], o, ops.op[j], acc[(int64_t)nspecs * i + j]);
+  }
+}
+
+extern "C" int gpuq_hash_agg_keys(void* stream, int64_t n,
+                                  const gpuq_col* key_cols, int32_t nkeys,
+                                  const gpuq_col* vals, const int32_t* spec_ops,
+                                  const int32_t* spec_cols, int32_t nspecs,
+                                  void* workspace, int64_t cap,
+                                  int32_t first_batch, int32_t finalize,
+                                  void* const* out_keys, uint8_t* out_kmask,
+                                  void* const* out_accs, int64_t* out_ngroups) {
+  hipStream_t s = (hipStream_t)stream;
+  if (cap <= 0 || (cap & (cap - 1)))
+    FAIL(GPUQ_ERR_INVALID, "aggk: capacity %lld not a power of two", (long long)cap);
+  if (nkeys < 1 || nkeys > AGG_MAX_KEYS)
+    FAIL(GPUQ_ERR_INVALID, "aggk: nkeys %d not in [1,%d]", nkeys, AGG_MAX_KEYS);
+  if (nspecs < 1 || nspecs > AGG_MAX_SPECS)
+    FAIL(GPUQ_ERR_INVALID, "aggk: nspecs %d not in [1,%d]", nspecs, AGG_MAX_SPECS);
+  agg_keycols kc = {};
+  for (int c = 0; c < nkeys; c++) {
+    if (key_cols[c].dtype != GPUQ_INT64)
+      FAIL(GPUQ_ERR_INVALID, "aggk: key col %d must be int64", c);
+    kc.k[c] = (const int64_t*)key_cols[c].data;
+    kc.v[c] = key_cols[c].validity;
+  }
+  agg_cols cols = {}; agg_ops_spec ops = {}; agg_valid av = {};
+  for (int j = 0; j < nspecs; j++) {
+    int op = spec_ops[j];
+    ops.op[j] = op;
+    if (op == 0 || op == 3 || (op >= 4 && op <= 7)) {
+      const gpuq_col& c = vals[spec_cols[j]];
+      int want = (op == 0 || op == 6 || op == 7) ? GPUQ_FLOAT64 : GPUQ_INT64;
+      if (c.dtype != want) FAIL(GPUQ_ERR_INVALID, "aggk: value col dtype mismatch op %d", op);
+      cols.p[j] = (const double*)c.data;
+      av.v[j] = c.validity;
+    } else if (op == 1) {
+      av.v[j] = vals[spec_cols[j]].validity;
+    } else if (op != 2) {
+      FAIL(GPUQ_ERR_INVALID, "aggk: bad op %d", op);
+    }
+  }
+  aggk_ws w; int64_t need;
+  aggk_ws_layout(cap, nkeys, nspecs, &w, (char*)workspace, &need);
+  if (first_batch) {
+    k_aggk_init<<<grid1d(cap), 256, 0, s>>>(cap, nspecs, ops, w.sig, w.acc, w.sp);
+    HIP_TRY(hipGetLastError());
+  }
+  if (n > 0) {
+    { hipEvent_t _pe = prof_begin(s);
+    k_aggk_build<<<hash_grid(n), 256, 0, s>>>(
+        n, kc, nkeys, cols, ops, av, nspecs, w.sig, w.keys, w.kmask, w.acc,
+        w.sp, cap, cap - 1);
+    prof_end("aggk_build", s, _pe); }
+    HIP_TRY(hipGetLastError());
+  }
+  if (finalize) {
+    agg_multi_special hsp;
+    HIP_TRY(hipMemcpyAsync(&hsp, w.sp, sizeof(hsp), hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "aggk: hash table overflow (capacity %lld)", (long long)cap);
+    gpuq_outcols okeys = {};
+    for (int c = 0; c < nkeys; c++) okeys.p[c] = (void*)out_keys[c];
+    agg_outs outs = {};
+    for (int j = 0; j < nspecs; j++) outs.p[j] = (void*)out_accs[j];
+    k_aggk_compact<<<grid1d(cap), 256, 0, s>>>(cap, nkeys, nspecs, w.sig, w.keys,
+                                               w.kmask, w.acc, w.sp, ops, okeys,
+                                               out_kmask, outs);
+    HIP_TRY(hipGetLastError());
+    agg_multi_special hsp2;
+    HIP_TRY(hipMemcpyAsync(&hsp2, w.sp, sizeof(hsp2), hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    *out_ngroups = (int64_t)hsp2.out_cursor;
+  }
+  return GPUQ_OK;
+}
+
+/* ============ multi-column partition ids (seed-chained Murmur3) =========== */
+/* pid = Pmod(Murmur3Hash(k1..kK, 42), n) with the hash chained column-wise
+ * and NULL columns passing the running seed through unchanged —
+ * HashPartitioning.partitionIdExpression (partitioning.scala:328) over
+ * HashExpression.eval (hash.scala:849-860). */
+
+__global__ void k_partition_pids_multi(int64_t n, agg_keycols kc, int nkeys,
+                                       int32_t nparts, uint64_t* pid_as_key,
+                                       uint32_t* idx,
+                                       unsigned long long* counts) {
+  __shared__ uint32_t h[256];
+  bool lds_counts = nparts <= 256;
+  if (lds_counts)
+    for (int b = threadIdx.x; b < nparts; b += blockDim.x) h[b] = 0;
+  __syncthreads();
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int32_t hsh = 42;
+    for (int c = 0; c < nkeys; c++)
+      if (bit_valid(kc.v[c], i)) hsh = mm3_hash_long(kc.k[c][i], hsh);
+    int pid = spark_pmod(hsh, nparts);
+    pid_as_key[i] = (uint64_t)pid;
+    idx[i] = (uint32_t)i;
+    if (lds_counts) atomicAdd(&h[pid], 1u);
+    else atomicAdd(&counts[pid], 1ull);
+  }
+  __syncthreads();
+  if (lds_counts)
+    for (int b = threadIdx.x; b < nparts; b += blockDim.x)
+      if (h[b]) atomicAdd(&counts[b], (unsigned long long)h[b]);
+}
+
+/* shared tail of the partition entry points: stable radix over the pid */
+static int partition_scatter_tail(hipStream_t s, int64_t n, sort_ws& w,
+                                  int32_t nparts, uint32_t* out_perm) {
+  scatter_geom geom = get_sort_geom();
+  int tile = geom.block * geom.items;
+  int64_t nb = sort_nblocks(n, tile);
+  int passes = nparts > 256 ? 2 : 1;
+  uint64_t *kin = w.ka, *kout = w.kb;
+  uint32_t *iin = w.ia, *iout = w.ib;
+  for (int p = 0; p < passes; p++) {
+    { hipEvent_t _pe = prof_begin(s);
+    k_radix_hist<0><<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, p * 8, w.hist, (int)nb, tile);
+    prof_end("radix_hist", s, _pe); }
+    HIP_TRY(hipGetLastError());
+    int rc = exclusive_scan_u32(s, (int64_t)256 * nb, w.hist, w.hist_scan, w.block_sums);
+    if (rc) return rc;
+    uint32_t* iout_pass = (p == passes - 1) ? out_perm : iout;
+    { hipEvent_t _pe = prof_begin(s);
+    launch_scatter<0, false>(s, geom, nb, n, kin, iin, kout, iout_pass, w.hist_scan,
+                             p * 8, 0, nullptr, nullptr, nullptr, 0);
+    prof_end("radix_scatter", s, _pe); }
+    HIP_TRY(hipGetLastError());
+    uint64_t* tk = kin; kin = kout; kout = tk;
+    uint32_t* ti = iin; iin = iout_pass; iout = ti;
+  }
+  return GPUQ_OK;
+}
+
+extern "C" int gpuq_partition_perm_multi(void* stream, int64_t n,
+                                         const gpuq_col* key_cols, int32_t nkeys,
+                                         int32_t nparts, uint32_t* out_perm,
+                                         int64_t* out_counts,
+                                         void* workspace, int64_t workspace_bytes) {
+  hipStream_t s = (hipStream_t)stream;
+  if (n > 0xFFFFFFFFLL) FAIL(GPUQ_ERR_INVALID, "partition: nrows %lld > 2^32", (long long)n);
+  if (nparts < 1 || nparts > 65536)
+    FAIL(GPUQ_ERR_INVALID, "partition: num_parts %d not in [1,65536]", nparts);
+  if (nkeys < 1 || nkeys > AGG_MAX_KEYS)
+    FAIL(GPUQ_ERR_INVALID, "partition: nkeys %d not in [1,%d]", nkeys, AGG_MAX_KEYS);
+  agg_keycols kc = {};
+  for (int c = 0; c < nkeys; c++) {
+    if (key_cols[c].dtype != GPUQ_INT64)
+      FAIL(GPUQ_ERR_INVALID, "partition: key col %d must be int64", c);
+    kc.k[c] = (const int64_t*)key_cols[c].data;
+    kc.v[c] = key_cols[c].validity;
+  }
+  sort_ws w; int64_t need;
+  sort_ws_layout(n, 256, &w, (char*)workspace, &need);
+  if (workspace_bytes < need)
+    FAIL(GPUQ_ERR_INVALID, "partition: workspace %lld < %lld", (long long)workspace_bytes, (long long)need);
+  HIP_TRY(hipMemsetAsync(out_counts, 0, (size_t)nparts * 8, s));
+  if (n == 0) return GPUQ_OK;
+  { hipEvent_t _pe = prof_begin(s);
+  k_partition_pids_multi<<<grid1d(n), 256, 0, s>>>(n, kc, nkeys, nparts, w.ka, w.ia,
+                                                   (unsigned long long*)out_counts);
+  prof_end("partition_pids", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  return partition_scatter_tail(s, n, w, nparts, out_perm);
+}
+
+/* ================= validity-bitmap utilities ================= */
+/* The ColumnVector contract carries NULLs everywhere (ColumnVector.java:
+ * 58-366); these kernels move Arrow validity bitmaps (LSB-first) through
+ * permutations and across the exchange (bitmaps travel as u8 columns in
+ * the all-to-all because row split points are not byte-aligned). */
+
+__global__ void k_gather_bits(int64_t n, const uint8_t* src, const uint32_t* perm,
+                              uint8_t* out) {
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;   /* output byte */
+  int64_t nbytes = (n + 7) >> 3;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; b < nbytes; b += gs) {
+    uint8_t v = 0;
+    int64_t row0 = b << 3;
+    int top = (int)((n - row0) < 8 ? (n - row0) : 8);
+    for (int t = 0; t < top; t++) {
+      uint32_t p = perm[row0 + t];
+      v |= (uint8_t)(((src[p >> 3] >> (p & 7)) & 1) << t);
+    }
+    out[b] = v;
+  }
+}
+
+extern "C" int gpuq_gather_bits(void* stream, int64_t nrows, const uint8_t* src_bits,
+                                const uint32_t* perm, uint8_t* out_bits) {
+  hipStream_t s = (hipStream_t)stream;
+  if (nrows == 0) return GPUQ_OK;
+  int64_t nbytes = (nrows + 7) >> 3;
+  k_gather_bits<<<grid1d(nbytes), 256, 0, s>>>(nrows, src_bits, perm, out_bits);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+__global__ void k_bits_to_u8(int64_t n, const uint8_t* bits, uint8_t* out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += gs) out[i] = (bits[i >> 3] >> (i & 7)) & 1;
+}
+
+extern "C" int gpuq_bits_to_u8(void* stream, int64_t nrows, const uint8_t* bits,
+                               uint8_t* out) {
+  hipStream_t s = (hipStream_t)stream;
+  if (nrows == 0) return GPUQ_OK;
+  k_bits_to_u8<<<grid1d(nrows), 256, 0, s>>>(nrows, bits, out);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+__global__ void k_u8_to_bits(int64_t n, const uint8_t* u8, uint8_t* bits) {
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nbytes = (n + 7) >> 3;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; b < nbytes; b += gs) {
+    uint8_t v = 0;
+    int64_t row0 = b << 3;
+    int top = (int)((n - row0) < 8 ? (n - row0) : 8);
+    for (int t = 0; t < top; t++) v |= (uint8_t)((u8[row0 + t] & 1) << t);
+    bits[b] = v;
+  }
+}
+
+extern "C" int gpuq_u8_to_bits(void* stream, int64_t nrows, const uint8_t* u8,
+                               uint8_t* out_bits) {
+  hipStream_t s = (hipStream_t)stream;
+  if (nrows == 0) return GPUQ_OK;
+  int64_t nbytes = (nrows + 7) >> 3;
+  k_u8_to_bits<<<grid1d(nbytes), 256, 0, s>>>(nrows, u8, out_bits);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+__global__ void k_nonzero_to_bits(int64_t n, const int64_t* in, uint8_t* bits) {
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nbytes = (n + 7) >> 3;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; b < nbytes; b += gs) {
+    uint8_t v = 0;
+    int64_t row0 = b << 3;
+    int top = (int)((n - row0) < 8 ? (n - row0) : 8);
+    for (int t = 0; t < top; t++) v |= (uint8_t)((in[row0 + t] != 0 ? 1 : 0) << t);
+    bits[b] = v;
+  }
+}
+
+/* validity bitmap from an int64 column: bit i = (in[i] != 0). Used to turn
+ * merged partial COUNTs into the NULL-ness of merged SUM/MIN/MAX results
+ * (Sum.scala: result is NULL iff no non-null input). */
+extern "C" int gpuq_nonzero_to_bits(void* stream, int64_t nrows, const int64_t* in,
+                                    uint8_t* out_bits) {
+  hipStream_t s = (hipStream_t)stream;
+  if (nrows == 0) return GPUQ_OK;
+  int64_t nbytes = (nrows + 7) >> 3;
+  k_nonzero_to_bits<<<grid1d(nbytes), 256, 0, s>>>(nrows, in, out_bits);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+__global__ void k_maskbit_to_bits(int64_t n, const uint8_t* mask, int bit,
+                                  uint8_t* bits) {
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nbytes = (n + 7) >> 3;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; b < nbytes; b += gs) {
+    uint8_t v = 0;
+    int64_t row0 = b << 3;
+    int top = (int)((n - row0) < 8 ? (n - row0) : 8);
+    for (int t = 0; t < top; t++)
+      v |= (uint8_t)(((mask[row0 + t] >> bit) & 1) << t);
+    bits[b] = v;
+  }
+}
+
+/* validity bitmap for key column `bit` from gpuq_hash_agg_keys' out_kmask */
+extern "C" int gpuq_maskbit_to_bits(void* stream, int64_t nrows,
+                                    const uint8_t* mask, int32_t bit,
+                                    uint8_t* out_bits) {
+  hipStream_t s = (hipStream_t)stream;
+  if (bit < 0 || bit > 7) FAIL(GPUQ_ERR_INVALID, "maskbit: bit %d", bit);
+  if (nrows == 0) return GPUQ_OK;
+  int64_t nbytes = (nrows + 7) >> 3;
+  k_maskbit_to_bits<<<grid1d(nbytes), 256, 0, s>>>(nrows, mask, bit, out_bits);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+/* ================= min/max reduction (int64 column) ================= */
+/* out_dev[0] = min (encoded u64), out_dev[1] = max (encoded), out_dev[2] =
+ * count of valid rows. Used by the host packing rule (narrow composite keys
+ * -> one i64) and for range sanity checks. */
+
+__global__ void k_minmax_i64(int64_t n, const int64_t* data, const uint8_t* validity,
+                             unsigned long long* out) {
+  __shared__ unsigned long long smin[256], smax[256];
+  __shared__ unsigned long long scnt;
+  if (threadIdx.x == 0) scnt = 0;
+  unsigned long long lmin = ~0ULL, lmax = 0, lcnt = 0;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += gs) {
+    if (!bit_valid(validity, i)) continue;
+    unsigned long long e = encode_i64(data[i]);
+    if (e < lmin) lmin = e;
+    if (e > lmax) lmax = e;
+    lcnt++;
+  }
+  smin[threadIdx.x] = lmin; smax[threadIdx.x] = lmax;
+  __syncthreads();
+  if (lcnt) atomicAdd(&scnt, lcnt);
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) {
+      if (smin[threadIdx.x + w] < smin[threadIdx.x]) smin[threadIdx.x] = smin[threadIdx.x + w];
+      if (smax[threadIdx.x + w] > smax[threadIdx.x]) smax[threadIdx.x] = smax[threadIdx.x + w];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    atomicMin(&out[0], smin[0]);
+    atomicMax(&out[1], smax[0]);
+    if (scnt) atomicAdd(&out[2], scnt);
+  }
+}
+
+extern "C" int gpuq_minmax_i64(void* stream, int64_t nrows, gpuq_col col,
+                               unsigned long long* out_dev /* [3] */) {
+  hipStream_t s = (hipStream_t)stream;
+  if (col.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "minmax: col must be int64");
+  HIP_TRY(hipMemsetAsync(out_dev, 0xFF, 8, s));       /* min := u64 max */
+  HIP_TRY(hipMemsetAsync(out_dev + 1, 0, 16, s));     /* max := 0, cnt := 0 */
+  if (nrows == 0) return GPUQ_OK;
+  k_minmax_i64<<<grid1d(nrows), 256, 0, s>>>(nrows, (const int64_t*)col.data,
+                                             col.validity, out_dev);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+/* ================= narrow-key pack/unpack ================= */
+/* Pack two int64 key columns with known small ranges into one int64:
+ * out = (a - a_bias) << shift | (b - b_bias). The host rule verifies
+ * 0 <= a-a_bias < 2^(63-shift) and 0 <= b-b_bias < 2^shift via
+ * gpuq_minmax_i64 first; the packed key feeds the fast single-key
+ * aggregation path (LDS tables at low cardinality), then unpack restores
+ * the tuple. NULLs must be handled by the caller (packing is only applied
+ * to non-null key columns). */
+
+__global__ void k_pack2_i64(int64_t n, const int64_t* a, const int64_t* b,
+                            int64_t a_bias, int64_t b_bias, int shift,
+                            int64_t* out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += gs)
+    out[i] = ((a[i] - a_bias) << shift) | (b[i] - b_bias);
+}
+
+extern "C" int gpuq_pack2_i64(void* stream, int64_t nrows, const int64_t* a,
+                              const int64_t* b, int64_t a_bias, int64_t b_bias,
+                              int32_t shift, int64_t* out) {
+  hipStream_t s = (hipStream_t)stream;
+  if (shift < 1 || shift > 62) FAIL(GPUQ_ERR_INVALID, "pack2: shift %d", shift);
+  if (nrows == 0) return GPUQ_OK;
+  k_pack2_i64<<<grid1d(nrows), 256, 0, s>>>(nrows, a, b, a_bias, b_bias, shift, out);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+__global__ void k_unpack2_i64(int64_t n, const int64_t* in, int64_t a_bias,
+                              int64_t b_bias, int shift, int64_t* a, int64_t* b) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += gs) {
+    int64_t v = in[i];
+    if (a) a[i] = (v >> shift) + a_bias;
+    if (b) b[i] = (v & (((int64_t)1 << shift) - 1)) + b_bias;
+  }
+}
+
+extern "C" int gpuq_unpack2_i64(void* stream, int64_t nrows, const int64_t* in,
+                                int64_t a_bias, int64_t b_bias, int32_t shift,
+                                int64_t* out_a, int64_t* out_b) {
+  hipStream_t s = (hipStream_t)stream;
+  if (shift < 1 || shift > 62) FAIL(GPUQ_ERR_INVALID, "unpack2: shift %d", shift);
+  if (nrows == 0) return GPUQ_OK;
+  k_unpack2_i64<<<grid1d(nrows), 256, 0, s>>>(nrows, in, a_bias, b_bias, shift,
+                                              out_a, out_b);
+  HIP_TRY(hipGetLastError());
   return GPUQ_OK;
 }
 

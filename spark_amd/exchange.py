@@ -45,23 +45,51 @@ def exchange_columns(cols: Dict[str, torch.Tensor], in_splits: List[int],
     return out, out_splits
 
 
+def _pack_validity_wire(gpuq, cols, validity, perm, n):
+    """Permute validity bitmaps and add them as u8 wire columns (bitmaps
+    can't ride the all-to-all directly: split points aren't byte-aligned)."""
+    for name, v in (validity or {}).items():
+        if v is None:
+            continue
+        pv = gpuq.gather_bits(v, perm)
+        cols[f"__valid__{name}"] = gpuq.bits_to_u8(pv, n)
+
+
+def _unpack_validity_wire(gpuq, out):
+    validity = {}
+    for name in list(out):
+        if name.startswith("__valid__"):
+            u8 = out.pop(name)
+            validity[name[len("__valid__"):]] = (
+                gpuq.u8_to_bits(u8) if u8.numel() else None)
+    return {k: v for k, v in validity.items() if v is not None}
+
+
 def shuffle_exchange_gpu(key: torch.Tensor, payload: Dict[str, torch.Tensor],
-                         group=None):
+                         group=None, key_validity=None, validity=None):
     """Full GPU exchange: partition on-device (gpuq) + RCCL all-to-all.
 
     num_partitions = world size (one partition per GPU, SURVEY §2 analog of
-    one task per partition). Returns (key, payload) columns now holding this
-    rank's partition. Fails loudly if the HIP engine is missing."""
+    one task per partition). Returns (key, payload, validity) columns now
+    holding this rank's partition; validity maps column names (and
+    "__key__") to received bitmaps. Fails loudly if the HIP engine is
+    missing."""
     from . import gpuq
     world = dist.get_world_size(group)
-    perm, counts = gpuq.partition_perm(key, world)
+    perm, counts = gpuq.partition_perm(key, world, key_validity=key_validity)
+    n = key.numel()
     cols = {"__key__": gpuq.gather(key, perm)}
     for name, t in payload.items():
         cols[name] = gpuq.gather(t, perm)
+    wire_validity = dict(validity or {})
+    if key_validity is not None:
+        wire_validity["__key__"] = key_validity
+    _pack_validity_wire(gpuq, cols, wire_validity, perm, n)
     in_splits = counts.cpu().tolist()
     out, _ = exchange_columns(cols, in_splits, group=group)
+    vout = _unpack_validity_wire(gpuq, out)
     k = out.pop("__key__")
-    return k, out
+    return k, out, vout
 
 
 def broadcast_gather(cols: Dict[str, torch.Tensor], group=None) -> Dict[str, torch.Tensor]:
@@ -91,28 +119,66 @@ def broadcast_gather(cols: Dict[str, torch.Tensor], group=None) -> Dict[str, tor
 
 
 def range_exchange(key: torch.Tensor, payload: Dict[str, torch.Tensor],
-                   desc=False, samples_per_rank: int = 4096, group=None):
+                   desc=False, nulls_first=None, samples_per_rank: int = 4096,
+                   group=None, key_validity=None, validity=None,
+                   key_name: str = "__key__"):
     """Global ORDER BY exchange (the RangePartitioning path,
     ShuffleExchangeExec.scala:379-400): sample keys, all-gather the samples,
     take quantile bounds, range-partition on device, all-to-all. After a
     local sort on each rank, rank-major order is the global sort order.
-    Returns (key, payload) holding this rank's range."""
+    NULL keys land on the first (nulls_first) / last rank, matching
+    SortOrder null placement. Returns (key, payload, validity) holding this
+    rank's range; validity is keyed by payload names + key_name."""
     from . import gpuq
     world = dist.get_world_size(group)
     n = key.numel()
-    # evenly-strided local sample, sorted on device with the GPU sort
+    if nulls_first is None:
+        nulls_first = not desc
+    # evenly-strided local sample, sorted on device with the GPU sort.
+    # NULL sample entries are fine: the bound picker sorts with the same
+    # null placement and NULL keys bypass the bounds in k_range_pids.
     stride = max(1, n // samples_per_rank)
     idx = torch.arange(0, n, stride, dtype=torch.int32, device=key.device)
     local_sample = gpuq.gather(key, idx)
-    gathered = broadcast_gather({"s": local_sample}, group=group)["s"]
-    _, sorted_samples = gpuq.sort_perm(gathered, desc=desc)
+    sample_valid = (gpuq.bits_to_u8(gpuq.gather_bits(key_validity, idx),
+                                    idx.numel())
+                    if key_validity is not None else None)
+    gcols = {"s": local_sample}
+    if sample_valid is not None:
+        gcols["sv"] = sample_valid
+    gathered = broadcast_gather(gcols, group=group)
+    gs = gathered["s"]
+    gsv = gpuq.u8_to_bits(gathered["sv"]) if sample_valid is not None else None
+    sperm, sorted_samples = gpuq.sort_perm(gs, desc=desc,
+                                           nulls_first=nulls_first,
+                                           key_validity=gsv)
     m = sorted_samples.numel()
-    bidx = torch.arange(1, world, dtype=torch.int32, device=key.device) * (m // world)
-    bounds = gpuq.gather(sorted_samples, bidx.clamp(max=m - 1))
-    perm, counts = gpuq.range_partition_perm(key, bounds, desc=desc)
+    # bounds are quantiles of the VALID sorted samples; NULL samples cluster
+    # at one end (SortOrder placement) and NULL keys bypass the bounds in
+    # k_range_pids, landing on the first/last partition directly
+    if gsv is not None:
+        nvalid = int(gpuq.bits_to_u8(gsv, m).sum().item())  # metadata-sized
+        base = m - nvalid if nulls_first else 0
+    else:
+        nvalid, base = m, 0
+    nvalid = max(nvalid, 1)
+    bidx = base + torch.arange(1, world, dtype=torch.int64,
+                               device=key.device) * (nvalid // world)
+    bidx = bidx.clamp(max=m - 1).to(torch.int32)
+    bounds = gpuq.gather(sorted_samples, bidx)
+    perm, counts = gpuq.range_partition_perm(key, bounds, desc=desc,
+                                             nulls_first=nulls_first,
+                                             key_validity=key_validity)
     cols = {"__key__": gpuq.gather(key, perm)}
     for name, t in payload.items():
         cols[name] = gpuq.gather(t, perm)
+    wire_validity = dict(validity or {})
+    if key_validity is not None:
+        wire_validity["__key__"] = key_validity
+    _pack_validity_wire(gpuq, cols, wire_validity, perm, n)
     out, _ = exchange_columns(cols, counts.cpu().tolist(), group=group)
+    vout = _unpack_validity_wire(gpuq, out)
     k = out.pop("__key__")
-    return k, out
+    if "__key__" in vout:
+        vout[key_name] = vout.pop("__key__")
+    return k, out, vout

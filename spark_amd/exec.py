@@ -78,6 +78,14 @@ class SparkPlan:
     def required_child_distribution(self) -> List[Distribution]:
         return [Distribution("unspecified") for _ in self.children]
 
+    @property
+    def output_ordering(self) -> list:
+        """outputOrdering (SparkPlan.scala:180 contract): the SortOrders
+        this node's output satisfies. A replacement node must deliver AT
+        LEAST what the node it replaces advertised — parents may have had
+        their sorts elided against it (EnsureRequirements)."""
+        return []
+
     def execute_columnar(self):
         """doExecuteColumnar (SparkPlan.scala:359)."""
         raise NotImplementedError(
@@ -115,17 +123,28 @@ class SortExec(_CpuNode):
     def output(self):
         return self.children[0].output
 
+    @property
+    def output_ordering(self):
+        return [self.sort_order]
+
 
 class HashAggregateExec(_CpuNode):
-    def __init__(self, group_key: str, aggs: List[Tuple[str, str]], mode: str, child,
+    """group_key: one column name or a tuple of names (composite GROUP BY,
+    the UnsafeRow key-tuple case)."""
+
+    def __init__(self, group_key, aggs: List[Tuple[str, str]], mode: str, child,
                  capacity: Optional[int] = None):
         super().__init__(child)
         self.group_key, self.aggs, self.mode = group_key, aggs, mode
         self.capacity = capacity
 
     @property
+    def group_keys(self):
+        return (self.group_key,) if isinstance(self.group_key, str) else tuple(self.group_key)
+
+    @property
     def output(self):
-        return [self.group_key] + [f"{fn}({col})" for fn, col in self.aggs]
+        return list(self.group_keys) + [f"{fn}({col})" for fn, col in self.aggs]
 
 
 class ShuffledHashJoinExec(_CpuNode):
@@ -153,6 +172,12 @@ class SortMergeJoinExec(_CpuNode):
     @property
     def output(self):
         return self.children[0].output + self.children[1].output
+
+    @property
+    def output_ordering(self):
+        # SMJ's output is ordered by the streamed-side join keys
+        # (SortMergeJoinExec.scala:83 outputOrdering via getKeyOrdering)
+        return [SortOrder(self.left_key)]
 
 
 class ShuffleExchangeExec(_CpuNode):
@@ -258,6 +283,10 @@ class GpuSortExec(SparkPlan):
         return self.children[0].output
 
     @property
+    def output_ordering(self):
+        return [self.sort_order]
+
+    @property
     def supports_columnar(self):
         return True
 
@@ -277,44 +306,96 @@ class GpuSortExec(SparkPlan):
                 # RangePartitioning insertion, exchange/EnsureRequirements.scala:296);
                 # after the local sort below, rank-major order is global order
                 from .exchange import range_exchange
-                assert batch.validity(o.key) is None, (
-                    "global ORDER BY with NULL keys across ranks: validity "
-                    "does not travel through the exchange yet (round-2)")
                 payload = {n_: t for n_, t in batch.columns().items()
                            if n_ != o.key}
-                k, payload = range_exchange(batch.column(o.key), payload,
-                                            desc=o.descending)
+                validity = {n_: batch.validity(n_) for n_ in batch.columns()
+                            if batch.validity(n_) is not None}
+                k, payload, validity = range_exchange(
+                    batch.column(o.key), payload, desc=o.descending,
+                    nulls_first=o.nulls_first,
+                    key_validity=batch.validity(o.key),
+                    validity={n_: v for n_, v in validity.items()
+                              if n_ != o.key},
+                    key_name=o.key)
                 cols = {o.key: k}
                 cols.update(payload)
                 batch.close()
-                batch = ColumnarBatch(cols)
+                batch = ColumnarBatch(cols, validity=validity or None)
             keys = batch.column(o.key)
+            kvalid = batch.validity(o.key)
             perm, skeys = gpuq.sort_perm(keys, desc=o.descending,
                                          nulls_first=o.nulls_first,
-                                         key_validity=batch.validity(o.key))
+                                         key_validity=kvalid)
             cols = {o.key: skeys}
+            validity = {}
+            if kvalid is not None:
+                validity[o.key] = gpuq.gather_bits(kvalid, perm)
             for name, t in batch.columns().items():
                 if name != o.key:
                     cols[name] = gpuq.gather(t, perm)
+                    v = batch.validity(name)
+                    if v is not None:
+                        validity[name] = gpuq.gather_bits(v, perm)
             batch.close()
-            yield ColumnarBatch(cols)
+            yield ColumnarBatch(cols, validity=validity or None)
 
 
 class GpuHashAggregateExec(SparkPlan):
     """Replaces HashAggregateExec (HashAggregateExec.scala:99-151).
-    mode: "partial" | "final" | "complete" (AggUtils.scala:126-195 split);
-    "complete" = single-node partial+final in one table."""
+    mode: "partial" | "final" | "complete" (AggUtils.scala:126-195 split).
 
-    def __init__(self, group_key: str, aggs: List[Tuple[str, str]], mode: str,
+    Partial mode outputs the aggregation BUFFER columns (the
+    aggBufferAttributes analog, e.g. Average.scala's (sum, count) pair;
+    Sum/Min/Max carry a count companion for NULL-ness), which travel
+    through the exchange; final mode merges them exactly: partial sums
+    merge with the same-dtype SUM op (int64 counts merge as wrapping i64
+    adds, never as f64 — Sum.scala mergeExpressions / Count.scala), and
+    partial MIN/MAX merge with MIN/MAX over non-NULL partials. A result is
+    NULL iff the merged count of non-null inputs is 0.
+
+    group_key: one name or a tuple (composite GROUP BY via
+    gpuq_hash_agg_keys; the reference's UnsafeRow key-tuple path,
+    UnsafeFixedWidthAggregationMap.java:39)."""
+
+    def __init__(self, group_key, aggs: List[Tuple[str, str]], mode: str,
                  child, capacity: Optional[int] = None):
         super().__init__(child)
         assert mode in ("partial", "final", "complete")
         self.group_key, self.aggs, self.mode = group_key, aggs, mode
         self.capacity = capacity
+        fns = {fn for fn, _ in self.aggs}
+        assert fns <= {"sum", "count", "count*", "avg", "min", "max"}, \
+            f"unsupported aggs {fns}"
+
+    @property
+    def group_keys(self):
+        if self.group_key is None:
+            return ()
+        return (self.group_key,) if isinstance(self.group_key, str) else tuple(self.group_key)
+
+    def _buffer_cols(self, fn: str, col: str) -> List[str]:
+        """agg buffer column names (partial-mode output schema)."""
+        if fn == "sum":
+            return [f"sum({col})", f"count({col})"]
+        if fn == "count":
+            return [f"count({col})"]
+        if fn == "count*":
+            return ["count(1)"]
+        if fn == "avg":
+            return [f"avg_sum({col})", f"count({col})"]
+        return [f"{fn}({col})", f"count({col})"]  # min/max
 
     @property
     def output(self):
-        return [self.group_key] + [f"{fn}({col})" for fn, col in self.aggs]
+        keys = list(self.group_keys)
+        if self.mode == "partial":
+            seen, bufs = set(), []
+            for fn, col in self.aggs:
+                for b in self._buffer_cols(fn, col):
+                    if b not in seen:
+                        seen.add(b); bufs.append(b)
+            return keys + bufs
+        return keys + [f"{fn}({col})" for fn, col in self.aggs]
 
     @property
     def supports_columnar(self):
@@ -322,83 +403,164 @@ class GpuHashAggregateExec(SparkPlan):
 
     def required_child_distribution(self):
         if self.mode == "final":
-            return [Distribution("clustered", (self.group_key,))]
+            return [Distribution("clustered", self.group_keys)]
         return [Distribution("unspecified")]
 
     def execute_columnar(self):
-        from . import gpuq
-        fns = {fn for fn, _ in self.aggs}
-        assert fns <= {"sum", "count", "avg"}, f"unsupported aggs {fns}"
-        cols_used = {col for _, col in self.aggs}
-        if len(cols_used) > 1 or len(self.aggs) > 2:
-            yield from self._execute_multi(gpuq)
-            return
-        val_col = next(col for fn, col in self.aggs)
-        ops = 0
-        if fns & {"sum", "avg"} or self.mode == "final":
-            ops |= gpuq.AGG_SUM
-        if fns & {"count", "avg"}:
-            ops |= gpuq.AGG_COUNT
         for batch in self.children[0].execute_columnar():
-            keys = batch.column(self.group_key)
-            vals = batch.column(val_col)
-            if vals.dtype == torch.int64:
-                # COUNT over any column / final-mode merge of partial counts:
-                # values ride as f64 (exact below 2^53); SUM(int64)'s int64
-                # result type is a round-2 item (Sum.scala resultType)
-                vals = gpuq.cast_i64_f64(vals)
-            n = keys.numel()
-            cap = self.capacity or (1 << max(10, int(n).bit_length()))
-            out = gpuq.hash_agg(keys, vals, cap, ops=ops or gpuq.AGG_SUM,
-                                key_validity=batch.validity(self.group_key),
-                                val_validity=batch.validity(val_col))
-            ok, okv, osum, osv, ocnt = out
-            cols = {self.group_key: ok}
+            yield self._agg_batch(batch)
+
+    # ---- spec construction ----
+
+    def _input_specs(self, batch):
+        """specs over RAW input rows (partial/complete) + per-buffer slots.
+        Returns (specs, buffer_slot: name -> spec index)."""
+        from . import gpuq
+        specs, slot = [], {}
+
+        def add(name, spec):
+            if name not in slot:
+                slot[name] = len(specs)
+                specs.append(spec)
+
+        for fn, col in self.aggs:
+            if fn == "count*":
+                add("count(1)", ("count*",))
+                continue
+            t = batch.column(col)
+            v = batch.validity(col)
+            if fn == "avg":
+                ts = gpuq.cast_i64_f64(t) if t.dtype == torch.int64 else t
+                add(f"avg_sum({col})", ("sum", ts, v))
+                add(f"count({col})", ("count", t, v))
+            elif fn == "count":
+                add(f"count({col})", ("count", t, v))
+            else:  # sum/min/max + count companion for NULL-ness
+                add(f"{fn}({col})", (fn, t, v))
+                add(f"count({col})", ("count", t, v))
+        return specs, slot
+
+    def _merge_specs(self, batch):
+        """specs over PARTIAL BUFFER columns (final mode): same-dtype SUM
+        for sums/counts, MIN/MAX over non-NULL partials."""
+        specs, slot = [], {}
+
+        def add(name, kind):
+            if name in slot:
+                return
+            t = batch.column(name)
+            v = batch.validity(name)
+            slot[name] = len(specs)
+            specs.append((kind, t, v))
+
+        for fn, col in self.aggs:
+            for b in self._buffer_cols(fn, col):
+                if b.startswith("count"):
+                    add(b, "sum")     # SUM_I64: exact merged counts
+                elif b.startswith("min"):
+                    add(b, "min")
+                elif b.startswith("max"):
+                    add(b, "max")
+                else:
+                    add(b, "sum")
+        return specs, slot
+
+    def _agg_batch(self, batch):
+        from . import gpuq
+        keys = self.group_keys
+        n = batch.num_rows()
+        cap = self.capacity or (1 << max(10, int(n).bit_length()))
+        if self.mode == "final":
+            specs, slot = self._merge_specs(batch)
+        else:
+            specs, slot = self._input_specs(batch)
+        mg = min(n, cap) + 2
+        if len(keys) == 0:
+            # global aggregate (empty grouping, HashAggregateExec.scala with
+            # no grouping expressions): one output row even on empty input —
+            # COUNT 0, SUM/MIN/MAX/AVG NULL (AggUtils emptyInputAggBuffer)
+            if n == 0:
+                cols, validity = {}, {}
+                dev = "cuda"
+
+                def src_dtype(name):
+                    base = name.split("(", 1)[1][:-1]
+                    try:
+                        return batch.column(base).dtype
+                    except KeyError:
+                        return torch.float64
+                names = (list(slot) if self.mode == "partial"
+                         else [f"{fn}({col})" for fn, col in self.aggs])
+                for name in names:
+                    if name.startswith("count"):
+                        cols[name] = torch.zeros(1, dtype=torch.int64, device=dev)
+                    elif name.startswith(("min(", "max(", "sum(")):
+                        cols[name] = torch.zeros(1, dtype=src_dtype(name),
+                                                 device=dev)
+                        validity[name] = torch.zeros(1, dtype=torch.uint8,
+                                                     device=dev)
+                    else:  # avg / avg_sum: f64
+                        cols[name] = torch.zeros(1, dtype=torch.float64,
+                                                 device=dev)
+                        validity[name] = torch.zeros(1, dtype=torch.uint8,
+                                                     device=dev)
+                batch.close()
+                return ColumnarBatch(cols, validity=validity or None)
+            key0 = gpuq.range_i64(n, 0, 0)   # constant key: one group
+            ok, okv, accs = gpuq.hash_agg_multi(key0, specs, cap, max_groups=mg)
+            key_cols, key_valid = {}, {}
+        elif len(keys) == 1:
+            k = keys[0]
+            ok, okv, accs = gpuq.hash_agg_multi(
+                batch.column(k), specs, cap,
+                key_validity=batch.validity(k), max_groups=mg)
+            key_cols = {k: ok}
+            # NULL-key group: okv==0 row. Only materialize a bitmap when the
+            # input key was nullable.
+            key_valid = {}
+            if batch.validity(k) is not None:
+                key_valid[k] = gpuq.u8_to_bits(okv)
+        else:
+            kc = [batch.column(k) for k in keys]
+            kv = [batch.validity(k) for k in keys]
+            okeys, kmask, accs = gpuq.hash_agg_keys(
+                kc, specs, cap, key_validities=kv, max_groups=mg)
+            key_cols = dict(zip(keys, okeys))
+            key_valid = {}
+            for c, k in enumerate(keys):
+                if kv[c] is not None:
+                    key_valid[k] = gpuq.maskbit_to_bits(kmask, c)
+        cols, validity = dict(key_cols), dict(key_valid)
+        if self.mode == "partial":
+            for name, j in slot.items():
+                cols[name] = accs[j]
+                # partial sum/min/max of an all-NULL group is NULL (its
+                # count companion is 0); counts themselves are never NULL
+                if not name.startswith("count"):
+                    base = name.split("(", 1)[1][:-1]
+                    validity[name] = gpuq.nonzero_to_bits(
+                        accs[slot[f"count({base})"]])
+        else:
             for fn, col in self.aggs:
+                out = f"{fn}({col})"
+                if fn == "count*":
+                    cols[out] = accs[slot["count(1)"]]
+                    continue
+                if fn == "count":
+                    cols[out] = accs[slot[f"count({col})"]]
+                    continue
+                cnt = accs[slot[f"count({col})"]]
                 if fn == "avg":
                     # Average.evaluateExpression: sum / cast(count)
-                    # (catalyst/.../expressions/aggregate/Average.scala)
-                    cols[f"{fn}({col})"] = gpuq.project_binop(
-                        osum, "/", b=gpuq.cast_i64_f64(ocnt))
+                    cols[out] = gpuq.project_binop(
+                        accs[slot[f"avg_sum({col})"]], "/",
+                        b=gpuq.cast_i64_f64(cnt))
                 else:
-                    cols[f"{fn}({col})"] = osum if fn == "sum" else ocnt
-            batch.close()
-            yield ColumnarBatch(cols, validity={self.group_key: None})
-
-    def _execute_multi(self, gpuq):
-        """multi-accumulator path (one gpuq_hash_agg_multi pass): several
-        aggregate expressions over multiple value columns (the Q1 shape)."""
-        for batch in self.children[0].execute_columnar():
-            keys = batch.column(self.group_key)
-            specs = []
-            slots = []  # (fn, col, acc indices)
-            for fn, col in self.aggs:
-                t = batch.column(col)
-                if fn == "avg" and t.dtype == torch.int64:
-                    t = gpuq.cast_i64_f64(t)   # AVG divides f64 sums
-                if fn == "sum":
-                    specs.append(("sum", t)); slots.append((fn, col, [len(specs) - 1]))
-                elif fn == "count":
-                    specs.append(("count", t, batch.validity(col)))
-                    slots.append((fn, col, [len(specs) - 1]))
-                else:  # avg = sum + count
-                    specs.append(("sum", t))
-                    specs.append(("count", t, batch.validity(col)))
-                    slots.append((fn, col, [len(specs) - 2, len(specs) - 1]))
-            n = keys.numel()
-            cap = self.capacity or (1 << max(10, int(n).bit_length()))
-            ok, okv, accs = gpuq.hash_agg_multi(
-                keys, specs, cap, key_validity=batch.validity(self.group_key),
-                max_groups=min(n, cap) + 2)
-            cols = {self.group_key: ok}
-            for fn, col, idx in slots:
-                if fn == "avg":
-                    cols[f"avg({col})"] = gpuq.project_binop(
-                        accs[idx[0]], "/", b=gpuq.cast_i64_f64(accs[idx[1]]))
-                else:
-                    cols[f"{fn}({col})"] = accs[idx[0]]
-            batch.close()
-            yield ColumnarBatch(cols, validity={self.group_key: None})
+                    cols[out] = accs[slot[f"{fn}({col})"]]
+                # SQL NULL iff no non-null input (Sum/Min/Max/Average.scala)
+                validity[out] = gpuq.nonzero_to_bits(cnt)
+        batch.close()
+        return ColumnarBatch(cols, validity=validity or None)
 
 
 class GpuShuffleExchangeExec(SparkPlan):
@@ -408,8 +570,8 @@ class GpuShuffleExchangeExec(SparkPlan):
 
     def __init__(self, keys: Tuple[str, ...], child):
         super().__init__(child)
-        assert len(keys) == 1, "round 1: single int64 partition key"
-        self.keys = keys
+        assert 1 <= len(keys) <= 4, "1..4 int64 partition key columns"
+        self.keys = tuple(keys)
 
     @property
     def output(self):
@@ -440,29 +602,71 @@ class GpuShuffleExchangeExec(SparkPlan):
         ShuffleExchangeLike.runtimeStatistics (:146)."""
         return dict(self._stats) if hasattr(self, "_stats") else None
 
+    def get_shuffle_partitions(self, specs):
+        """AQE getShuffleRDD(partitionSpecs) analog
+        (ShuffleExchangeExec.scala:141, ShuffledRowRDD.scala:33
+        CoalescedPartitionSpec): serve coalesced partition ranges straight
+        from this rank's partition-contiguous map output, no re-partition.
+        specs: list of (start_partition, end_partition) half-open ranges.
+        Returns one ColumnarBatch per spec (this rank's contribution)."""
+        assert hasattr(self, "_map_output"), "execute_columnar first"
+        cols, validity, offsets = self._map_output
+        out = []
+        for start, end in specs:
+            lo, hi = offsets[start], offsets[end]
+            sl = {n_: t[lo:hi] for n_, t in cols.items()}
+            va = {n_: None for n_ in sl}
+            for n_, u8 in validity.items():
+                from . import gpuq
+                va[n_] = gpuq.u8_to_bits(u8[lo:hi]) if hi > lo else None
+            out.append(ColumnarBatch(
+                sl, validity={k: v for k, v in va.items() if v is not None}))
+        return out
+
     def execute_columnar(self):
         import torch.distributed as dist
         from . import gpuq
         from .exchange import exchange_columns
         assert dist.is_initialized(), "GpuShuffleExchangeExec needs torch.distributed"
-        key_name = self.keys[0]
         world = self.num_partitions
         for batch in self.children[0].execute_columnar():
-            key = batch.column(key_name)
-            perm, counts = gpuq.partition_perm(key, world,
-                                               key_validity=batch.validity(key_name))
-            cols = {key_name: gpuq.gather(key, perm)}
+            kcols = [batch.column(k) for k in self.keys]
+            kvals = [batch.validity(k) for k in self.keys]
+            if len(self.keys) == 1:
+                perm, counts = gpuq.partition_perm(kcols[0], world,
+                                                   key_validity=kvals[0])
+            else:
+                perm, counts = gpuq.partition_perm_multi(kcols, world,
+                                                         key_validities=kvals)
+            cols, vcols = {}, {}
             row_bytes = 0
             for name, t in batch.columns().items():
-                if name != key_name:
-                    cols[name] = gpuq.gather(t, perm)
+                cols[name] = gpuq.gather(t, perm)
                 row_bytes += t.element_size()
+                v = batch.validity(name)
+                if v is not None:
+                    # bitmaps travel as u8 columns: partition split points
+                    # are not byte-aligned (repacked on receive)
+                    pv = gpuq.gather_bits(v, perm)
+                    vcols[name] = gpuq.bits_to_u8(pv, t.numel())
+                    row_bytes += 1
             in_splits = counts.cpu().tolist()
             self._stats = {"bytes_by_partition": [c * row_bytes for c in in_splits],
                            "rows_written": int(sum(in_splits))}
-            out, _ = exchange_columns(cols, in_splits)
+            offsets = [0]
+            for c in in_splits:
+                offsets.append(offsets[-1] + c)
+            self._map_output = (cols, vcols, offsets)
+            wire = dict(cols)
+            wire.update({f"__valid__{n}": u8 for n, u8 in vcols.items()})
+            out, _ = exchange_columns(wire, in_splits)
             batch.close()
-            yield ColumnarBatch(out)
+            validity = {}
+            for name in list(out):
+                if name.startswith("__valid__"):
+                    u8 = out.pop(name)
+                    validity[name[len("__valid__"):]] = gpuq.u8_to_bits(u8)
+            yield ColumnarBatch(out, validity=validity or None)
 
 
 class GpuShuffledHashJoinExec(SparkPlan):
@@ -510,15 +714,21 @@ class GpuShuffledHashJoinExec(SparkPlan):
             if op is not None:
                 break
             out_cap = nm + 64
-        cols = {}
+        cols, validity = {}, {}
         for name, t in build.columns().items():
             cols[name] = gpuq.gather(t, ob)
+            v = build.validity(name)
+            if v is not None:
+                validity[name] = gpuq.gather_bits(v, ob)
         for name, t in probe.columns().items():
+            v = probe.validity(name)
             if name in cols:
                 name = f"{name}#probe"
             cols[name] = gpuq.gather(t, op)
+            if v is not None:
+                validity[name] = gpuq.gather_bits(v, op)
         lb.close(), rb.close()
-        yield ColumnarBatch(cols)
+        yield ColumnarBatch(cols, validity=validity or None)
 
 
 class GpuRangeExec(SparkPlan):
@@ -556,6 +766,11 @@ class GpuFilterExec(SparkPlan):
         return self.children[0].output
 
     @property
+    def output_ordering(self):
+        # stable compaction preserves the child's order (FilterExec does)
+        return self.children[0].output_ordering
+
+    @property
     def supports_columnar(self):
         return True
 
@@ -565,10 +780,14 @@ class GpuFilterExec(SparkPlan):
             perm, cnt = gpuq.filter_cmp(batch.column(self.col), self.op,
                                         self.literal,
                                         validity=batch.validity(self.col))
-            cols = {name: gpuq.gather(t, perm)
-                    for name, t in batch.columns().items()}
+            cols, validity = {}, {}
+            for name, t in batch.columns().items():
+                cols[name] = gpuq.gather(t, perm)
+                v = batch.validity(name)
+                if v is not None and cnt:
+                    validity[name] = gpuq.gather_bits(v, perm)
             batch.close()
-            yield ColumnarBatch(cols)
+            yield ColumnarBatch(cols, validity=validity or None)
 
 
 class GpuProjectExec(SparkPlan):
@@ -623,12 +842,20 @@ class GpuBroadcastExchangeExec(SparkPlan):
 
     def execute_columnar(self):
         import torch.distributed as dist
-        from .exchange import broadcast_gather
+        from . import gpuq
+        from .exchange import broadcast_gather, _unpack_validity_wire
         for batch in self.children[0].execute_columnar():
             if dist.is_initialized() and dist.get_world_size() > 1:
-                cols = broadcast_gather(batch.columns())
+                wire = batch.columns()
+                n = batch.num_rows()
+                for name in list(wire):
+                    v = batch.validity(name)
+                    if v is not None:
+                        wire[f"__valid__{name}"] = gpuq.bits_to_u8(v, n)
+                cols = broadcast_gather(wire)
+                validity = _unpack_validity_wire(gpuq, cols)
                 batch.close()
-                yield ColumnarBatch(cols)
+                yield ColumnarBatch(cols, validity=validity or None)
             else:
                 yield batch
 
@@ -652,6 +879,9 @@ class GpuColumnarRule:
     preColumnarTransitions swaps CPU nodes for GPU subclasses — the
     SparkSessionExtensionSuite.scala:959-1000 pattern."""
 
+    def __init__(self, preserve_smj_ordering: bool = True):
+        self.preserve_smj_ordering = preserve_smj_ordering
+
     def pre_columnar_transitions(self, plan: SparkPlan) -> SparkPlan:
         children = [self.pre_columnar_transitions(c) for c in plan.children]
         if isinstance(plan, SortExec):
@@ -664,9 +894,16 @@ class GpuColumnarRule:
                                            plan.build_side, *children)
         if isinstance(plan, SortMergeJoinExec):
             # SMJ -> GPU hash join (same slot, same required distribution;
-            # build on the right side as SHJ's default would choose)
-            return GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
-                                           "right", *children)
+            # build on the right side as SHJ's default would choose). SMJ
+            # advertises outputOrdering on the streamed keys — parents may
+            # have had sorts elided against it — so re-sort the hash join's
+            # output to honor the contract (one radix pass; the GPU sort
+            # node's output_ordering then matches what SMJ declared).
+            j = GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
+                                        "right", *children)
+            if self.preserve_smj_ordering:
+                return GpuSortExec(SortOrder(plan.left_key), False, j)
+            return j
         if isinstance(plan, ShuffleExchangeExec):
             return GpuShuffleExchangeExec(plan.keys, *children)
         if isinstance(plan, FilterExec):

@@ -84,6 +84,32 @@ def lib() -> ctypes.CDLL:
         L.gpuq_partition_workspace_bytes.argtypes = [i64, i32]
         L.gpuq_partition_perm.restype = i32
         L.gpuq_partition_perm.argtypes = [vp, i64, _Col, i32, vp, vp, vp, i64]
+        L.gpuq_partition_perm_multi.restype = i32
+        L.gpuq_partition_perm_multi.argtypes = [vp, i64, ctypes.POINTER(_Col), i32,
+                                                i32, vp, vp, vp, i64]
+        L.gpuq_hash_agg_keys_workspace_bytes.restype = i64
+        L.gpuq_hash_agg_keys_workspace_bytes.argtypes = [i64, i32, i32]
+        L.gpuq_hash_agg_keys.restype = i32
+        L.gpuq_hash_agg_keys.argtypes = [vp, i64, ctypes.POINTER(_Col), i32,
+                                         ctypes.POINTER(_Col), vp, vp, i32,
+                                         vp, i64, i32, i32,
+                                         vp, vp, vp, ctypes.POINTER(i64)]
+        L.gpuq_gather_bits.restype = i32
+        L.gpuq_gather_bits.argtypes = [vp, i64, vp, vp, vp]
+        L.gpuq_bits_to_u8.restype = i32
+        L.gpuq_bits_to_u8.argtypes = [vp, i64, vp, vp]
+        L.gpuq_u8_to_bits.restype = i32
+        L.gpuq_u8_to_bits.argtypes = [vp, i64, vp, vp]
+        L.gpuq_nonzero_to_bits.restype = i32
+        L.gpuq_nonzero_to_bits.argtypes = [vp, i64, vp, vp]
+        L.gpuq_maskbit_to_bits.restype = i32
+        L.gpuq_maskbit_to_bits.argtypes = [vp, i64, vp, i32, vp]
+        L.gpuq_minmax_i64.restype = i32
+        L.gpuq_minmax_i64.argtypes = [vp, i64, _Col, vp]
+        L.gpuq_pack2_i64.restype = i32
+        L.gpuq_pack2_i64.argtypes = [vp, i64, vp, vp, i64, i64, i32, vp]
+        L.gpuq_unpack2_i64.restype = i32
+        L.gpuq_unpack2_i64.argtypes = [vp, i64, vp, i64, i64, i32, vp, vp]
         L.gpuq_range_partition_perm.restype = i32
         L.gpuq_range_partition_perm.argtypes = [vp, i64, _Col, i32, i32, vp, i32,
                                                 vp, vp, vp, i64]
@@ -296,9 +322,33 @@ BINOP = {"+": 0, "-": 1, "*": 2, "/": 3}
 
 def filter_cmp(col: torch.Tensor, op: str, literal, workspace=None, validity=None):
     """Stable filter: returns (perm[:count], count) — passing rows in input
-    order (FilterExec replacement for col OP literal predicates)."""
+    order (FilterExec replacement for col OP literal predicates).
+
+    An int64 column compared against a fractional literal follows Spark's
+    cast-to-double comparison semantics: the predicate is rewritten to an
+    equivalent integer comparison (e.g. k < 0.5 == k <= 0), never truncated
+    (k < 0.5 must keep k == 0)."""
     n = col.numel()
     dev = col.device
+    if col.dtype == torch.int64 and isinstance(literal, float) \
+            and literal != int(literal):
+        import math
+        f = math.floor(literal)
+        I64_MIN, I64_MAX = -(1 << 63), (1 << 63) - 1
+        if op in ("==",):
+            op, literal = "<", I64_MIN          # never true
+        elif op in ("!=",):
+            op, literal = ">=", I64_MIN         # true for every valid row
+        elif op in ("<", "<="):
+            # col < L  <=>  col <= floor(L)
+            op, literal = ("<=", f) if f >= I64_MIN else ("<", I64_MIN)
+            if f > I64_MAX:
+                op, literal = ">=", I64_MIN     # all pass
+        else:  # > / >=  : col > L <=> col >= floor(L)+1
+            g = f + 1
+            op, literal = (">=", g) if g <= I64_MAX else ("<", I64_MIN)
+            if g < I64_MIN:
+                op, literal = ">=", I64_MIN     # all pass
     if workspace is None:
         workspace = torch.empty(lib().gpuq_filter_workspace_bytes(n),
                                 dtype=torch.uint8, device=dev)
@@ -336,47 +386,58 @@ def range_i64(n: int, start: int = 0, step: int = 1, device="cuda") -> torch.Ten
     return out
 
 
-def hash_agg_multi(keys: torch.Tensor, specs, capacity: int, key_validity=None,
-                   max_groups=None):
-    """specs: list of ("sum", col_tensor) / ("count", col_tensor, validity) /
-    ("count*", None). Returns (keys, key_valid, [acc_j...]) sliced to
-    ngroups; SUM accs f64, COUNT accs i64."""
-    n = keys.numel()
-    dev = keys.device
-    nspecs = len(specs)
+def _build_agg_specs(specs, dev, mg, placeholder):
+    """specs: list of (kind, col_tensor_or_None[, validity]) with kind in
+    {"sum","count","count*","min","max"}. Op codes per gpuq.h:
+    0=SUM_F64 1=COUNT(col) 2=COUNT(*) 3=SUM_I64 4=MIN_I64 5=MAX_I64
+    6=MIN_F64 7=MAX_F64. Returns (cols_arr, ops_arr, cid_arr, outs)."""
     cols, ops, colidx, outs = [], [], [], []
     for s in specs:
         kind = s[0]
-        if kind == "sum":
-            t = s[1]
-            if t.dtype == torch.int64:
-                ops.append(3)  # SUM(int64) -> int64, wrapping (Sum.scala)
-                outs.append(torch.empty(max_groups or capacity + 2,
-                                        dtype=torch.int64, device=dev))
-            else:
-                ops.append(0)
-                outs.append(torch.empty(max_groups or capacity + 2,
-                                        dtype=torch.float64, device=dev))
-            colidx.append(len(cols)); cols.append(_col(t))
-        elif kind == "count":
-            ops.append(1); colidx.append(len(cols))
-            cols.append(_col(s[1], s[2] if len(s) > 2 else None))
-            outs.append(torch.empty(max_groups or capacity + 2,
-                                    dtype=torch.int64, device=dev))
-        else:
+        t = s[1] if len(s) > 1 else None
+        v = s[2] if len(s) > 2 else None
+        if kind == "count*":
             ops.append(2); colidx.append(0)
-            outs.append(torch.empty(max_groups or capacity + 2,
-                                    dtype=torch.int64, device=dev))
+            outs.append(torch.empty(mg, dtype=torch.int64, device=dev))
+            continue
+        if kind == "count":
+            ops.append(1)
+        elif kind == "sum":
+            ops.append(3 if t.dtype == torch.int64 else 0)
+        elif kind == "min":
+            ops.append(4 if t.dtype == torch.int64 else 6)
+        elif kind == "max":
+            ops.append(5 if t.dtype == torch.int64 else 7)
+        else:
+            raise ValueError(f"bad agg spec kind {kind}")
+        out_dtype = torch.float64 if ops[-1] in (0, 6, 7) else torch.int64
+        outs.append(torch.empty(mg, dtype=out_dtype, device=dev))
+        colidx.append(len(cols))
+        cols.append(_col(t, v))
     if not cols:
-        cols = [_col(keys)]  # placeholder, unused
+        cols = [_col(placeholder)]  # placeholder, unused
+    nspecs = len(specs)
     cols_arr = (_Col * len(cols))(*cols)
     ops_arr = (ctypes.c_int32 * nspecs)(*ops)
     cid_arr = (ctypes.c_int32 * nspecs)(*colidx)
+    return cols_arr, ops_arr, cid_arr, outs
+
+
+def hash_agg_multi(keys: torch.Tensor, specs, capacity: int, key_validity=None,
+                   max_groups=None):
+    """Multi-accumulator GROUP BY (single int64 key). specs per
+    _build_agg_specs. Returns (keys, key_valid, [acc_j...]) sliced to
+    ngroups."""
+    n = keys.numel()
+    dev = keys.device
+    nspecs = len(specs)
+    mg = max_groups or capacity + 2
+    cols_arr, ops_arr, cid_arr, outs = _build_agg_specs(specs, dev, mg, keys)
     outp_arr = (ctypes.c_void_p * nspecs)(*[t.data_ptr() for t in outs])
     ws = torch.empty(lib().gpuq_hash_agg_multi_workspace_bytes(capacity, nspecs),
                      dtype=torch.uint8, device=dev)
-    ok = torch.empty(max_groups or capacity + 2, dtype=torch.int64, device=dev)
-    okv = torch.empty(max_groups or capacity + 2, dtype=torch.uint8, device=dev)
+    ok = torch.empty(mg, dtype=torch.int64, device=dev)
+    okv = torch.empty(mg, dtype=torch.uint8, device=dev)
     ng = ctypes.c_int64(0)
     _check(lib().gpuq_hash_agg_multi(
         _stream(), n, _col(keys, key_validity), cols_arr, ops_arr, cid_arr,
@@ -384,6 +445,133 @@ def hash_agg_multi(keys: torch.Tensor, specs, capacity: int, key_validity=None,
         outp_arr, ctypes.byref(ng)))
     gn = ng.value
     return ok[:gn], okv[:gn], [t[:gn] for t in outs]
+
+
+def hash_agg_keys(key_cols, specs, capacity: int, key_validities=None,
+                  max_groups=None):
+    """Composite-key GROUP BY (k1..kK), K <= 4 int64 columns.
+    key_validities: optional list (same length) of validity bitmaps.
+    Returns (key_out_cols list, kmask u8 tensor, [acc_j...]) sliced to
+    ngroups; kmask bit c = key column c non-NULL in that group."""
+    nkeys = len(key_cols)
+    n = key_cols[0].numel()
+    dev = key_cols[0].device
+    kv = key_validities or [None] * nkeys
+    nspecs = len(specs)
+    mg = max_groups or capacity + 2
+    cols_arr, ops_arr, cid_arr, outs = _build_agg_specs(specs, dev, mg, key_cols[0])
+    outp_arr = (ctypes.c_void_p * nspecs)(*[t.data_ptr() for t in outs])
+    keys_arr = (_Col * nkeys)(*[_col(k, v) for k, v in zip(key_cols, kv)])
+    okeys = [torch.empty(mg, dtype=torch.int64, device=dev) for _ in range(nkeys)]
+    okp_arr = (ctypes.c_void_p * nkeys)(*[t.data_ptr() for t in okeys])
+    omask = torch.empty(mg, dtype=torch.uint8, device=dev)
+    ws = torch.empty(lib().gpuq_hash_agg_keys_workspace_bytes(capacity, nkeys, nspecs),
+                     dtype=torch.uint8, device=dev)
+    ng = ctypes.c_int64(0)
+    _check(lib().gpuq_hash_agg_keys(
+        _stream(), n, keys_arr, nkeys, cols_arr, ops_arr, cid_arr, nspecs,
+        ws.data_ptr(), capacity, 1, 1, okp_arr, omask.data_ptr(),
+        outp_arr, ctypes.byref(ng)))
+    gn = ng.value
+    return [t[:gn] for t in okeys], omask[:gn], [t[:gn] for t in outs]
+
+
+def partition_perm_multi(key_cols, nparts: int, workspace=None,
+                         key_validities=None):
+    """Stable group-by-partition permutation over a composite key tuple
+    (seed-chained Murmur3, hash.scala:849-860)."""
+    nkeys = len(key_cols)
+    n = key_cols[0].numel()
+    dev = key_cols[0].device
+    kv = key_validities or [None] * nkeys
+    if workspace is None:
+        workspace = partition_workspace(n, nparts, dev)
+    keys_arr = (_Col * nkeys)(*[_col(k, v) for k, v in zip(key_cols, kv)])
+    perm = torch.empty(n, dtype=torch.int32, device=dev)
+    counts = torch.empty(nparts, dtype=torch.int64, device=dev)
+    _check(lib().gpuq_partition_perm_multi(
+        _stream(), n, keys_arr, nkeys, nparts, perm.data_ptr(),
+        counts.data_ptr(), workspace.data_ptr(), workspace.numel()))
+    return perm, counts
+
+
+def _bitmap_bytes(n: int) -> int:
+    return (n + 7) // 8
+
+
+def gather_bits(bits: torch.Tensor, perm: torch.Tensor) -> torch.Tensor:
+    """Permute a validity bitmap: out bit i = bits[perm[i]]."""
+    n = perm.numel()
+    out = torch.empty(_bitmap_bytes(n), dtype=torch.uint8, device=perm.device)
+    _check(lib().gpuq_gather_bits(_stream(), n, bits.data_ptr(), perm.data_ptr(),
+                                  out.data_ptr()))
+    return out
+
+
+def bits_to_u8(bits: torch.Tensor, n: int) -> torch.Tensor:
+    out = torch.empty(n, dtype=torch.uint8, device=bits.device)
+    _check(lib().gpuq_bits_to_u8(_stream(), n, bits.data_ptr(), out.data_ptr()))
+    return out
+
+
+def u8_to_bits(u8: torch.Tensor) -> torch.Tensor:
+    n = u8.numel()
+    out = torch.empty(_bitmap_bytes(n), dtype=torch.uint8, device=u8.device)
+    _check(lib().gpuq_u8_to_bits(_stream(), n, u8.data_ptr(), out.data_ptr()))
+    return out
+
+
+def nonzero_to_bits(col: torch.Tensor) -> torch.Tensor:
+    """bit i = (col[i] != 0) — NULL-ness of merged aggregates from a merged
+    COUNT column (Sum.scala: NULL iff no non-null input)."""
+    n = col.numel()
+    out = torch.empty(_bitmap_bytes(n), dtype=torch.uint8, device=col.device)
+    _check(lib().gpuq_nonzero_to_bits(_stream(), n, col.data_ptr(), out.data_ptr()))
+    return out
+
+
+def maskbit_to_bits(mask: torch.Tensor, bit: int) -> torch.Tensor:
+    """Per-key-column validity bitmap from hash_agg_keys' kmask."""
+    n = mask.numel()
+    out = torch.empty(_bitmap_bytes(n), dtype=torch.uint8, device=mask.device)
+    _check(lib().gpuq_maskbit_to_bits(_stream(), n, mask.data_ptr(), bit,
+                                      out.data_ptr()))
+    return out
+
+
+def minmax_i64(col: torch.Tensor, validity=None):
+    """(min, max, valid_count) of an int64 column; (None, None, 0) if no
+    valid rows."""
+    out = torch.empty(3, dtype=torch.int64, device=col.device)
+    _check(lib().gpuq_minmax_i64(_stream(), col.numel(), _col(col, validity),
+                                 out.data_ptr()))
+    enc = out.cpu().tolist()
+    sign = 1 << 63
+    cnt = enc[2]
+    if cnt == 0:
+        return None, None, 0
+    # decode_i64: e ^ SIGNBIT (two's complement)
+    def dec(e):
+        v = (e & 0xFFFFFFFFFFFFFFFF) ^ sign
+        return v - (1 << 64) if v >= sign else v
+    return dec(enc[0]), dec(enc[1]), cnt
+
+
+def pack2_i64(a: torch.Tensor, b: torch.Tensor, a_bias: int, b_bias: int,
+              shift: int) -> torch.Tensor:
+    out = torch.empty(a.numel(), dtype=torch.int64, device=a.device)
+    _check(lib().gpuq_pack2_i64(_stream(), a.numel(), a.data_ptr(), b.data_ptr(),
+                                a_bias, b_bias, shift, out.data_ptr()))
+    return out
+
+
+def unpack2_i64(packed: torch.Tensor, a_bias: int, b_bias: int, shift: int):
+    n = packed.numel()
+    a = torch.empty(n, dtype=torch.int64, device=packed.device)
+    b = torch.empty(n, dtype=torch.int64, device=packed.device)
+    _check(lib().gpuq_unpack2_i64(_stream(), n, packed.data_ptr(), a_bias, b_bias,
+                                  shift, a.data_ptr(), b.data_ptr()))
+    return a, b
 
 
 def range_partition_perm(keys: torch.Tensor, bounds: torch.Tensor, desc=False,

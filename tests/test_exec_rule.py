@@ -63,14 +63,26 @@ def test_batch_lifetime():
 
 def test_sort_merge_join_replaced_by_gpu_hash_join():
     # SMJ and SHJ fill the same equi-join plan slot; the
-    # rule maps both to the GPU hash join
+    # rule maps both to the GPU hash join. SMJ additionally advertises
+    # outputOrdering on the streamed keys (SortMergeJoinExec.scala:83), so
+    # the rule re-sorts the hash join's output to honor that contract —
+    # the replacement's output_ordering must cover what SMJ declared.
     scan = gx.InputBatches.__new__(gx.InputBatches)
     gx.SparkPlan.__init__(scan)
     scan._batches = []
     smj = gx.SortMergeJoinExec("a", "b", scan, scan)
+    declared = smj.output_ordering
+    assert [o.key for o in declared] == ["a"]
     plan = gx.GpuColumnarRule().pre_columnar_transitions(smj)
-    assert isinstance(plan, gx.GpuShuffledHashJoinExec)
-    assert plan.left_key == "a" and plan.right_key == "b"
+    assert isinstance(plan, gx.GpuSortExec)
+    assert [o.key for o in plan.output_ordering] == ["a"]
+    join = plan.children[0]
+    assert isinstance(join, gx.GpuShuffledHashJoinExec)
+    assert join.left_key == "a" and join.right_key == "b"
+    # opt-out for plans that provably don't rely on SMJ ordering
+    plan2 = gx.GpuColumnarRule(preserve_smj_ordering=False) \
+        .pre_columnar_transitions(gx.SortMergeJoinExec("a", "b", scan, scan))
+    assert isinstance(plan2, gx.GpuShuffledHashJoinExec)
 
 
 def test_broadcast_join_rule():
