@@ -115,9 +115,17 @@ class _CpuNode(SparkPlan):
 
 
 class SortExec(_CpuNode):
-    def __init__(self, sort_order: SortOrder, global_sort: bool, child):
+    """sort_order: one SortOrder or a list (multi-key ORDER BY,
+    SortExec.scala sortOrder: Seq[SortOrder])."""
+
+    def __init__(self, sort_order, global_sort: bool, child):
         super().__init__(child)
         self.sort_order, self.global_sort = sort_order, global_sort
+
+    @property
+    def sort_orders(self):
+        return ([self.sort_order] if isinstance(self.sort_order, SortOrder)
+                else list(self.sort_order))
 
     @property
     def output(self):
@@ -125,7 +133,7 @@ class SortExec(_CpuNode):
 
     @property
     def output_ordering(self):
-        return [self.sort_order]
+        return self.sort_orders
 
 
 class HashAggregateExec(_CpuNode):
@@ -271,12 +279,20 @@ class InputBatches(SparkPlan):
 # ---------------- GPU exec nodes ----------------
 
 class GpuSortExec(SparkPlan):
-    """Replaces SortExec (SortExec.scala:75-126): radix-eligible single-key
-    sort (canUseRadixSort analog: int64/float64 key)."""
+    """Replaces SortExec (SortExec.scala:75-126): radix-eligible
+    int64/float64 keys (canUseRadixSort analog). Multi-key ORDER BY
+    composes as successive STABLE radix passes from the last sort key to
+    the first — the stability the kernel guarantees makes the composition
+    exactly the lexicographic order SortExec produces."""
 
-    def __init__(self, sort_order: SortOrder, global_sort: bool, child):
+    def __init__(self, sort_order, global_sort: bool, child):
         super().__init__(child)
         self.sort_order, self.global_sort = sort_order, global_sort
+
+    @property
+    def sort_orders(self):
+        return ([self.sort_order] if isinstance(self.sort_order, SortOrder)
+                else list(self.sort_order))
 
     @property
     def output(self):
@@ -284,7 +300,7 @@ class GpuSortExec(SparkPlan):
 
     @property
     def output_ordering(self):
-        return [self.sort_order]
+        return self.sort_orders
 
     @property
     def supports_columnar(self):
@@ -292,20 +308,49 @@ class GpuSortExec(SparkPlan):
 
     def required_child_distribution(self):
         if self.global_sort:
-            return [Distribution("ordered", (self.sort_order.key,))]
+            return [Distribution("ordered", tuple(o.key for o in self.sort_orders))]
         return [Distribution("unspecified")]
+
+    @staticmethod
+    def _sort_pass(batch, o):
+        """one stable sort of the whole batch by o (keys + payload +
+        validity bitmaps through the same permutation)."""
+        from . import gpuq
+        keys = batch.column(o.key)
+        kvalid = batch.validity(o.key)
+        perm, skeys = gpuq.sort_perm(keys, desc=o.descending,
+                                     nulls_first=o.nulls_first,
+                                     key_validity=kvalid)
+        cols = {o.key: skeys}
+        validity = {}
+        if kvalid is not None:
+            validity[o.key] = gpuq.gather_bits(kvalid, perm)
+        for name, t in batch.columns().items():
+            if name != o.key:
+                cols[name] = gpuq.gather(t, perm)
+                v = batch.validity(name)
+                if v is not None:
+                    validity[name] = gpuq.gather_bits(v, perm)
+        batch.close()
+        return ColumnarBatch(cols, validity=validity or None)
 
     def execute_columnar(self):
         import torch.distributed as dist
-        from . import gpuq
-        o = self.sort_order
+        from . import gpuq  # noqa: F401 (engine presence check)
+        orders = self.sort_orders
         for batch in self.children[0].execute_columnar():
             if (self.global_sort and dist.is_initialized()
                     and dist.get_world_size() > 1):
-                # global ORDER BY: range exchange first (the EnsureRequirements
-                # RangePartitioning insertion, exchange/EnsureRequirements.scala:296);
-                # after the local sort below, rank-major order is global order
+                # global ORDER BY: range exchange on the PRIMARY key first
+                # (the EnsureRequirements RangePartitioning insertion,
+                # exchange/EnsureRequirements.scala:296); after the local
+                # sort below, rank-major order is global order. Multi-key
+                # global sort would need tie-aware range bounds — the
+                # single-key contract is asserted.
+                assert len(orders) == 1, \
+                    "multi-key global ORDER BY across ranks unsupported"
                 from .exchange import range_exchange
+                o = orders[0]
                 payload = {n_: t for n_, t in batch.columns().items()
                            if n_ != o.key}
                 validity = {n_: batch.validity(n_) for n_ in batch.columns()
@@ -321,23 +366,9 @@ class GpuSortExec(SparkPlan):
                 cols.update(payload)
                 batch.close()
                 batch = ColumnarBatch(cols, validity=validity or None)
-            keys = batch.column(o.key)
-            kvalid = batch.validity(o.key)
-            perm, skeys = gpuq.sort_perm(keys, desc=o.descending,
-                                         nulls_first=o.nulls_first,
-                                         key_validity=kvalid)
-            cols = {o.key: skeys}
-            validity = {}
-            if kvalid is not None:
-                validity[o.key] = gpuq.gather_bits(kvalid, perm)
-            for name, t in batch.columns().items():
-                if name != o.key:
-                    cols[name] = gpuq.gather(t, perm)
-                    v = batch.validity(name)
-                    if v is not None:
-                        validity[name] = gpuq.gather_bits(v, perm)
-            batch.close()
-            yield ColumnarBatch(cols, validity=validity or None)
+            for o in reversed(orders):
+                batch = self._sort_pass(batch, o)
+            yield batch
 
 
 class GpuHashAggregateExec(SparkPlan):
@@ -373,6 +404,10 @@ class GpuHashAggregateExec(SparkPlan):
             return ()
         return (self.group_key,) if isinstance(self.group_key, str) else tuple(self.group_key)
 
+    @staticmethod
+    def agg_out_name(fn: str, col) -> str:
+        return "count(1)" if fn == "count*" else f"{fn}({col})"
+
     def _buffer_cols(self, fn: str, col: str) -> List[str]:
         """agg buffer column names (partial-mode output schema)."""
         if fn == "sum":
@@ -395,7 +430,7 @@ class GpuHashAggregateExec(SparkPlan):
                     if b not in seen:
                         seen.add(b); bufs.append(b)
             return keys + bufs
-        return keys + [f"{fn}({col})" for fn, col in self.aggs]
+        return keys + [self.agg_out_name(fn, col) for fn, col in self.aggs]
 
     @property
     def supports_columnar(self):
@@ -542,7 +577,7 @@ class GpuHashAggregateExec(SparkPlan):
                         accs[slot[f"count({base})"]])
         else:
             for fn, col in self.aggs:
-                out = f"{fn}({col})"
+                out = self.agg_out_name(fn, col)
                 if fn == "count*":
                     cols[out] = accs[slot["count(1)"]]
                     continue
