@@ -184,3 +184,60 @@ def test_range_exchange_logic_gloo_world2():
     mp.start_processes(_worker_range, args=(29534, q), nprocs=WORLD,
                        join=True, start_method="spawn")
     assert q.empty(), q.get()
+
+
+def _worker_validity(rank, port, fail_q):
+    try:
+        import torch
+        import torch.distributed as dist
+        os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                          RANK=str(rank), WORLD_SIZE=str(WORLD))
+        dist.init_process_group("gloo")
+        import oracle
+        from spark_amd.exchange import exchange_columns
+
+        # NULL payloads travel as u8 wire columns alongside the data (the
+        # exec node's transport form: split points aren't byte-aligned, so
+        # bitmaps are unpacked for the all-to-all and repacked on receive —
+        # ColumnVector null contract, ColumnVector.java:58-366)
+        n = 8_000
+        keys = oracle.gen_i64(seed=300 + rank, n=n)
+        pay = oracle.gen_i64(seed=400 + rank, n=n)
+        null_mask = (oracle.gen_i64(seed=500 + rank, n=n, range_=4) == 0)
+        pids = oracle.partition_ids(keys, WORLD)
+        perm = np.argsort(pids, kind="stable")
+        counts = np.bincount(pids, minlength=WORLD).tolist()
+        cols = {"k": torch.from_numpy(keys[perm]),
+                "p": torch.from_numpy(pay[perm]),
+                "__valid__p": torch.from_numpy(
+                    (~null_mask[perm]).astype(np.uint8))}
+        out, _ = exchange_columns(cols, counts)
+        got_k = out["k"].numpy()
+        got_p = out["p"].numpy()
+        got_v = out["__valid__p"].numpy().astype(bool)
+        exp_k, exp_p, exp_v = [], [], []
+        for src in range(WORLD):
+            sk = oracle.gen_i64(seed=300 + src, n=n)
+            sp = oracle.gen_i64(seed=400 + src, n=n)
+            sm = (oracle.gen_i64(seed=500 + src, n=n, range_=4) == 0)
+            sel = oracle.partition_ids(sk, WORLD) == rank
+            exp_k.append(sk[sel]); exp_p.append(sp[sel]); exp_v.append(~sm[sel])
+        exp_k = np.concatenate(exp_k)
+        exp_p = np.concatenate(exp_p)
+        exp_v = np.concatenate(exp_v)
+        assert (got_k == exp_k).all()
+        assert (got_v == exp_v).all()
+        # data under NULL positions is garbage by contract; compare valid only
+        assert (got_p[got_v] == exp_p[exp_v]).all()
+        dist.destroy_process_group()
+    except Exception as e:
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}")
+        raise
+
+
+def test_exchange_validity_wire_gloo_world2():
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    mp.start_processes(_worker_validity, args=(29534, q), nprocs=WORLD,
+                       join=True, start_method="spawn")
+    assert q.empty(), q.get()
