@@ -186,10 +186,67 @@ class JoinWorkload:
         dist.all_to_all_single(rp, pp, out_splits, in_splits)
         return rk, rp
 
+    def _exchange_tiled(self, sides, ntiles=4):
+        """Overlapped exchange (north star: the all-to-all rides a side HIP
+        stream, overlapped with the partition kernel): each table side is
+        split into tiles; tile i's RCCL all-to-all runs on the comm stream
+        while tile i+1 is partitioned + gathered on the compute stream.
+        Join semantics don't depend on row order, so per-tile arrival
+        order is free concat fodder."""
+        gq = self.gq
+        import torch.distributed as dist
+        if not hasattr(self, "_comm_stream"):
+            self._comm_stream = torch.cuda.Stream()
+        comm = self._comm_stream
+        compute = torch.cuda.current_stream()
+        recv = [[] for _ in sides]
+        jobs = []
+        for si, (keys, pay) in enumerate(sides):
+            n = keys.numel()
+            step = (n + ntiles - 1) // ntiles
+            for t0 in range(0, n, step):
+                jobs.append((si, keys[t0:t0 + step], pay[t0:t0 + step]))
+        out_counts = torch.empty(self.world, dtype=torch.int64, device="cuda")
+        for si, k, p in jobs:
+            # compute stream: partition + gather THIS tile (overlaps the
+            # previous tile's data all-to-all already queued on `comm`)
+            perm, counts = gq.partition_perm(k, self.world)
+            pk = gq.gather(k, perm)
+            pp = gq.gather(p, perm)
+            # these tensors are consumed by the comm stream after their
+            # python refs drop next iteration — pin them in the caching
+            # allocator against compute-stream reuse until comm is done
+            pk.record_stream(comm)
+            pp.record_stream(comm)
+            counts.record_stream(comm)
+            ev = torch.cuda.Event()
+            ev.record(compute)
+            in_splits = counts.cpu().tolist()
+            with torch.cuda.stream(comm):
+                comm.wait_event(ev)
+                dist.all_to_all_single(out_counts, counts)
+                out_splits = out_counts.cpu().tolist()
+                total = sum(out_splits)
+                rk = torch.empty(total, dtype=torch.int64, device="cuda")
+                rp = torch.empty(total, dtype=torch.int64, device="cuda")
+                dist.all_to_all_single(rk, pk, out_splits, in_splits)
+                dist.all_to_all_single(rp, pp, out_splits, in_splits)
+            recv[si].append((rk, rp))
+        compute.wait_stream(comm)
+        out = []
+        for parts in recv:
+            out.append((torch.cat([a for a, _ in parts]),
+                        torch.cat([b for _, b in parts])))
+        return out
+
     def step(self):
         gq = self.gq
-        bk, bp = self._exchange(self.bkeys, self.bpay)
-        pk, pp = self._exchange(self.pkeys, self.ppay)
+        if self.world > 1 and not os.environ.get("GPUQ_NO_OVERLAP"):
+            (bk, bp), (pk, pp) = self._exchange_tiled(
+                [(self.bkeys, self.bpay), (self.pkeys, self.ppay)])
+        else:
+            bk, bp = self._exchange(self.bkeys, self.bpay)
+            pk, pp = self._exchange(self.pkeys, self.ppay)
         bn = bk.numel()
         gq.lib().gpuq_join_build_i64(gq._stream(), bn, gq._col(bk),
                                      self.join_ws.data_ptr(), self.cap)
@@ -219,7 +276,7 @@ class GlobalSortWorkload:
     def step(self):
         from spark_amd.exchange import range_exchange
         gq = self.gq
-        k, payload = range_exchange(self.keys, {"p": self.pay})
+        k, payload, _ = range_exchange(self.keys, {"p": self.pay})
         perm, skeys = gq.sort_perm(k)
         gq.gather(payload["p"], perm)
         self.nrows_local = k.numel()

@@ -473,6 +473,10 @@ class GpuHashAggregateExec(SparkPlan):
             else:  # sum/min/max + count companion for NULL-ness
                 add(f"{fn}({col})", (fn, t, v))
                 add(f"count({col})", ("count", t, v))
+        if not specs:
+            # pure GROUP BY (no aggregate expressions — the SELECT DISTINCT
+            # shape): the kernel wants >= 1 spec; count rows and discard
+            add("__rows__", ("count*",))
         return specs, slot
 
     def _merge_specs(self, batch):
@@ -498,6 +502,9 @@ class GpuHashAggregateExec(SparkPlan):
                     add(b, "max")
                 else:
                     add(b, "sum")
+        if not specs:
+            slot["__rows__"] = len(specs)
+            specs.append(("count*",))
         return specs, slot
 
     def _agg_batch(self, batch):
@@ -568,6 +575,8 @@ class GpuHashAggregateExec(SparkPlan):
         cols, validity = dict(key_cols), dict(key_valid)
         if self.mode == "partial":
             for name, j in slot.items():
+                if name == "__rows__":
+                    continue
                 cols[name] = accs[j]
                 # partial sum/min/max of an all-NULL group is NULL (its
                 # count companion is 0); counts themselves are never NULL

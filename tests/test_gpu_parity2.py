@@ -121,7 +121,7 @@ def test_composite_key_agg_parity(gq, nkeys):
     okeys, kmask, accs = gq.hash_agg_keys(
         [to_dev(k) for k in kc],
         [("sum", to_dev(vals), None), ("count*",)],
-        1 << 14,
+        1 << (14 if nkeys == 2 else 19),   # ~51^nkeys distinct tuples
         key_validities=[pack_validity(v) for v in kvalid])
     gk = [t.cpu().numpy() for t in okeys]
     gm = kmask.cpu().numpy()
