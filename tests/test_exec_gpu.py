@@ -185,3 +185,46 @@ def test_broadcast_hash_join_single_rank():
     o = np.lexsort((bkeys[ob], pkeys[op]))
     assert (got["lk"][g] == pkeys[op][o]).all()
     assert (got["rk"][g] == bkeys[ob][o]).all()
+
+
+def test_aqe_get_shuffle_partitions_coalesce():
+    """AQE getShuffleRDD(partitionSpecs) analog (ShuffleExchangeExec.scala:
+    141, ShuffledRowRDD.scala:33 CoalescedPartitionSpec): coalescing 8 map
+    partitions into 3 ranges is served straight from the partition-
+    contiguous map output — no re-partition kernel runs."""
+    from spark_amd import gpuq
+    n, nparts = 120_000, 8
+    keys = oracle.gen_i64(seed=400, n=n)
+    pay = oracle.gen_i64(seed=401, n=n)
+    valid = oracle.gen_i64(seed=402, n=n, range_=5) != 0
+    kd, pd = torch.from_numpy(keys).cuda(), torch.from_numpy(pay).cuda()
+    perm, counts = gpuq.partition_perm(kd, nparts)
+    cols = {"k": gpuq.gather(kd, perm), "p": gpuq.gather(pd, perm)}
+    vbits = torch.from_numpy(
+        np.packbits(valid, bitorder="little")).cuda()
+    vcols = {"p": gpuq.bits_to_u8(gpuq.gather_bits(vbits, perm), n)}
+    splits = counts.cpu().tolist()
+    offsets = [0]
+    for c in splits:
+        offsets.append(offsets[-1] + c)
+
+    node = gx.GpuShuffleExchangeExec(("k",), gx.InputBatches([]))
+    node._map_output = (cols, vcols, offsets)
+    specs = [(0, 3), (3, 4), (4, 8)]
+    batches = node.get_shuffle_partitions(specs)
+    pids = oracle.partition_ids(keys, nparts)
+    order = np.argsort(pids, kind="stable")
+    for (lo_p, hi_p), b in zip(specs, batches):
+        sel = order[(pids[order] >= lo_p) & (pids[order] < hi_p)]
+        assert b.num_rows() == len(sel)
+        got_k = b.column("k").cpu().numpy()
+        got_p = b.column("p").cpu().numpy()
+        assert (got_k == keys[sel]).all()
+        gv = b.validity("p")
+        got_valid = (np.unpackbits(gv.cpu().numpy(), count=len(sel),
+                                   bitorder="little").astype(bool)
+                     if gv is not None else np.ones(len(sel), bool))
+        assert (got_valid == valid[sel]).all()
+        assert (got_p[got_valid] == pay[sel][got_valid]).all()
+    # total coverage: the three ranges partition the full map output
+    assert sum(b.num_rows() for b in batches) == n
