@@ -530,6 +530,9 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
   __shared__ uint32_t bin_start[BINS];    /* in-block exclusive start per bin */
   __shared__ uint32_t bin_gbase[BINS];    /* global dest minus local start    */
   __shared__ uint32_t wtot[WAVES <= 4 ? 4 : WAVES];
+  /* narrow-digit cooperative lookback needs each bin's block count visible
+   * to its resolver wave */
+  __shared__ uint32_t bin_cnt[(LOOKBACK && RADIX_BITS < 8) ? BINS : 1];
   __shared__ uint64_t stage_k[TILE];
   __shared__ PayT stage_i[TILE];
 
@@ -605,6 +608,7 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
       __hip_atomic_store(&state[(int64_t)blockIdx.x * 256 + bin],
                          ((unsigned long long)epoch << 56) | OSW_AGG | acc,
                          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      if (RADIX_BITS < 8) bin_cnt[bin] = acc;
     }
     uint32_t inc = wave_inclusive_scan(acc);
     if (lane == WAVE - 1) wtot[wave] = inc;
@@ -634,7 +638,69 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
       stage_i[pos] = id[r];
     }
   }
-  if (LOOKBACK && tid < 256) {
+  if (LOOKBACK && RADIX_BITS < 8) {
+    /* full-wave cooperative lookback: each wave resolves whole bins — all
+     * 64 lanes poll one predecessor status word each per round (the
+     * per-thread serial walk left only BINS walker lanes active, which
+     * measured as the 4-bit mode's wall: 83% WAIT_ANY). Lane j inspects
+     * predecessor blockIdx.x-1-j; a ballot finds the newest PREFIX and the
+     * first not-ready word, a shuffle reduction sums the consumable run. */
+    constexpr int BPW = (BINS + WAVES - 1) / WAVES;
+    for (int bi = 0; bi < BPW; bi++) {
+      int bin = wave * BPW + bi;
+      if (bin >= BINS) break;
+      uint32_t excl = bin_gbase[bin];
+      unsigned long long pred = 0;
+      uint32_t spins = 0;
+      int p = (int)blockIdx.x - 1;
+      bool failed = false;
+      while (p >= 0) {
+        int cnt = (p + 1 < WAVE) ? p + 1 : WAVE;
+        unsigned long long v = 0;
+        bool ready = false, ispfx = false;
+        if (lane < cnt) {
+          v = __hip_atomic_load(&state[(int64_t)(p - lane) * 256 + bin],
+                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+          ready = OSW_EPOCH(v) == epoch && (v & (OSW_AGG | OSW_PFX));
+          ispfx = ready && (v & OSW_PFX);
+        }
+        uint64_t nr_mask = __ballot(lane < cnt && !ready);
+        uint64_t pfx_mask = __ballot(ispfx);
+        int first_nr = nr_mask ? __ffsll((unsigned long long)nr_mask) - 1 : 64;
+        int first_pfx = pfx_mask ? __ffsll((unsigned long long)pfx_mask) - 1 : 64;
+        bool done = first_pfx < first_nr;
+        int take = done ? first_pfx + 1 : (first_nr < cnt ? first_nr : cnt);
+        if (take == 0) {
+          if (++spins > OSW_SPIN_LIMIT) {
+            if (lane == 0)
+              __hip_atomic_store(err_flag, 1ull, __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT);
+            failed = true;
+            break;
+          }
+          __builtin_amdgcn_s_sleep(1);
+          continue;
+        }
+        unsigned long long contrib = (lane < take) ? OSW_VAL(v) : 0;
+        #pragma unroll
+        for (int off = WAVE / 2; off > 0; off >>= 1)
+          contrib += __shfl_down(contrib, off);
+        contrib = __shfl(contrib, 0);
+        pred += contrib;
+        spins = 0;
+        if (done) break;
+        p -= take;
+      }
+      (void)failed;
+      if (lane == 0) {
+        __hip_atomic_store(&state[(int64_t)blockIdx.x * 256 + bin],
+                           ((unsigned long long)epoch << 56) | OSW_PFX |
+                               (pred + bin_cnt[bin]),
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        bin_gbase[bin] = gbase[bin] + (uint32_t)pred - excl;
+      }
+    }
+  } else if (LOOKBACK && tid < 256) {
     int bin = tid;
     uint32_t excl = bin_gbase[bin];
     uint32_t own = acc;   /* this thread's phase-2 block count for its bin */
