@@ -86,18 +86,25 @@ class SortWorkload:
         self.out2 = torch.empty(rows, dtype=torch.float64, device="cuda")
         self.perm = torch.empty(rows, dtype=torch.int32, device="cuda")
         self.skeys = torch.empty(rows, dtype=torch.int64, device="cuda")
+        self.pair_scratch = torch.empty(rows * 2, dtype=torch.int64,
+                                        device="cuda")
 
     def step(self):
         gq = self.gq
         perm, skeys = gq.sort_perm(self.keys, workspace=self.ws,
                                    out_perm=self.perm, out_keys=self.skeys)
-        gq.lib().gpuq_gather2_i64(gq._stream(), self.rows,
-                                  self.pay1.data_ptr(), self.pay2.data_ptr(),
-                                  perm.data_ptr(), self.out1.data_ptr(),
-                                  self.out2.data_ptr())
+        # interleaved-pair payload gather: one random b128 load per row
+        # serves both payload columns (the interleave pass runs inside the
+        # step — it is part of the operator's work)
+        gq._check(gq.lib().gpuq_gather2_i64_fast(
+            gq._stream(), self.rows,
+            self.pay1.data_ptr(), self.pay2.data_ptr(),
+            perm.data_ptr(), self.out1.data_ptr(), self.out2.data_ptr(),
+            self.pair_scratch.data_ptr()))
 
     def free(self):
-        del self.keys, self.pay1, self.pay2, self.ws, self.out1, self.out2, self.perm, self.skeys
+        del self.keys, self.pay1, self.pay2, self.ws, self.out1, self.out2, \
+            self.perm, self.skeys, self.pair_scratch
 
 
 class AggWorkload:

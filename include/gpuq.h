@@ -114,6 +114,15 @@ int gpuq_gather(void* stream, int64_t nrows, gpuq_col col,
 int gpuq_gather2_i64(void* stream, int64_t nrows, const void* a, const void* b,
                      const uint32_t* perm, void* out_a, void* out_b);
 
+/* two 8-byte columns through one permutation via an interleaved staging
+ * buffer (scratch: nrows * 16 B device): one streaming interleave pass,
+ * then ONE random b128 load per row serving both columns — halves the
+ * random-request count and line amplification of the plain two-column
+ * gather. */
+int gpuq_gather2_i64_fast(void* stream, int64_t nrows, const void* a,
+                          const void* b, const uint32_t* perm,
+                          void* out_a, void* out_b, void* scratch);
+
 /* ---------------------------------------------------------------- */
 /* HASH AGGREGATE — replaces HashAggregateExec                       */
 /* (execution/aggregate/HashAggregateExec.scala:99-151) for          */
