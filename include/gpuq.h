@@ -160,12 +160,12 @@ int gpuq_hash_agg_i64_f64(void* stream, int64_t nrows,
                           int64_t* out_counts, int64_t* out_ngroups);
 
 /* Partitioned aggregation: same results contract as gpuq_hash_agg_i64_f64
- * for a single batch with NON-NULL values — rows are first ordered by a
- * 16-bit Murmur bucket (two stable ranked-scatter passes over (key,val)
- * pairs), then each chunk aggregates in an LDS table and merges once per
- * distinct key; sidesteps the global-atomic throughput wall at mid/high
- * cardinality. GPUQ_ERR_OVERFLOW with *out_ngroups = -1 => a chunk held
- * more distinct keys than the LDS table; fall back to the direct path. */
+ * for a single batch with NON-NULL values — rows are bucket-partitioned
+ * by 13 Murmur bits in ONE non-stable pass (histogram + per-block range
+ * reservation + contiguous per-bucket runs), then each 16 K-row chunk
+ * aggregates in an LDS table that flushes to the global table mid-chunk
+ * if a skewed/boundary chunk overflows it; sidesteps the global-atomic
+ * throughput wall at mid/high cardinality and never needs a fallback. */
 int64_t gpuq_hash_agg_part_workspace_bytes(int64_t nrows, int64_t capacity);
 int gpuq_hash_agg_partitioned(void* stream, int64_t nrows,
                               gpuq_col key, gpuq_col val,

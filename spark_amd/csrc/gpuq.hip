@@ -2323,85 +2323,96 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
 
 /* ---- partitioned aggregation (mid/high cardinality) ----
  * The direct table is atomic-throughput-bound (~22 G f64-adds/s at 10M
- * groups, tools/diag_hash.py). Here rows are first ordered by a 16-bit
- * Murmur bucket (two stable ranked-scatter passes carrying (key, val)
- * pairs), so any contiguous chunk holds few DISTINCT keys; each block then
- * aggregates its chunk in a small LDS table and merges once per distinct
- * key into the global table (a group spans at most a few chunks). */
+ * groups, tools/diag_hash.py). Round-2 design: ONE non-stable 13-bit
+ * bucket partition (8192 buckets; aggregation does not need stability, so
+ * the two 16-bit ranked-scatter passes of round 1 — 2x the scatter
+ * kernel's internal ceiling — are replaced by a histogram + reserve +
+ * direct scatter whose per-block per-bucket runs are ~0.5 KB contiguous
+ * stores), then per-chunk LDS-table aggregation: a 16 K-row chunk of
+ * bucket-ordered data holds at most ~n_groups/8192 (+ one bucket boundary)
+ * distinct keys, and a chunk that ever overflows the LDS table flushes it
+ * to the global table mid-chunk and continues (skew-safe, no global
+ * fallback). */
 
+#define PAGG_BUCKETS 8192
+#define PAGG_BUCKET_SHIFT 19      /* bucket = (mm3 >> 19) & 8191 */
+#define PAGG_TILE (256 * 1024)    /* rows partitioned per block */
+#define PAGG_SCATTER_BLOCK 512
 #define PAGG_CHUNK 16384
 #define PAGG_LDS_SLOTS 2048
 
-template <int OPS, int SLOT>
-__global__ __launch_bounds__(256)
-void k_agg_part(int64_t n, const uint64_t* keys, const uint64_t* vals,
-                unsigned long long* tab, agg_special* sp, int64_t cap_mask) {
-  __shared__ unsigned long long lt[PAGG_LDS_SLOTS * 3];
-  for (int j = threadIdx.x; j < PAGG_LDS_SLOTS * 3; j += blockDim.x)
-    lt[j] = (j % 3 == 0) ? AGG_EMPTY : 0;
-  __syncthreads();
-  const int64_t base = (int64_t)blockIdx.x * PAGG_CHUNK;
-  bool overflow = false;
-  for (int r = 0; r < PAGG_CHUNK / 256 && !overflow; r++) {
-    int64_t i = base + r * 256 + threadIdx.x;
-    if (i >= n) break;
-    if (keys[i] == AGG_EMPTY) continue;   /* special rows, handled upstream */
-    int64_t k = (int64_t)keys[i];
-    double v = __longlong_as_double((long long)vals[i]);
-    uint64_t slot = ((uint32_t)mm3_hash_long(k, 42)) & (PAGG_LDS_SLOTS - 1);
-    for (int probes = 0;; probes++) {
-      unsigned long long cur = lt[3 * slot];
-      if (cur == (unsigned long long)k) break;
-      if (cur == AGG_EMPTY) {
-        unsigned long long prev = atomicCAS(&lt[3 * slot], AGG_EMPTY,
-                                            (unsigned long long)k);
-        if (prev == AGG_EMPTY || prev == (unsigned long long)k) break;
-      }
-      slot = (slot + 1) & (PAGG_LDS_SLOTS - 1);
-      if (probes >= PAGG_LDS_SLOTS) { overflow = true; break; }
-    }
-    if (overflow) { atomicMax(&sp->overflow, 2ull); break; }
-    if (OPS & AGG_OP_SUM) atomicAdd((double*)&lt[3 * slot + 1], v);
-    if (OPS & AGG_OP_COUNT) atomicAdd(&lt[3 * slot + 2], 1ull);
-  }
-  __syncthreads();
-  /* merge occupied LDS slots into the global table */
-  for (int j = threadIdx.x; j < PAGG_LDS_SLOTS; j += blockDim.x) {
-    unsigned long long k = lt[3 * j];
-    if (k == AGG_EMPTY) continue;
-    uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)k, 42)) & (uint64_t)cap_mask;
-    for (int probes = 0;; probes++) {
-      unsigned long long cur = __hip_atomic_load(&tab[SLOT * slot], __ATOMIC_RELAXED,
-                                                 __HIP_MEMORY_SCOPE_AGENT);
-      if (cur == k) break;
-      if (cur == AGG_EMPTY) {
-        unsigned long long prev = atomicCAS(&tab[SLOT * slot], AGG_EMPTY, k);
-        if (prev == AGG_EMPTY || prev == k) break;
-      }
-      slot = (slot + 1) & (uint64_t)cap_mask;
-      if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
-    }
-    if (OPS & AGG_OP_SUM)
-      atomicAdd((double*)&tab[SLOT * slot + 1],
-                __longlong_as_double((long long)lt[3 * j + 1]));
-    if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], lt[3 * j + 2]);
-  }
+DEV int pagg_bucket(int64_t k) {
+  return (int)(((uint32_t)mm3_hash_long(k, 42) >> PAGG_BUCKET_SHIFT) &
+               (PAGG_BUCKETS - 1));
 }
 
-/* pack (key, f64-val-bits) pairs for the bucket ordering, position-aligned
- * with the input. NULL-key and real -1-key rows accumulate into the special
- * slots here and become EMPTY pairs (skipped by the chunk aggregator).
- * Requires non-null values (the partitioned path is gated on that). */
-template <int OPS>
-__global__ void k_agg_part_pairs(int64_t n, const int64_t* keys, const uint8_t* kvalid,
-                                 const double* vals,
-                                 uint64_t* out_k, uint64_t* out_v, agg_special* sp) {
+/* global bucket histogram; NULL-key rows count via key 0, real -1 keys via
+ * their own hash — the scatter uses identical bucketing and marks those
+ * records EMPTY, so layout and content agree. */
+__global__ void k_pagg_ghist(int64_t n, const int64_t* keys, const uint8_t* kvalid,
+                             unsigned int* ghist) {
+  __shared__ unsigned int h[PAGG_BUCKETS];
+  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x) h[b] = 0;
+  __syncthreads();
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += gs) {
+    int64_t k = bit_valid(kvalid, i) ? keys[i] : 0;
+    atomicAdd(&h[pagg_bucket(k)], 1u);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+    if (h[b]) atomicAdd(&ghist[b], h[b]);
+}
+
+/* single-block exclusive scan of the bucket histogram -> running cursors */
+__global__ void k_pagg_scan(const unsigned int* ghist, unsigned int* gcursor) {
+  __shared__ unsigned int buf[PAGG_BUCKETS];
+  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+    buf[b] = ghist[b];
+  __syncthreads();
+  if (threadIdx.x == 0) {           /* 8192 adds, once per aggregate call */
+    unsigned int run = 0;
+    for (int b = 0; b < PAGG_BUCKETS; b++) {
+      unsigned int t = buf[b];
+      buf[b] = run;
+      run += t;
+    }
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+    gcursor[b] = buf[b];
+}
+
+/* bucket scatter: per block, histogram its tile, reserve per-bucket ranges
+ * with ONE global atomicAdd per (block, bucket), then write (key, val)
+ * records bucket-contiguously. Special rows (NULL key / key == -1 ==
+ * AGG_EMPTY) accumulate into the special slots and leave EMPTY records. */
+template <int OPS>
+__global__ __launch_bounds__(PAGG_SCATTER_BLOCK)
+void k_pagg_scatter(int64_t n, const int64_t* keys, const uint8_t* kvalid,
+                    const double* vals, unsigned int* gcursor,
+                    ulonglong2* recs, agg_special* sp) {
+  __shared__ unsigned int h[PAGG_BUCKETS];
+  __shared__ unsigned int base[PAGG_BUCKETS];
+  const int64_t t0 = (int64_t)blockIdx.x * PAGG_TILE;
+  const int64_t t1 = min(t0 + (int64_t)PAGG_TILE, n);
+  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x) h[b] = 0;
+  __syncthreads();
+  for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x) {
+    int64_t k = bit_valid(kvalid, i) ? keys[i] : 0;
+    atomicAdd(&h[pagg_bucket(k)], 1u);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+    base[b] = h[b] ? atomicAdd(&gcursor[b], h[b]) : 0u;
+  __syncthreads();
+  for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x) {
     bool kv = bit_valid(kvalid, i);
     int64_t k = kv ? keys[i] : 0;
     double v = vals[i];
+    unsigned int pos = atomicAdd(&base[pagg_bucket(k)], 1u);
+    ulonglong2 rec;
     if (!kv || (unsigned long long)k == AGG_EMPTY) {
       unsigned long long* pseen = kv ? &sp->m1_seen : &sp->nul_seen;
       double* psum = kv ? &sp->m1_sum : &sp->nul_sum;
@@ -2409,52 +2420,113 @@ __global__ void k_agg_part_pairs(int64_t n, const int64_t* keys, const uint8_t* 
       atomicMax(pseen, 1ull);
       if (OPS & AGG_OP_SUM) atomicAdd(psum, v);
       if (OPS & AGG_OP_COUNT) atomicAdd(pcnt, 1ull);
-      out_k[i] = AGG_EMPTY;
-      out_v[i] = 0;
-      continue;
+      rec.x = AGG_EMPTY;
+      rec.y = 0;
+    } else {
+      rec.x = (unsigned long long)k;
+      rec.y = (unsigned long long)__double_as_longlong(v);
     }
-    out_k[i] = (uint64_t)k;
-    out_v[i] = (unsigned long long)__double_as_longlong(v);
+    recs[pos] = rec;
   }
 }
 
-/* u64-payload scatter launcher (fixed 512x8 geometry: 16 B/element staging
- * keeps two blocks per CU) */
-template <int BIN_MODE>
-static void launch_scatter64(hipStream_t s, int64_t nb, int64_t n,
-                             const uint64_t* kin, const uint64_t* iin,
-                             uint64_t* kout, uint64_t* iout,
-                             const uint32_t* scanned, int shift) {
-  k_radix_scatter<BIN_MODE, 512, 8, false, uint64_t><<<dim3((uint32_t)nb), 512, 0, s>>>(
-      n, kin, iin, kout, iout, scanned, shift, (int)nb, 0,
-      nullptr, nullptr, nullptr, 0, nullptr, 0);
+/* chunk aggregation over bucket-ordered records with mid-chunk flush: when
+ * any thread cannot place its key (LDS table full — a skewed or
+ * boundary-spanning chunk), the whole block merges the table into the
+ * global one, clears it, and the failed inserts retry. */
+template <int OPS, int SLOT>
+__global__ __launch_bounds__(256)
+void k_agg_part(int64_t n, const ulonglong2* recs,
+                unsigned long long* tab, agg_special* sp, int64_t cap_mask) {
+  __shared__ unsigned long long lt[PAGG_LDS_SLOTS * 3];
+  const int64_t base = (int64_t)blockIdx.x * PAGG_CHUNK;
+  for (int j = threadIdx.x; j < PAGG_LDS_SLOTS * 3; j += blockDim.x)
+    lt[j] = (j % 3 == 0) ? AGG_EMPTY : 0;
+  __syncthreads();
+  auto merge_flush = [&]() {
+    for (int j = threadIdx.x; j < PAGG_LDS_SLOTS; j += blockDim.x) {
+      unsigned long long k = lt[3 * j];
+      if (k == AGG_EMPTY) continue;
+      uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)k, 42)) & (uint64_t)cap_mask;
+      for (int64_t probes = 0;; probes++) {
+        unsigned long long cur = __hip_atomic_load(&tab[SLOT * slot], __ATOMIC_RELAXED,
+                                                   __HIP_MEMORY_SCOPE_AGENT);
+        if (cur == k) break;
+        if (cur == AGG_EMPTY) {
+          unsigned long long prev = atomicCAS(&tab[SLOT * slot], AGG_EMPTY, k);
+          if (prev == AGG_EMPTY || prev == k) break;
+        }
+        slot = (slot + 1) & (uint64_t)cap_mask;
+        if (probes > cap_mask) { atomicMax(&sp->overflow, 1ull); return; }
+      }
+      if (OPS & AGG_OP_SUM)
+        atomicAdd((double*)&tab[SLOT * slot + 1],
+                  __longlong_as_double((long long)lt[3 * j + 1]));
+      if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], lt[3 * j + 2]);
+    }
+  };
+  for (int r = 0; r < PAGG_CHUNK / 256; r++) {
+    int64_t i = base + (int64_t)r * 256 + threadIdx.x;
+    bool have = i < n;
+    ulonglong2 rec;
+    if (have) {
+      rec = recs[i];
+      have = rec.x != AGG_EMPTY;
+    }
+    for (;;) {
+      bool fail = false;
+      if (have) {
+        uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)rec.x, 42)) &
+                        (PAGG_LDS_SLOTS - 1);
+        int probes = 0;
+        for (;; probes++) {
+          unsigned long long cur = lt[3 * slot];
+          if (cur == rec.x) break;
+          if (cur == AGG_EMPTY) {
+            unsigned long long prev = atomicCAS(&lt[3 * slot], AGG_EMPTY, rec.x);
+            if (prev == AGG_EMPTY || prev == rec.x) break;
+          }
+          slot = (slot + 1) & (PAGG_LDS_SLOTS - 1);
+          if (probes >= PAGG_LDS_SLOTS) { fail = true; break; }
+        }
+        if (!fail) {
+          if (OPS & AGG_OP_SUM)
+            atomicAdd((double*)&lt[3 * slot + 1],
+                      __longlong_as_double((long long)rec.y));
+          if (OPS & AGG_OP_COUNT) atomicAdd(&lt[3 * slot + 2], 1ull);
+          have = false;
+        }
+      }
+      if (__syncthreads_count(fail ? 1 : 0) == 0) break;
+      merge_flush();
+      __syncthreads();
+      for (int j = threadIdx.x; j < PAGG_LDS_SLOTS * 3; j += blockDim.x)
+        lt[j] = (j % 3 == 0) ? AGG_EMPTY : 0;
+      __syncthreads();
+    }
+  }
+  __syncthreads();
+  merge_flush();
 }
 
 struct pagg_ws {
-  uint64_t *pk_a, *pv_a, *pk_b, *pv_b;
-  uint32_t *hist, *hist_scan, *block_sums;
+  ulonglong2* recs;
+  unsigned int* ghist;
+  unsigned int* gcursor;
   unsigned long long* tab;
   agg_special* sp;
 };
 
 static void pagg_ws_layout(int64_t n, int64_t cap, pagg_ws* w, char* base, int64_t* total) {
-  const int tile = 512 * 8;
-  int64_t nb = sort_nblocks(n, tile);
-  int64_t hist_n = 256 * nb;
-  int64_t scan_blocks = (hist_n + SCAN_TILE - 1) / SCAN_TILE + 1;
   int64_t off = 0;
   auto take = [&](int64_t bytes) {
     char* p = base ? base + off : nullptr;
     off += (bytes + 255) & ~255LL;
     return p;
   };
-  w->pk_a = (uint64_t*)take(n * 8);
-  w->pv_a = (uint64_t*)take(n * 8);
-  w->pk_b = (uint64_t*)take(n * 8);
-  w->pv_b = (uint64_t*)take(n * 8);
-  w->hist = (uint32_t*)take(hist_n * 4);
-  w->hist_scan = (uint32_t*)take(hist_n * 4);
-  w->block_sums = (uint32_t*)take(scan_blocks * 4);
+  w->recs = (ulonglong2*)take(n * 16);
+  w->ghist = (unsigned int*)take(PAGG_BUCKETS * 4);
+  w->gcursor = (unsigned int*)take(PAGG_BUCKETS * 4);
   w->tab = (unsigned long long*)take(cap * 24);
   w->sp = (agg_special*)take(sizeof(agg_special));
   *total = off;
@@ -2467,9 +2539,7 @@ extern "C" int64_t gpuq_hash_agg_part_workspace_bytes(int64_t n, int64_t cap) {
 }
 
 /* Partitioned aggregation: same results contract as gpuq_hash_agg_i64_f64
- * (single-shot; requires non-null values). Returns GPUQ_ERR_OVERFLOW with
- * *out_ngroups = -1 when a chunk exceeded the LDS table (caller falls back
- * to the direct path). */
+ * (single-shot; requires non-null values). */
 extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
                                          gpuq_col key, gpuq_col val,
                                          void* workspace, int64_t cap, int32_t ops,
@@ -2479,6 +2549,8 @@ extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
   hipStream_t s = (hipStream_t)stream;
   if (cap <= 0 || (cap & (cap - 1)))
     FAIL(GPUQ_ERR_INVALID, "pagg: capacity %lld not a power of two", (long long)cap);
+  if (n > 0xFFFFFFFELL)
+    FAIL(GPUQ_ERR_INVALID, "pagg: nrows %lld > 2^32 (u32 bucket cursors)", (long long)n);
   if (key.dtype != GPUQ_INT64 || val.dtype != GPUQ_FLOAT64)
     FAIL(GPUQ_ERR_INVALID, "pagg: expected int64 key + float64 val");
   if (val.validity) FAIL(GPUQ_ERR_INVALID, "pagg: values must be non-null");
@@ -2487,52 +2559,41 @@ extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
   k_agg_init<3><<<grid1d(cap), 256, 0, s>>>(cap, w.tab);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_special), s));
+  HIP_TRY(hipMemsetAsync(w.ghist, 0, PAGG_BUCKETS * 4, s));
   if (n > 0) {
     { hipEvent_t _pe = prof_begin(s);
-    if (ops == AGG_OP_SUM)
-      k_agg_part_pairs<AGG_OP_SUM><<<grid1d(n), 256, 0, s>>>(
-          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
-          w.pk_a, w.pv_a, w.sp);
-    else
-      k_agg_part_pairs<3><<<grid1d(n), 256, 0, s>>>(
-          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
-          w.pk_a, w.pv_a, w.sp);
-    prof_end("pagg_pairs", s, _pe); }
+    k_pagg_ghist<<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data,
+                                           key.validity, w.ghist);
+    prof_end("pagg_ghist", s, _pe); }
     HIP_TRY(hipGetLastError());
-    const int tile = 512 * 8;
-    int64_t nb = sort_nblocks(n, tile);
-    uint64_t *kin = w.pk_a, *vin = w.pv_a, *kout = w.pk_b, *vout = w.pv_b;
-    for (int p = 0; p < 2; p++) {
-      { hipEvent_t _pe = prof_begin(s);
-      k_radix_hist<3><<<dim3((uint32_t)nb), 256, 0, s>>>(n, kin, p * 8, w.hist, (int)nb, tile);
-      prof_end("pagg_hist", s, _pe); }
-      HIP_TRY(hipGetLastError());
-      int rc = exclusive_scan_u32(s, 256 * nb, w.hist, w.hist_scan, w.block_sums);
-      if (rc) return rc;
-      { hipEvent_t _pe = prof_begin(s);
-      launch_scatter64<3>(s, nb, n, kin, vin, kout, vout, w.hist_scan, p * 8);
-      prof_end("pagg_scatter", s, _pe); }
-      HIP_TRY(hipGetLastError());
-      uint64_t* t;
-      t = kin; kin = kout; kout = t;
-      t = vin; vin = vout; vout = t;
-    }
+    k_pagg_scan<<<1, 256, 0, s>>>(w.ghist, w.gcursor);
+    HIP_TRY(hipGetLastError());
+    int64_t nblocks = (n + PAGG_TILE - 1) / PAGG_TILE;
+    { hipEvent_t _pe = prof_begin(s);
+    if (ops == AGG_OP_SUM)
+      k_pagg_scatter<AGG_OP_SUM><<<dim3((uint32_t)nblocks), PAGG_SCATTER_BLOCK, 0, s>>>(
+          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
+          w.gcursor, w.recs, w.sp);
+    else
+      k_pagg_scatter<3><<<dim3((uint32_t)nblocks), PAGG_SCATTER_BLOCK, 0, s>>>(
+          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
+          w.gcursor, w.recs, w.sp);
+    prof_end("pagg_scatter", s, _pe); }
+    HIP_TRY(hipGetLastError());
     { hipEvent_t _pe = prof_begin(s);
     int64_t nchunks = (n + PAGG_CHUNK - 1) / PAGG_CHUNK;
     if (ops == AGG_OP_SUM)
       k_agg_part<AGG_OP_SUM, 3><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
-          n, kin, vin, w.tab, w.sp, cap - 1);
+          n, w.recs, w.tab, w.sp, cap - 1);
     else
       k_agg_part<3, 3><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
-          n, kin, vin, w.tab, w.sp, cap - 1);
+          n, w.recs, w.tab, w.sp, cap - 1);
     prof_end("pagg_chunks", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
   agg_special hsp;
   HIP_TRY(hipMemcpyAsync(&hsp, w.sp, sizeof(hsp), hipMemcpyDeviceToHost, s));
   HIP_TRY(hipStreamSynchronize(s));
-  if (hsp.overflow == 2ull) { *out_ngroups = -1;
-    FAIL(GPUQ_ERR_OVERFLOW, "pagg: chunk exceeded the LDS table (fall back to direct)"); }
   if (hsp.overflow) FAIL(GPUQ_ERR_OVERFLOW, "pagg: table overflow (capacity %lld)", (long long)cap);
   dim3 cgrid((uint32_t)((cap + AGGC_CHUNK - 1) / AGGC_CHUNK));
   if (ops == AGG_OP_SUM)

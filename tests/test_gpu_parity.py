@@ -396,12 +396,15 @@ def test_partitioned_agg_parity(gq, n, ngroups):
     agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt)
 
 
-def test_partitioned_agg_signals_lds_overflow(gq):
-    # nearly-all-distinct keys in a small input: chunks exceed the LDS table
-    # and the call must fail loudly with the fall-back signal
-    from spark_amd.gpuq import GpuqError
-    n = 100_000
-    keys = oracle.gen_i64(seed=1, n=n, range_=90_000)
+def test_partitioned_agg_flush_path(gq):
+    # nearly-all-distinct keys: chunks exceed the LDS table and must take
+    # the mid-chunk flush path (merge + clear + retry), staying correct —
+    # the round-1 design errored here; the round-2 one is skew-safe
+    n = 400_000
+    keys = oracle.gen_i64(seed=1, n=n, range_=380_000)
     vals = oracle.gen_f64_unit(seed=2, n=n)
-    with pytest.raises(GpuqError, match="fall back"):
-        gq.hash_agg_partitioned(to_dev(keys), to_dev(vals), 1 << 18)
+    cap = 1 << 20
+    gk, gkv, gs, gsv, gc = (t.cpu().numpy() for t in
+                            gq.hash_agg_partitioned(to_dev(keys), to_dev(vals), cap))
+    ok, okv, osum, osv, ocnt = oracle.hash_agg(keys, vals)
+    agg_compare(gk, gkv, gs, gsv, gc, ok, okv, osum, osv, ocnt)
