@@ -563,8 +563,11 @@ static int exclusive_scan_u32(hipStream_t s, int64_t n, const uint32_t* in,
 #define OSW_EPOCH(x) ((x) >> 56)
 #define OSW_SPIN_LIMIT (1u << 22)
 
+typedef unsigned long long ull2_a8 __attribute__((vector_size(16), aligned(8)));
+typedef uint32_t u32v2_a4 __attribute__((vector_size(8), aligned(4)));
+
 template <int BIN_MODE, int BLOCK, int ITEMS, bool LOOKBACK, typename PayT = uint32_t,
-          int RADIX_BITS = 8>
+          int RADIX_BITS = 8, bool CONTIG = false>
 __global__ __launch_bounds__(BLOCK)
 void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
                      uint64_t* kout, PayT* iout,
@@ -805,7 +808,50 @@ void k_radix_scatter(int64_t n, const uint64_t* kin, const PayT* iin,
   }
   __syncthreads();
 
-  /* drain LDS linearly -> coalesced global runs per bin */
+  /* drain LDS -> coalesced global runs per bin */
+  if (CONTIG) {
+    /* contiguous per-thread drain: each thread owns ITEMS consecutive
+     * staged elements; same-bin neighbours have consecutive destinations,
+     * so pairs collapse into ONE 16 B key store + ONE 8 B id store — the
+     * same bytes in HALF the store instructions (the padded-record test
+     * showed extra BYTES lose; this halves requests at equal bytes).
+     * Needs odd ITEMS for tolerable LDS bank conflicts (stride 2*ITEMS
+     * words, gcd 2 -> 2-way). PayT must be 4 B. */
+    auto out_word = [&](uint64_t kk) -> uint64_t {
+      if (!decode_mode) return kk;
+      uint64_t e = (decode_mode & 4) ? ~kk : kk;
+      if (decode_mode & 2) {
+        uint64_t mask = ((e >> 63) ? 0 : ~0ULL) | SIGNBIT;
+        return e ^ mask;
+      }
+      return e ^ SIGNBIT;
+    };
+    uint64_t* kbase = decode_mode ? (uint64_t*)decode_out : kout;
+    int j0 = tid * ITEMS;
+    int jend = j0 + ITEMS < tile_n ? j0 + ITEMS : tile_n;
+    for (int j = j0; j < jend;) {
+      uint64_t k0 = stage_k[j];
+      int b0 = compute_bin<BIN_MODE>(k0, shift, nparts);
+      if (RADIX_BITS < 8) b0 &= (1 << RADIX_BITS) - 1;
+      uint32_t dst = bin_gbase[b0] + (uint32_t)j;
+      if (j + 1 < jend) {
+        uint64_t k1 = stage_k[j + 1];
+        int b1 = compute_bin<BIN_MODE>(k1, shift, nparts);
+        if (RADIX_BITS < 8) b1 &= (1 << RADIX_BITS) - 1;
+        if (b1 == b0) {
+          *(ull2_a8*)&kbase[dst] = (ull2_a8){out_word(k0), out_word(k1)};
+          *(u32v2_a4*)&iout[dst] =
+              (u32v2_a4){(uint32_t)stage_i[j], (uint32_t)stage_i[j + 1]};
+          j += 2;
+          continue;
+        }
+      }
+      kbase[dst] = out_word(k0);
+      iout[dst] = stage_i[j];
+      j += 1;
+    }
+    return;
+  }
   for (int r = 0; r < ITEMS; r++) {
     int j = r * BLOCK + tid;
     if (j < tile_n) {
@@ -872,7 +918,28 @@ static void launch_scatter(hipStream_t s, scatter_geom g, int64_t nb,
 #define LS(B, I) k_radix_scatter<BIN_MODE, B, I, LOOKBACK><<<grid, B, 0, s>>>( \
       n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase, \
       err_flag, epoch, decode_out, decode_mode)
-  if (g.block == 256 && g.items == 16) LS(256, 16);
+  static int contig = -1;
+  if (contig < 0) contig = getenv("GPUQ_DRAIN_CONTIG") != nullptr;
+  if (contig && g.block == 512 && (g.items == 9 || g.items == 10 || g.items == 11)) {
+    if (g.items == 9)
+      k_radix_scatter<BIN_MODE, 512, 9, LOOKBACK, uint32_t, 8, true><<<grid, 512, 0, s>>>(
+          n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+          err_flag, epoch, decode_out, decode_mode);
+    else if (g.items == 11)
+      k_radix_scatter<BIN_MODE, 512, 11, LOOKBACK, uint32_t, 8, true><<<grid, 512, 0, s>>>(
+          n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+          err_flag, epoch, decode_out, decode_mode);
+    else
+      k_radix_scatter<BIN_MODE, 512, 10, LOOKBACK, uint32_t, 8, true><<<grid, 512, 0, s>>>(
+          n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+          err_flag, epoch, decode_out, decode_mode);
+    return;
+  }
+  if (g.block == 512 && g.items == 11)
+    k_radix_scatter<BIN_MODE, 512, 11, LOOKBACK><<<grid, 512, 0, s>>>(
+      n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+      err_flag, epoch, decode_out, decode_mode);
+  else if (g.block == 256 && g.items == 16) LS(256, 16);
   else if (g.block == 512 && g.items == 8) LS(512, 8);
   else if (g.block == 512 && g.items == 16) LS(512, 16);
   else if (g.block == 1024 && g.items == 8) LS(1024, 8);
@@ -2334,53 +2401,52 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
  * to the global table mid-chunk and continues (skew-safe, no global
  * fallback). */
 
-#define PAGG_BUCKETS 8192
-#define PAGG_BUCKET_SHIFT 19      /* bucket = (mm3 >> 19) & 8191 */
+#define PAGG_BUCKETS_MAX 16384
 #define PAGG_TILE (256 * 1024)    /* rows partitioned per block */
-#define PAGG_SCATTER_BLOCK 512
-#define PAGG_CHUNK 16384
-#define PAGG_LDS_SLOTS 2048
 
+template <int BUCKETS>
 DEV int pagg_bucket(int64_t k) {
-  return (int)(((uint32_t)mm3_hash_long(k, 42) >> PAGG_BUCKET_SHIFT) &
-               (PAGG_BUCKETS - 1));
+  /* top bits below the sign of the 32-bit murmur */
+  return (int)(((uint32_t)mm3_hash_long(k, 42) >> 12) & (BUCKETS - 1));
 }
 
 /* global bucket histogram; NULL-key rows count via key 0, real -1 keys via
  * their own hash — the scatter uses identical bucketing and marks those
  * records EMPTY, so layout and content agree. */
+template <int BUCKETS>
 __global__ void k_pagg_ghist(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                              unsigned int* ghist) {
-  __shared__ unsigned int h[PAGG_BUCKETS];
-  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x) h[b] = 0;
+  __shared__ unsigned int h[BUCKETS];
+  for (int b = threadIdx.x; b < BUCKETS; b += blockDim.x) h[b] = 0;
   __syncthreads();
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t gs = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += gs) {
     int64_t k = bit_valid(kvalid, i) ? keys[i] : 0;
-    atomicAdd(&h[pagg_bucket(k)], 1u);
+    atomicAdd(&h[pagg_bucket<BUCKETS>(k)], 1u);
   }
   __syncthreads();
-  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+  for (int b = threadIdx.x; b < BUCKETS; b += blockDim.x)
     if (h[b]) atomicAdd(&ghist[b], h[b]);
 }
 
 /* single-block exclusive scan of the bucket histogram -> running cursors */
+template <int BUCKETS>
 __global__ void k_pagg_scan(const unsigned int* ghist, unsigned int* gcursor) {
-  __shared__ unsigned int buf[PAGG_BUCKETS];
-  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+  __shared__ unsigned int buf[BUCKETS];
+  for (int b = threadIdx.x; b < BUCKETS; b += blockDim.x)
     buf[b] = ghist[b];
   __syncthreads();
-  if (threadIdx.x == 0) {           /* 8192 adds, once per aggregate call */
+  if (threadIdx.x == 0) {           /* serial adds, once per aggregate call */
     unsigned int run = 0;
-    for (int b = 0; b < PAGG_BUCKETS; b++) {
+    for (int b = 0; b < BUCKETS; b++) {
       unsigned int t = buf[b];
       buf[b] = run;
       run += t;
     }
   }
   __syncthreads();
-  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+  for (int b = threadIdx.x; b < BUCKETS; b += blockDim.x)
     gcursor[b] = buf[b];
 }
 
@@ -2388,30 +2454,30 @@ __global__ void k_pagg_scan(const unsigned int* ghist, unsigned int* gcursor) {
  * with ONE global atomicAdd per (block, bucket), then write (key, val)
  * records bucket-contiguously. Special rows (NULL key / key == -1 ==
  * AGG_EMPTY) accumulate into the special slots and leave EMPTY records. */
-template <int OPS>
-__global__ __launch_bounds__(PAGG_SCATTER_BLOCK)
+template <int OPS, int BUCKETS, int SBLOCK, int ST /*0 plain,1 sc1,2 nt*/>
+__global__ __launch_bounds__(SBLOCK)
 void k_pagg_scatter(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                     const double* vals, unsigned int* gcursor,
                     ulonglong2* recs, agg_special* sp) {
-  __shared__ unsigned int h[PAGG_BUCKETS];
-  __shared__ unsigned int base[PAGG_BUCKETS];
+  __shared__ unsigned int h[BUCKETS];
+  __shared__ unsigned int base[BUCKETS];
   const int64_t t0 = (int64_t)blockIdx.x * PAGG_TILE;
   const int64_t t1 = min(t0 + (int64_t)PAGG_TILE, n);
-  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x) h[b] = 0;
+  for (int b = threadIdx.x; b < BUCKETS; b += blockDim.x) h[b] = 0;
   __syncthreads();
   for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x) {
     int64_t k = bit_valid(kvalid, i) ? keys[i] : 0;
-    atomicAdd(&h[pagg_bucket(k)], 1u);
+    atomicAdd(&h[pagg_bucket<BUCKETS>(k)], 1u);
   }
   __syncthreads();
-  for (int b = threadIdx.x; b < PAGG_BUCKETS; b += blockDim.x)
+  for (int b = threadIdx.x; b < BUCKETS; b += blockDim.x)
     base[b] = h[b] ? atomicAdd(&gcursor[b], h[b]) : 0u;
   __syncthreads();
   for (int64_t i = t0 + threadIdx.x; i < t1; i += blockDim.x) {
     bool kv = bit_valid(kvalid, i);
     int64_t k = kv ? keys[i] : 0;
     double v = vals[i];
-    unsigned int pos = atomicAdd(&base[pagg_bucket(k)], 1u);
+    unsigned int pos = atomicAdd(&base[pagg_bucket<BUCKETS>(k)], 1u);
     ulonglong2 rec;
     if (!kv || (unsigned long long)k == AGG_EMPTY) {
       unsigned long long* pseen = kv ? &sp->m1_seen : &sp->nul_seen;
@@ -2426,7 +2492,20 @@ void k_pagg_scatter(int64_t n, const int64_t* keys, const uint8_t* kvalid,
       rec.x = (unsigned long long)k;
       rec.y = (unsigned long long)__double_as_longlong(v);
     }
-    recs[pos] = rec;
+    if (ST == 1) {
+      /* write-through: the per-bucket runs are written temporally
+       * scattered across the tile scan, so cached lines rarely collect a
+       * full 64 B before eviction — skip the write-allocate RFO */
+      __hip_atomic_store(&recs[pos].x, rec.x, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_store(&recs[pos].y, rec.y, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    } else if (ST == 2) {
+      __builtin_nontemporal_store(rec.x, &recs[pos].x);
+      __builtin_nontemporal_store(rec.y, &recs[pos].y);
+    } else {
+      recs[pos] = rec;
+    }
   }
 }
 
@@ -2434,17 +2513,17 @@ void k_pagg_scatter(int64_t n, const int64_t* keys, const uint8_t* kvalid,
  * any thread cannot place its key (LDS table full — a skewed or
  * boundary-spanning chunk), the whole block merges the table into the
  * global one, clears it, and the failed inserts retry. */
-template <int OPS, int SLOT>
-__global__ __launch_bounds__(256)
+template <int OPS, int SLOT, int SLOTS, int CHUNK, int CBLOCK>
+__global__ __launch_bounds__(CBLOCK)
 void k_agg_part(int64_t n, const ulonglong2* recs,
                 unsigned long long* tab, agg_special* sp, int64_t cap_mask) {
-  __shared__ unsigned long long lt[PAGG_LDS_SLOTS * 3];
-  const int64_t base = (int64_t)blockIdx.x * PAGG_CHUNK;
-  for (int j = threadIdx.x; j < PAGG_LDS_SLOTS * 3; j += blockDim.x)
+  __shared__ unsigned long long lt[SLOTS * 3];
+  const int64_t base = (int64_t)blockIdx.x * CHUNK;
+  for (int j = threadIdx.x; j < SLOTS * 3; j += blockDim.x)
     lt[j] = (j % 3 == 0) ? AGG_EMPTY : 0;
   __syncthreads();
   auto merge_flush = [&]() {
-    for (int j = threadIdx.x; j < PAGG_LDS_SLOTS; j += blockDim.x) {
+    for (int j = threadIdx.x; j < SLOTS; j += blockDim.x) {
       unsigned long long k = lt[3 * j];
       if (k == AGG_EMPTY) continue;
       uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)k, 42)) & (uint64_t)cap_mask;
@@ -2465,19 +2544,29 @@ void k_agg_part(int64_t n, const ulonglong2* recs,
       if (OPS & AGG_OP_COUNT) atomicAdd(&tab[SLOT * slot + 2], lt[3 * j + 2]);
     }
   };
-  for (int r = 0; r < PAGG_CHUNK / 256; r++) {
-    int64_t i = base + (int64_t)r * 256 + threadIdx.x;
-    bool have = i < n;
-    ulonglong2 rec;
-    if (have) {
-      rec = recs[i];
-      have = rec.x != AGG_EMPTY;
+  static_assert((CHUNK / CBLOCK) % 4 == 0, "round grouping");
+  for (int r0 = 0; r0 < CHUNK / CBLOCK; r0 += 4) {
+    /* batch 4 record loads in flight before the dependent hash->LDS-probe
+     * chain (the LDS table path is latency-, not queue-bound) */
+    ulonglong2 recv[4];
+    bool havev[4];
+    #pragma unroll
+    for (int q = 0; q < 4; q++) {
+      int64_t i = base + (int64_t)(r0 + q) * CBLOCK + threadIdx.x;
+      havev[q] = i < n;
+      if (havev[q]) recv[q] = recs[i];
     }
+    #pragma unroll
+    for (int q = 0; q < 4; q++)
+      havev[q] = havev[q] && recv[q].x != AGG_EMPTY;
+    for (int q = 0; q < 4; q++) {
+    ulonglong2 rec = recv[q];
+    bool have = havev[q];
     for (;;) {
       bool fail = false;
       if (have) {
         uint64_t slot = ((uint32_t)mm3_hash_long((int64_t)rec.x, 42)) &
-                        (PAGG_LDS_SLOTS - 1);
+                        (SLOTS - 1);
         int probes = 0;
         for (;; probes++) {
           unsigned long long cur = lt[3 * slot];
@@ -2486,8 +2575,8 @@ void k_agg_part(int64_t n, const ulonglong2* recs,
             unsigned long long prev = atomicCAS(&lt[3 * slot], AGG_EMPTY, rec.x);
             if (prev == AGG_EMPTY || prev == rec.x) break;
           }
-          slot = (slot + 1) & (PAGG_LDS_SLOTS - 1);
-          if (probes >= PAGG_LDS_SLOTS) { fail = true; break; }
+          slot = (slot + 1) & (SLOTS - 1);
+          if (probes >= SLOTS) { fail = true; break; }
         }
         if (!fail) {
           if (OPS & AGG_OP_SUM)
@@ -2500,9 +2589,10 @@ void k_agg_part(int64_t n, const ulonglong2* recs,
       if (__syncthreads_count(fail ? 1 : 0) == 0) break;
       merge_flush();
       __syncthreads();
-      for (int j = threadIdx.x; j < PAGG_LDS_SLOTS * 3; j += blockDim.x)
+      for (int j = threadIdx.x; j < SLOTS * 3; j += blockDim.x)
         lt[j] = (j % 3 == 0) ? AGG_EMPTY : 0;
       __syncthreads();
+    }
     }
   }
   __syncthreads();
@@ -2525,8 +2615,8 @@ static void pagg_ws_layout(int64_t n, int64_t cap, pagg_ws* w, char* base, int64
     return p;
   };
   w->recs = (ulonglong2*)take(n * 16);
-  w->ghist = (unsigned int*)take(PAGG_BUCKETS * 4);
-  w->gcursor = (unsigned int*)take(PAGG_BUCKETS * 4);
+  w->ghist = (unsigned int*)take(PAGG_BUCKETS_MAX * 4);
+  w->gcursor = (unsigned int*)take(PAGG_BUCKETS_MAX * 4);
   w->tab = (unsigned long long*)take(cap * 24);
   w->sp = (agg_special*)take(sizeof(agg_special));
   *total = off;
@@ -2559,35 +2649,74 @@ extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
   k_agg_init<3><<<grid1d(cap), 256, 0, s>>>(cap, w.tab);
   HIP_TRY(hipGetLastError());
   HIP_TRY(hipMemsetAsync(w.sp, 0, sizeof(agg_special), s));
-  HIP_TRY(hipMemsetAsync(w.ghist, 0, PAGG_BUCKETS * 4, s));
+  HIP_TRY(hipMemsetAsync(w.ghist, 0, PAGG_BUCKETS_MAX * 4, s));
+  /* variant sweep knob: GPUQ_PAGG=0 (8192 buckets / 2048-slot 16K chunks)
+   * 1 (16384 buckets / 4096-slot chunks) 2 (8192 buckets / 4096-slot) */
+  static int variant = -1;
+  if (variant < 0) {
+    const char* e = getenv("GPUQ_PAGG");
+    variant = e ? atoi(e) : 2;  /* 8192 buckets / 4096-slot chunks measured best */
+  }
   if (n > 0) {
     { hipEvent_t _pe = prof_begin(s);
-    k_pagg_ghist<<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data,
-                                           key.validity, w.ghist);
+    if (variant == 1)
+      k_pagg_ghist<16384><<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data,
+                                                    key.validity, w.ghist);
+    else
+      k_pagg_ghist<8192><<<grid1d(n), 256, 0, s>>>(n, (const int64_t*)key.data,
+                                                   key.validity, w.ghist);
     prof_end("pagg_ghist", s, _pe); }
     HIP_TRY(hipGetLastError());
-    k_pagg_scan<<<1, 256, 0, s>>>(w.ghist, w.gcursor);
+    if (variant == 1)
+      k_pagg_scan<16384><<<1, 256, 0, s>>>(w.ghist, w.gcursor);
+    else
+      k_pagg_scan<8192><<<1, 256, 0, s>>>(w.ghist, w.gcursor);
     HIP_TRY(hipGetLastError());
     int64_t nblocks = (n + PAGG_TILE - 1) / PAGG_TILE;
+    static int st = -1;
+    if (st < 0) {
+      const char* e = getenv("GPUQ_PAGG_ST");
+      st = e ? atoi(e) : 0;  /* plain stores measured best (sc1 +105%, nt +120%) */
+    }
     { hipEvent_t _pe = prof_begin(s);
-    if (ops == AGG_OP_SUM)
-      k_pagg_scatter<AGG_OP_SUM><<<dim3((uint32_t)nblocks), PAGG_SCATTER_BLOCK, 0, s>>>(
-          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
-          w.gcursor, w.recs, w.sp);
-    else
-      k_pagg_scatter<3><<<dim3((uint32_t)nblocks), PAGG_SCATTER_BLOCK, 0, s>>>(
-          n, (const int64_t*)key.data, key.validity, (const double*)val.data,
-          w.gcursor, w.recs, w.sp);
+    #define PAGG_LAUNCH(OPS_, B_, T_) do { \
+      if (st == 1) k_pagg_scatter<OPS_, B_, T_, 1><<<dim3((uint32_t)nblocks), T_, 0, s>>>( \
+          n, (const int64_t*)key.data, key.validity, (const double*)val.data, \
+          w.gcursor, w.recs, w.sp); \
+      else if (st == 2) k_pagg_scatter<OPS_, B_, T_, 2><<<dim3((uint32_t)nblocks), T_, 0, s>>>( \
+          n, (const int64_t*)key.data, key.validity, (const double*)val.data, \
+          w.gcursor, w.recs, w.sp); \
+      else k_pagg_scatter<OPS_, B_, T_, 0><<<dim3((uint32_t)nblocks), T_, 0, s>>>( \
+          n, (const int64_t*)key.data, key.validity, (const double*)val.data, \
+          w.gcursor, w.recs, w.sp); \
+    } while (0)
+    if (variant == 1) {
+      if (ops == AGG_OP_SUM) PAGG_LAUNCH(AGG_OP_SUM, 16384, 1024);
+      else PAGG_LAUNCH(3, 16384, 1024);
+    } else {
+      if (ops == AGG_OP_SUM) PAGG_LAUNCH(AGG_OP_SUM, 8192, 512);
+      else PAGG_LAUNCH(3, 8192, 512);
+    }
+    #undef PAGG_LAUNCH
     prof_end("pagg_scatter", s, _pe); }
     HIP_TRY(hipGetLastError());
     { hipEvent_t _pe = prof_begin(s);
-    int64_t nchunks = (n + PAGG_CHUNK - 1) / PAGG_CHUNK;
-    if (ops == AGG_OP_SUM)
-      k_agg_part<AGG_OP_SUM, 3><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
-          n, w.recs, w.tab, w.sp, cap - 1);
-    else
-      k_agg_part<3, 3><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
-          n, w.recs, w.tab, w.sp, cap - 1);
+    int64_t nchunks = (n + 16384 - 1) / 16384;
+    if (variant >= 1) {
+      if (ops == AGG_OP_SUM)
+        k_agg_part<AGG_OP_SUM, 3, 4096, 16384, 512><<<dim3((uint32_t)nchunks), 512, 0, s>>>(
+            n, w.recs, w.tab, w.sp, cap - 1);
+      else
+        k_agg_part<3, 3, 4096, 16384, 512><<<dim3((uint32_t)nchunks), 512, 0, s>>>(
+            n, w.recs, w.tab, w.sp, cap - 1);
+    } else {
+      if (ops == AGG_OP_SUM)
+        k_agg_part<AGG_OP_SUM, 3, 2048, 16384, 256><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
+            n, w.recs, w.tab, w.sp, cap - 1);
+      else
+        k_agg_part<3, 3, 2048, 16384, 256><<<dim3((uint32_t)nchunks), 256, 0, s>>>(
+            n, w.recs, w.tab, w.sp, cap - 1);
+    }
     prof_end("pagg_chunks", s, _pe); }
     HIP_TRY(hipGetLastError());
   }

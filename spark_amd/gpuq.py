@@ -242,9 +242,20 @@ def hash_agg(keys: torch.Tensor, vals: torch.Tensor, capacity: int,
              workspace=None, max_groups=None,
              key_validity=None, val_validity=None, ops=AGG_SUM | AGG_COUNT):
     """One-shot aggregate of a single batch. Returns (keys, key_valid, sums,
-    sum_valid, counts) tensors sliced to ngroups."""
+    sum_valid, counts) tensors sliced to ngroups.
+
+    Mid-cardinality routing: when the capacity hint falls in the band where
+    the direct table's atomic contention dominates (measured on MI355X at
+    1B rows: 100K groups 60->40 ms, 1M groups 49->43 ms) and the batch is
+    single-shot with non-null values, the bucket-partitioned kernel runs
+    instead (identical results contract)."""
     n = keys.numel()
     dev = keys.device
+    if (val_validity is None and (1 << 16) <= capacity <= (1 << 22)
+            and n >= (1 << 24) and os.environ.get("GPUQ_NO_PART_AGG") is None):
+        return hash_agg_partitioned(keys, vals, capacity,
+                                    max_groups=max_groups,
+                                    key_validity=key_validity, ops=ops)
     if workspace is None:
         workspace = agg_workspace(capacity, dev)
     mg = max_groups if max_groups is not None else min(n + 2, capacity + 2)

@@ -4,7 +4,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from spark_amd import gpuq as gq
 
-n, groups = 1_000_000_000, 10_000_000
+n, groups = 1_000_000_000, int(os.environ.get("AB_GROUPS", 10_000_000))
 cap = 1 << (groups * 2 - 1).bit_length()
 keys = gq.gen_i64(seed=52, n=n, range_=groups)
 vals = gq.gen_f64_unit(seed=53, n=n)
@@ -38,7 +38,8 @@ def part():
 
 gq.profiling(True)
 t("direct", direct)
-t("partitioned", part)
+os.environ["GPUQ_PAGG"] = os.environ.get("AB_PAGG", "1")
+t("partitioned v" + os.environ["GPUQ_PAGG"], part)
 for k in ("pagg_ghist", "pagg_scatter", "pagg_chunks", "agg_build"):
     ms, cnt = gq.kernel_stats(k)
     if cnt: print(f"  {k}: {ms/cnt:.2f} ms avg x{cnt}")
@@ -50,3 +51,5 @@ o1 = torch.argsort(k1); o2 = torch.argsort(k2)
 assert torch.equal(k1[o1], k2[o2])
 assert torch.allclose(s1[o1], s2[o2], rtol=1e-9)
 print(f"parity OK ({g1} groups)")
+
+# variant sweep via GPUQ_PAGG (static-cached per process: report only)
