@@ -902,6 +902,20 @@ static scatter_geom get_sort_geom(void) {
     int b = 0, it = 0;
     if (e && sscanf(e, "%dx%d", &b, &it) == 2) { g.block = b; g.items = it; }
     else { g.block = 512; g.items = 10; }
+    /* only geometries with a dispatch entry are legal: an unknown pair
+     * would silently run a different tile than the block count assumed
+     * (incomplete sort). Clamp to the default. */
+    static const int known[][2] = {{256,16},{512,8},{512,16},{1024,8},{512,4},
+                                   {512,6},{256,12},{512,12},{1024,5},{512,10},
+                                   {1024,2},{1024,6},{1024,4},{512,11},
+                                   {256,20},{256,22}};
+    bool ok = false;
+    for (auto& k : known) ok = ok || (k[0] == g.block && k[1] == g.items);
+    if (!ok) {
+      fprintf(stderr, "gpuq: unsupported GPUQ_SORT_GEOM %dx%d, using 512x10\n",
+              g.block, g.items);
+      g.block = 512; g.items = 10;
+    }
   }
   return g;
 }
@@ -935,7 +949,15 @@ static void launch_scatter(hipStream_t s, scatter_geom g, int64_t nb,
           err_flag, epoch, decode_out, decode_mode);
     return;
   }
-  if (g.block == 512 && g.items == 11)
+  if (g.block == 256 && g.items == 20)
+    k_radix_scatter<BIN_MODE, 256, 20, LOOKBACK><<<grid, 256, 0, s>>>(
+      n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+      err_flag, epoch, decode_out, decode_mode);
+  else if (g.block == 256 && g.items == 22)
+    k_radix_scatter<BIN_MODE, 256, 22, LOOKBACK><<<grid, 256, 0, s>>>(
+      n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+      err_flag, epoch, decode_out, decode_mode);
+  else if (g.block == 512 && g.items == 11)
     k_radix_scatter<BIN_MODE, 512, 11, LOOKBACK><<<grid, 512, 0, s>>>(
       n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
       err_flag, epoch, decode_out, decode_mode);

@@ -23,9 +23,10 @@ def t(label, reps=5):
 gq.profiling(True)
 t(f"geom={os.environ.get('GPUQ_SORT_GEOM','512x10')} contig={bool(os.environ.get('GPUQ_DRAIN_CONTIG'))}")
 # parity vs reference run handled by the existing suite; quick check here:
-m = 50_000_000
-kk = gq.gen_i64(seed=9, n=m)
-ws2 = gq.sort_workspace(m)
-p1, _ = gq.sort_perm(kk, workspace=ws2, out_keys=False)
-import subprocess
-print("done")
+# verify the result IS a sorted permutation (a silent geometry fallback
+# once produced fast-but-wrong timings — never trust a sweep without this)
+perm, skeys = gq.sort_perm(keys, workspace=ws)
+assert bool((skeys[1:] >= skeys[:-1]).all()), "output not sorted!"
+idx = torch.randint(0, n, (1_000_000,), device="cuda")
+assert bool((skeys[idx] == keys[perm[idx].long()]).all()), "not a permutation!"
+print("sorted-permutation check OK")
