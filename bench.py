@@ -293,22 +293,34 @@ class GlobalSortWorkload:
 
 
 class Q1Workload:
-    """Config-5 progress probe (single-GPU slice of a TPC-H Q1-shaped plan):
-    filter (shipdate <= cut) -> project (disc_price = price*(1-disc)) ->
-    GROUP BY returnflag (low cardinality => the LDS pre-agg path) with
-    SUM(disc_price) + COUNT. Driven through the exec-node layer (the same
-    operator objects a ColumnarRule would produce)."""
+    """Config-5 single-GPU slice: the REAL TPC-H Q1 shape
+    (resources/tpch/q1.sql) over a lineitem-shaped synthetic batch —
+    decimal(12,2) columns as scaled int64 (Spark's compact-long Decimal),
+    shipdate as int64 days, returnflag/linestatus as dictionary ids.
+    Plan: filter (l_shipdate <= date'1998-12-01' - 90 days) -> projected
+    decimal products (scale-4 disc_price, scale-6 charge, exact i64) ->
+    GROUP BY (l_returnflag, l_linestatus) — the composite tuple rides the
+    narrow-key pack rule into the LDS low-cardinality path — with
+    sum_qty/sum_base_price/sum_disc_price/sum_charge (wrapping-exact i64),
+    avg_disc (f64 of exact sums) and count_order, then ORDER BY the
+    grouping keys. Driven through the exec-node layer (the same operator
+    objects a ColumnarRule would produce)."""
 
     def __init__(self, gq, rows, rank):
         from spark_amd import exec as gx
         self.gx = gx
         off = rank * 7
         self.cols = dict(
-            retflag=gq.gen_i64(seed=71 + off, n=rows, range_=6),
-            qty=gq.gen_f64_unit(seed=72 + off, n=rows),
-            price=gq.gen_f64_unit(seed=73 + off, n=rows),
-            disc=gq.gen_f64_unit(seed=74 + off, n=rows),
-            shipdate=gq.gen_f64_unit(seed=75 + off, n=rows),
+            l_returnflag=gq.gen_i64(seed=71 + off, n=rows, range_=3),
+            l_linestatus=gq.gen_i64(seed=76 + off, n=rows, range_=2),
+            l_quantity=gq.gen_i64(seed=72 + off, n=rows, range_=5000),
+            l_extendedprice=gq.gen_i64(seed=73 + off, n=rows,
+                                       range_=10_000_000),
+            l_discount=gq.gen_i64(seed=74 + off, n=rows, range_=11),
+            l_tax=gq.gen_i64(seed=77 + off, n=rows, range_=9),
+            # days offset into the 1992..1998 window (values 0..2555);
+            # the Q1 cutoff (date '1998-12-01' - 90 days) sits at ~95%
+            l_shipdate=gq.gen_i64(seed=75 + off, n=rows, range_=2556),
         )
         self.rows = rows
         self.ngroups = 0
@@ -316,22 +328,33 @@ class Q1Workload:
     def step(self):
         gx = self.gx
         scan = gx.InputBatches([gx.ColumnarBatch(dict(self.cols))])
-        plan = gx.HashAggregateExec(
-            "retflag",
-            [("sum", "qty"), ("sum", "price"), ("sum", "disc_price"),
-             ("avg", "disc"), ("count", "disc_price")],
-            "complete",
-            gx.ProjectExec(
-                ["retflag", "qty", "price", "disc",
-                 ("disc_price", "price", "*", "one_minus_disc", None)],
+        passthru = ["l_returnflag", "l_linestatus", "l_quantity",
+                    "l_extendedprice", "l_discount"]
+        plan = gx.SortExec(
+            [gx.SortOrder("l_returnflag"), gx.SortOrder("l_linestatus")],
+            False,
+            gx.HashAggregateExec(
+                ("l_returnflag", "l_linestatus"),
+                [("sum", "l_quantity"), ("sum", "l_extendedprice"),
+                 ("sum", "disc_price"), ("sum", "charge"),
+                 ("avg", "l_discount"), ("count*", None)],
+                "complete",
                 gx.ProjectExec(
-                    ["retflag", "qty", "price", "disc",
-                     ("one_minus_disc", "neg_disc", "+", None, 1)],
+                    passthru + ["disc_price",
+                                ("charge", "disc_price", "*",
+                                 "__one_plus_tax", None)],
                     gx.ProjectExec(
-                        ["retflag", "qty", "price", "disc",
-                         ("neg_disc", "disc", "*", None, -1)],
-                        gx.FilterExec("shipdate", "<=", 0.98, scan)))),
-            capacity=64)
+                        passthru + ["__one_plus_tax",
+                                    ("disc_price", "l_extendedprice", "*",
+                                     "__one_minus_disc", None)],
+                        gx.ProjectExec(
+                            passthru + [
+                                ("__one_minus_disc", "l_discount", "rsub",
+                                 None, 100),
+                                ("__one_plus_tax", "l_tax", "+", None, 100)],
+                            gx.FilterExec("l_shipdate", "<=", 2435,
+                                          scan)))),
+                capacity=64))
         plan = gx.GpuColumnarRule().pre_columnar_transitions(plan)
         out = next(plan.execute_columnar())
         self.ngroups = out.num_rows()
@@ -497,7 +520,9 @@ def main():
             torch.cuda.empty_cache()
             results["q1"] = {"sec_per_step": sec,
                              "rows_per_sec": args.join_rows * world / sec,
-                             "plan": "filter->project x3->groupby(3 sums, avg, count) one pass",
+                             "plan": "real Q1: filter(shipdate)->decimal "
+                                     "projects->groupby(retflag,linestatus) "
+                                     "4 sums+avg+count*->orderby",
                              "ngroups": ng}
 
 

@@ -2773,7 +2773,7 @@ extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
  * non-null (per-acc NULL tracking pairs a SUM with a COUNT spec).
  */
 
-#define AGG_MAX_SPECS 6
+#define AGG_MAX_SPECS 10
 
 struct agg_cols { const double* p[AGG_MAX_SPECS]; };
 /* ops (Sum.scala:113-180, Count.scala, Min/Max.scala null-skipping update
@@ -3737,13 +3737,14 @@ __global__ void k_project_binop(int64_t n, const void* a, const void* b,
     if (DTYPE == GPUQ_FLOAT64) {
       double x = ((const double*)a)[i];
       double y = B_IS_LIT ? lit_f : ((const double*)b)[i];
-      double r = OP == 0 ? x + y : OP == 1 ? x - y : OP == 2 ? x * y : x / y;
+      double r = OP == 0 ? x + y : OP == 1 ? x - y : OP == 2 ? x * y
+                 : OP == 4 ? y - x : x / y;
       ((double*)out)[i] = r;
     } else {
       int64_t x = ((const int64_t*)a)[i];
       int64_t y = B_IS_LIT ? lit_i : ((const int64_t*)b)[i];
       int64_t r = OP == 0 ? x + y : OP == 1 ? x - y : OP == 2 ? x * y
-                  : (y == 0 ? 0 : x / y);
+                  : OP == 4 ? y - x : (y == 0 ? 0 : x / y);
       ((int64_t*)out)[i] = r;
     }
   }
@@ -3754,7 +3755,7 @@ extern "C" int gpuq_project_binop(void* stream, int64_t n, gpuq_col a,
                                   double lit_f, int64_t lit_i, int32_t op,
                                   void* out) {
   hipStream_t s = (hipStream_t)stream;
-  if (op < 0 || op > 3) FAIL(GPUQ_ERR_INVALID, "project: bad op %d", op);
+  if (op < 0 || op > 4) FAIL(GPUQ_ERR_INVALID, "project: bad op %d", op);
   if (a.validity) FAIL(GPUQ_ERR_INVALID, "project: validity not yet supported");
   dim3 g = grid1d(n);
 #define PJ(DT, OPV) do { \
@@ -3763,10 +3764,12 @@ extern "C" int gpuq_project_binop(void* stream, int64_t n, gpuq_col a,
   } while (0)
   if (a.dtype == GPUQ_FLOAT64) {
     if (op == 0) PJ(GPUQ_FLOAT64, 0); else if (op == 1) PJ(GPUQ_FLOAT64, 1);
-    else if (op == 2) PJ(GPUQ_FLOAT64, 2); else PJ(GPUQ_FLOAT64, 3);
+    else if (op == 2) PJ(GPUQ_FLOAT64, 2); else if (op == 4) PJ(GPUQ_FLOAT64, 4);
+    else PJ(GPUQ_FLOAT64, 3);
   } else if (a.dtype == GPUQ_INT64) {
     if (op == 0) PJ(GPUQ_INT64, 0); else if (op == 1) PJ(GPUQ_INT64, 1);
-    else if (op == 2) PJ(GPUQ_INT64, 2); else PJ(GPUQ_INT64, 3);
+    else if (op == 2) PJ(GPUQ_INT64, 2); else if (op == 4) PJ(GPUQ_INT64, 4);
+    else PJ(GPUQ_INT64, 3);
   } else {
     FAIL(GPUQ_ERR_INVALID, "project: unsupported dtype %d", a.dtype);
   }

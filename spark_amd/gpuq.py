@@ -330,7 +330,7 @@ def join_probe(probe_keys: torch.Tensor, workspace: torch.Tensor, capacity: int,
 
 
 CMP = {"==": 0, "<": 1, "<=": 2, ">": 3, ">=": 4, "!=": 5}
-BINOP = {"+": 0, "-": 1, "*": 2, "/": 3}
+BINOP = {"+": 0, "-": 1, "*": 2, "/": 3, "rsub": 4}
 
 
 def filter_cmp(col: torch.Tensor, op: str, literal, workspace=None, validity=None):
