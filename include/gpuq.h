@@ -182,7 +182,7 @@ int gpuq_hash_agg_partitioned(void* stream, int64_t nrows,
  * overflow wraps), 4=MIN(int64), 5=MAX(int64), 6=MIN(float64),
  * 7=MAX(float64) (Min/Max.scala; float ordering = Java Double.compare:
  * NaN greatest, -0.0 < 0.0);
- * up to 10 specs; spec_cols[j] indexes vals[] (ignored for COUNT(*)). out_accs[j] is a
+ * up to 12 specs; spec_cols[j] indexes vals[] (ignored for COUNT(*)). out_accs[j] is a
  * device array per spec: f64 for SUM_F64/MIN_F64/MAX_F64, i64 otherwise.
  * Small tables (cap*(1+nspecs)*8 <= 64 KB) aggregate per-block in LDS
  * first. Value ops skip NULL rows; an all-NULL group emits the op's

@@ -2773,7 +2773,7 @@ extern "C" int gpuq_hash_agg_partitioned(void* stream, int64_t n,
  * non-null (per-acc NULL tracking pairs a SUM with a COUNT spec).
  */
 
-#define AGG_MAX_SPECS 10
+#define AGG_MAX_SPECS 12
 
 struct agg_cols { const double* p[AGG_MAX_SPECS]; };
 /* ops (Sum.scala:113-180, Count.scala, Min/Max.scala null-skipping update

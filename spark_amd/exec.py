@@ -520,9 +520,13 @@ class GpuHashAggregateExec(SparkPlan):
                 add(f"count({col})", ("count", t, v))
             elif fn == "count":
                 add(f"count({col})", ("count", t, v))
-            else:  # sum/min/max + count companion for NULL-ness
+            else:
                 add(f"{fn}({col})", (fn, t, v))
-                add(f"count({col})", ("count", t, v))
+                # count companion tracks NULL-ness (result NULL iff no
+                # non-null input). In complete mode over a non-null column
+                # every group is non-empty, so the companion is redundant.
+                if v is not None or self.mode != "complete":
+                    add(f"count({col})", ("count", t, v))
         if not specs:
             # pure GROUP BY (no aggregate expressions — the SELECT DISTINCT
             # shape): the kernel wants >= 1 spec; count rows and discard
@@ -665,16 +669,18 @@ class GpuHashAggregateExec(SparkPlan):
                 if fn == "count":
                     cols[out] = accs[slot[f"count({col})"]]
                     continue
-                cnt = accs[slot[f"count({col})"]]
+                cnt_slot = slot.get(f"count({col})")
                 if fn == "avg":
                     # Average.evaluateExpression: sum / cast(count)
                     cols[out] = gpuq.project_binop(
                         accs[slot[f"avg_sum({col})"]], "/",
-                        b=gpuq.cast_i64_f64(cnt))
+                        b=gpuq.cast_i64_f64(accs[cnt_slot]))
                 else:
                     cols[out] = accs[slot[f"{fn}({col})"]]
-                # SQL NULL iff no non-null input (Sum/Min/Max/Average.scala)
-                validity[out] = gpuq.nonzero_to_bits(cnt)
+                # SQL NULL iff no non-null input (Sum/Min/Max/Average.scala);
+                # no companion => non-null complete-mode input, always valid
+                if cnt_slot is not None:
+                    validity[out] = gpuq.nonzero_to_bits(accs[cnt_slot])
         batch.close()
         return ColumnarBatch(cols, validity=validity or None)
 
