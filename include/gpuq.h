@@ -284,6 +284,19 @@ int gpuq_join_build_i64(void* stream, int64_t build_rows, gpuq_col build_key,
  * after setting *out_nmatches (call again with bigger buffers).
  * probe_workspace (gpuq_join_probe_workspace_bytes) enables the
  * hash-ordered probe stream; pass NULL to probe in input order. */
+/* join_type (ShuffledHashJoinExec.scala joinType dispatch): 0=Inner,
+ * 1=probe-side Outer (unmatched probe rows pair with build rid 0xFFFFFFFF
+ * => NULL build columns; gather with gpuq_gather_nullable), 2=LeftSemi
+ * (probe row once iff matched; out_b = 0xFFFFFFFF), 3=LeftAnti (iff
+ * unmatched; NULL probe keys emit — the non-null-aware anti). */
+int gpuq_join_probe_i64_typed(void* stream, int64_t probe_rows, gpuq_col probe_key,
+                              const void* workspace, int64_t capacity,
+                              int64_t build_rows, void* probe_workspace,
+                              int64_t probe_workspace_bytes, int32_t join_type,
+                              uint32_t* out_probe_rid, uint32_t* out_build_rid,
+                              int64_t out_capacity, int64_t* out_nmatches);
+
+/* inner-join compatibility wrapper (join_type = 0) */
 int gpuq_join_probe_i64(void* stream, int64_t probe_rows, gpuq_col probe_key,
                         const void* workspace, int64_t capacity, int64_t build_rows,
                         void* probe_workspace, int64_t probe_ws_bytes,
@@ -342,6 +355,11 @@ int gpuq_bits_to_u8(void* stream, int64_t nrows, const uint8_t* bits,
                     uint8_t* out);
 int gpuq_u8_to_bits(void* stream, int64_t nrows, const uint8_t* u8,
                     uint8_t* out_bits);
+
+/* out[i] = col[perm[i]], with perm[i] == 0xFFFFFFFF producing NULL (the
+ * outer-join build side); out_bits = src validity AND not-NIL. */
+int gpuq_gather_nullable(void* stream, int64_t nrows, gpuq_col col,
+                         const uint32_t* perm, void* out, uint8_t* out_bits);
 
 /* validity bitmap for key column `bit` of a composite-key result:
  * out bit i = (mask[i] >> bit) & 1 (see gpuq_hash_agg_keys out_kmask) */

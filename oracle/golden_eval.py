@@ -131,9 +131,14 @@ def evaluate(plan: dict, tables: Dict[str, dict]) -> Frame:
         keep = [i for i in range(f.n) if _cmp(c[i], plan["cmp"], plan["lit"])]
         return Frame({k: [v[i] for i in keep] for k, v in f.cols.items()})
     if op == "join":
+        # kinds (joins/*.scala JoinType dispatch): inner | left | right |
+        # semi | anti. NULL keys never match; left/right preserve the
+        # named side with NULLs on the other; semi/anti emit left rows
+        # once by match existence (anti includes NULL-key left rows).
         lf = evaluate(plan["left"], tables)
         rf = evaluate(plan["right"], tables)
         lk, rk = lf.col(plan["lkey"]), rf.col(plan["rkey"])
+        kind = plan.get("kind", "inner")
         from collections import defaultdict
         buckets = defaultdict(list)
         for j, v in enumerate(rk):
@@ -141,18 +146,36 @@ def evaluate(plan: dict, tables: Dict[str, dict]) -> Frame:
                 buckets[v].append(j)
         li, ri = [], []
         for i, v in enumerate(lk):
-            if v is NULL:
+            hits = buckets.get(v, ()) if v is not NULL else ()
+            if kind == "semi":
+                if hits:
+                    li.append(i); ri.append(None)
                 continue
-            for j in buckets.get(v, ()):
-                li.append(i); ri.append(j)
+            if kind == "anti":
+                if not hits:
+                    li.append(i); ri.append(None)
+                continue
+            if hits:
+                for j in hits:
+                    li.append(i); ri.append(j)
+            elif kind in ("left",):
+                li.append(i); ri.append(None)
+        if kind == "right":
+            matched_r = set(j for j in ri if j is not None)
+            for j in range(rf.n):
+                if j not in matched_r and rk[j] is not NULL:
+                    li.append(None); ri.append(j)
+                elif rk[j] is NULL:
+                    li.append(None); ri.append(j)
         cols = {}
         for k, v in lf.cols.items():
-            cols[k] = [v[i] for i in li]
-        for k, v in rf.cols.items():
-            # NATURAL/USING joins keep both key columns here; the extractor
-            # disambiguates references (bare key -> left key) and expands *
-            # with the merged key shown once, matching Spark's output schema
-            cols[k] = [v[j] for j in ri]
+            cols[k] = [v[i] if i is not None else NULL for i in li]
+        if kind not in ("semi", "anti"):
+            for k, v in rf.cols.items():
+                # NATURAL/USING joins keep both key columns; the extractor
+                # disambiguates references (bare key -> preserved side) and
+                # expands * with the merged key once, matching Spark
+                cols[k] = [v[j] if j is not None else NULL for j in ri]
         return Frame(cols)
     if op == "agg":
         f = evaluate(plan["child"], tables)

@@ -2234,6 +2234,21 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
   }
 }
 
+/* join types over the PROBE (streamed) side — the reference's
+ * ShuffledHashJoinExec joinType dispatch (ShuffledHashJoinExec.scala /
+ * HashJoin.scala join() : inner, outer, semi, anti):
+ * 0 = Inner, 1 = probe-side Outer (unmatched probe rows emit one pair
+ * with build rid = JOIN_NIL -> NULL build columns), 2 = LeftSemi (probe
+ * row emitted once iff matched), 3 = LeftAnti (emitted iff unmatched;
+ * NULL probe keys never match, so they emit — the non-null-aware anti). */
+DEV uint32_t jt_emit_count(int jt, uint32_t m) {
+  if (jt == 1) return m ? m : 1u;
+  if (jt == 2) return m ? 1u : 0u;
+  if (jt == 3) return m ? 0u : 1u;
+  return m;
+}
+
+template <int JT>
 __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kvalid,
                              const unsigned long long* slots, const unsigned int* next,
                              const unsigned int* brid_map, const unsigned int* prid_map,
@@ -2284,7 +2299,8 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
         }
       }
     }
-    lane_total += cnt[r];
+    if (JT == 0) lane_total += cnt[r];
+    else if (i < n) lane_total += jt_emit_count(JT, cnt[r]);
   }
   /* phase B: block-level reservation (emit order within the block is
    * arbitrary — join output order is nondeterministic by contract) */
@@ -2301,8 +2317,19 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
   __syncthreads();
   int64_t o = (int64_t)block_base + wtot[wave] + (incl - lane_total);
   for (int r = 0; r < ROUNDS; r++) {
-    if (!cnt[r]) continue;
     int64_t i = base + r * 256 + threadIdx.x;
+    if (JT != 0) {
+      if (i >= n || !jt_emit_count(JT, cnt[r])) continue;
+      uint32_t pr = prid_map ? prid_map[i] : (uint32_t)i;
+      if (JT == 2 || JT == 3 || cnt[r] == 0) {
+        /* semi/anti emit the probe row once; outer's unmatched row pairs
+         * with NIL (NULL build columns downstream) */
+        if (o < out_cap) { out_p[o] = pr; out_b[o] = JOIN_NIL; }
+        o++;
+        continue;
+      }
+    }
+    if (!cnt[r]) continue;
     uint32_t pr = prid_map ? prid_map[i] : (uint32_t)i;
     if (o < out_cap) { out_p[o] = pr; out_b[o] = brid_map ? brid_map[c0[r]] : c0[r]; }
     o++;
@@ -2356,12 +2383,15 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   return GPUQ_OK;
 }
 
-extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
+extern "C" int gpuq_join_probe_i64_typed(void* stream, int64_t prows, gpuq_col pkey,
                                    const void* workspace, int64_t cap, int64_t brows,
                                    void* probe_workspace, int64_t probe_ws_bytes,
+                                   int32_t join_type,
                                    uint32_t* out_p, uint32_t* out_b,
                                    int64_t out_cap, int64_t* out_nmatches) {
   hipStream_t s = (hipStream_t)stream;
+  if (join_type < 0 || join_type > 3)
+    FAIL(GPUQ_ERR_INVALID, "join: bad join_type %d", join_type);
   if (pkey.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "join: key must be int64");
   join_ws w; int64_t need;
   join_ws_layout(cap, brows, &w, (char*)workspace, &need);
@@ -2393,10 +2423,20 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
     }
     int64_t nchunks = (prows + JOIN_CHUNK - 1) / JOIN_CHUNK;
     { hipEvent_t _pe = prof_begin(s);
-    k_join_probe<<<dim3((uint32_t)nchunks), 256, 0, s>>>(
-        prows, pkeys, pvalid, w.slots, w.next,
-        build_bucketed ? w.brid : nullptr, prid_map, w.sp, cap - 1,
-        out_p, out_b, out_cap);
+    const unsigned int* bm = build_bucketed ? w.brid : nullptr;
+    dim3 jg((uint32_t)nchunks);
+    if (join_type == 1)
+      k_join_probe<1><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
+                                         bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);
+    else if (join_type == 2)
+      k_join_probe<2><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
+                                         bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);
+    else if (join_type == 3)
+      k_join_probe<3><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
+                                         bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);
+    else
+      k_join_probe<0><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
+                                         bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);
     prof_end("join_probe", s, _pe); }
     HIP_TRY(hipGetLastError());
   }
@@ -2408,6 +2448,16 @@ extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
     FAIL(GPUQ_ERR_OVERFLOW, "join: %lld matches exceed out_cap %lld",
          (long long)hsp.cursor, (long long)out_cap);
   return GPUQ_OK;
+}
+
+extern "C" int gpuq_join_probe_i64(void* stream, int64_t prows, gpuq_col pkey,
+                                   const void* workspace, int64_t cap, int64_t brows,
+                                   void* probe_workspace, int64_t probe_ws_bytes,
+                                   uint32_t* out_p, uint32_t* out_b,
+                                   int64_t out_cap, int64_t* out_nmatches) {
+  return gpuq_join_probe_i64_typed(stream, prows, pkey, workspace, cap, brows,
+                                   probe_workspace, probe_ws_bytes, 0,
+                                   out_p, out_b, out_cap, out_nmatches);
 }
 
 /* ---- partitioned aggregation (mid/high cardinality) ----
@@ -3395,6 +3445,46 @@ extern "C" int gpuq_partition_perm_multi(void* stream, int64_t n,
   prof_end("partition_pids", s, _pe); }
   HIP_TRY(hipGetLastError());
   return partition_scatter_tail(s, n, w, nparts, out_perm);
+}
+
+/* out[i] = col[perm[i]] with perm[i] == 0xFFFFFFFF (JOIN_NIL) producing a
+ * NULL output row (outer-join build side); out_bits = resulting validity
+ * (src validity AND not-NIL). */
+__global__ void k_gather_nullable(int64_t n, const uint64_t* col,
+                                  const uint8_t* src_valid,
+                                  const uint32_t* perm, uint64_t* out,
+                                  uint8_t* out_bits) {
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nbytes = (n + 7) >> 3;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  for (; b < nbytes; b += gs) {
+    uint8_t v = 0;
+    int64_t row0 = b << 3;
+    int top = (int)((n - row0) < 8 ? (n - row0) : 8);
+    for (int t = 0; t < top; t++) {
+      uint32_t p = perm[row0 + t];
+      bool ok = p != 0xFFFFFFFFu;
+      out[row0 + t] = ok ? col[p] : 0;
+      if (ok && (!src_valid || ((src_valid[p >> 3] >> (p & 7)) & 1)))
+        v |= (uint8_t)(1 << t);
+    }
+    out_bits[b] = v;
+  }
+}
+
+extern "C" int gpuq_gather_nullable(void* stream, int64_t nrows, gpuq_col col,
+                                    const uint32_t* perm, void* out,
+                                    uint8_t* out_bits) {
+  hipStream_t s = (hipStream_t)stream;
+  if (col.dtype != GPUQ_INT64 && col.dtype != GPUQ_FLOAT64)
+    FAIL(GPUQ_ERR_INVALID, "gather_nullable: unsupported dtype %d", col.dtype);
+  if (nrows == 0) return GPUQ_OK;
+  int64_t nbytes = (nrows + 7) >> 3;
+  k_gather_nullable<<<grid1d(nbytes), 256, 0, s>>>(
+      nrows, (const uint64_t*)col.data, col.validity, perm, (uint64_t*)out,
+      out_bits);
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
 }
 
 /* ================= validity-bitmap utilities ================= */

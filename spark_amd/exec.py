@@ -156,13 +156,21 @@ class HashAggregateExec(_CpuNode):
 
 
 class ShuffledHashJoinExec(_CpuNode):
+    """join_type: "inner" | "left_outer" | "right_outer" | "left_semi" |
+    "left_anti" (ShuffledHashJoinExec.scala joinType; the preserved side
+    must stream, so left-preserving types build on the right and
+    vice versa)."""
+
     def __init__(self, left_key: str, right_key: str, build_side: str,
-                 left, right):
+                 left, right, join_type: str = "inner"):
         super().__init__(left, right)
         self.left_key, self.right_key, self.build_side = left_key, right_key, build_side
+        self.join_type = join_type
 
     @property
     def output(self):
+        if self.join_type in ("left_semi", "left_anti"):
+            return self.children[0].output
         return self.children[0].output + self.children[1].output
 
 
@@ -173,9 +181,11 @@ class SortMergeJoinExec(_CpuNode):
     57-69), and on GPU the hash build/probe dominates sort+merge for
     co-partitioned batches."""
 
-    def __init__(self, left_key: str, right_key: str, left, right):
+    def __init__(self, left_key: str, right_key: str, left, right,
+                 join_type: str = "inner"):
         super().__init__(left, right)
         self.left_key, self.right_key = left_key, right_key
+        self.join_type = join_type
 
     @property
     def output(self):
@@ -798,13 +808,24 @@ class GpuShuffledHashJoinExec(SparkPlan):
     exchanges, as EnsureRequirements would arrange."""
 
     def __init__(self, left_key: str, right_key: str, build_side: str,
-                 left, right):
+                 left, right, join_type: str = "inner"):
         super().__init__(left, right)
         assert build_side in ("left", "right")
+        assert join_type in ("inner", "left_outer", "right_outer",
+                             "left_semi", "left_anti")
+        # the preserved/streamed side must be the PROBE side
+        # (HashJoin.scala buildSide constraints)
+        if join_type in ("left_outer", "left_semi", "left_anti"):
+            assert build_side == "right", f"{join_type} builds on the right"
+        if join_type == "right_outer":
+            assert build_side == "left", "right_outer builds on the left"
         self.left_key, self.right_key, self.build_side = left_key, right_key, build_side
+        self.join_type = join_type
 
     @property
     def output(self):
+        if self.join_type in ("left_semi", "left_anti"):
+            return self.children[0].output
         return self.children[0].output + self.children[1].output
 
     @property
@@ -824,6 +845,9 @@ class GpuShuffledHashJoinExec(SparkPlan):
         build, probe = (lb, rb) if self.build_side == "left" else (rb, lb)
         bkey = self.left_key if self.build_side == "left" else self.right_key
         pkey = self.right_key if self.build_side == "left" else self.left_key
+        jt = {"inner": gpuq.JOIN_INNER, "left_outer": gpuq.JOIN_OUTER,
+              "right_outer": gpuq.JOIN_OUTER, "left_semi": gpuq.JOIN_SEMI,
+              "left_anti": gpuq.JOIN_ANTI}[self.join_type]
         bk = build.column(bkey)
         bn = bk.numel()
         cap = 1 << max(4, int(bn * 2 - 1).bit_length() if bn else 4)
@@ -832,16 +856,25 @@ class GpuShuffledHashJoinExec(SparkPlan):
         out_cap = max(int(probe.num_rows() * 2) + 64, 64)
         while True:
             op, ob, nm = gpuq.join_probe(pk, ws, cap, bn, out_cap,
-                                         key_validity=probe.validity(pkey))
+                                         key_validity=probe.validity(pkey),
+                                         join_type=jt)
             if op is not None:
                 break
             out_cap = nm + 64
         cols, validity = {}, {}
-        for name, t in build.columns().items():
-            cols[name] = gpuq.gather(t, ob)
-            v = build.validity(name)
-            if v is not None:
-                validity[name] = gpuq.gather_bits(v, ob)
+        semi = self.join_type in ("left_semi", "left_anti")
+        outer = self.join_type in ("left_outer", "right_outer")
+        if not semi:
+            for name, t in build.columns().items():
+                if outer:
+                    # unmatched probe rows pair with NIL -> NULL build cols
+                    cols[name], validity[name] = gpuq.gather_nullable(
+                        t, ob, validity=build.validity(name))
+                else:
+                    cols[name] = gpuq.gather(t, ob)
+                    v = build.validity(name)
+                    if v is not None:
+                        validity[name] = gpuq.gather_bits(v, ob)
         for name, t in probe.columns().items():
             v = probe.validity(name)
             if name in cols:
@@ -1013,7 +1046,8 @@ class GpuColumnarRule:
                                         *children, capacity=plan.capacity)
         if isinstance(plan, ShuffledHashJoinExec):
             return GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
-                                           plan.build_side, *children)
+                                           plan.build_side, *children,
+                                           join_type=plan.join_type)
         if isinstance(plan, SortMergeJoinExec):
             # SMJ -> GPU hash join (same slot, same required distribution;
             # build on the right side as SHJ's default would choose). SMJ
@@ -1021,8 +1055,10 @@ class GpuColumnarRule:
             # have had sorts elided against it — so re-sort the hash join's
             # output to honor the contract (one radix pass; the GPU sort
             # node's output_ordering then matches what SMJ declared).
+            build = "left" if plan.join_type == "right_outer" else "right"
             j = GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
-                                        "right", *children)
+                                        build, *children,
+                                        join_type=plan.join_type)
             if self.preserve_smj_ordering:
                 return GpuSortExec(SortOrder(plan.left_key), False, j)
             return j

@@ -124,6 +124,12 @@ def lib() -> ctypes.CDLL:
         L.gpuq_join_probe_i64.restype = i32
         L.gpuq_join_probe_i64.argtypes = [vp, i64, _Col, vp, i64, i64, vp, i64,
                                           vp, vp, i64, ctypes.POINTER(i64)]
+        L.gpuq_join_probe_i64_typed.restype = i32
+        L.gpuq_join_probe_i64_typed.argtypes = [vp, i64, _Col, vp, i64, i64, vp,
+                                                i64, i32, vp, vp, i64,
+                                                ctypes.POINTER(i64)]
+        L.gpuq_gather_nullable.restype = i32
+        L.gpuq_gather_nullable.argtypes = [vp, i64, _Col, vp, vp, vp]
         L.gpuq_filter_workspace_bytes.restype = i64
         L.gpuq_filter_workspace_bytes.argtypes = [i64]
         L.gpuq_filter_cmp.restype = i32
@@ -310,17 +316,23 @@ def join_probe_workspace(probe_rows: int, device="cuda") -> torch.Tensor:
                        dtype=torch.uint8, device=device)
 
 
+JOIN_INNER, JOIN_OUTER, JOIN_SEMI, JOIN_ANTI = 0, 1, 2, 3
+JOIN_NIL = 0xFFFFFFFF
+
+
 def join_probe(probe_keys: torch.Tensor, workspace: torch.Tensor, capacity: int,
-               build_rows: int, out_cap: int, key_validity=None, probe_ws=None):
+               build_rows: int, out_cap: int, key_validity=None, probe_ws=None,
+               join_type: int = JOIN_INNER):
     pn = probe_keys.numel()
     dev = probe_keys.device
     op = torch.empty(out_cap, dtype=torch.int32, device=dev)
     ob = torch.empty(out_cap, dtype=torch.int32, device=dev)
     nm = ctypes.c_int64(0)
-    rc = lib().gpuq_join_probe_i64(_stream(), pn, _col(probe_keys, key_validity),
+    rc = lib().gpuq_join_probe_i64_typed(_stream(), pn, _col(probe_keys, key_validity),
                                    workspace.data_ptr(), capacity, build_rows,
                                    _dp(probe_ws),
                                    probe_ws.numel() if probe_ws is not None else 0,
+                                   join_type,
                                    op.data_ptr(), ob.data_ptr(), out_cap,
                                    ctypes.byref(nm))
     if rc == 3:  # GPUQ_ERR_OVERFLOW: caller retries with nm.value capacity
@@ -331,6 +343,18 @@ def join_probe(probe_keys: torch.Tensor, workspace: torch.Tensor, capacity: int,
 
 CMP = {"==": 0, "<": 1, "<=": 2, ">": 3, ">=": 4, "!=": 5}
 BINOP = {"+": 0, "-": 1, "*": 2, "/": 3, "rsub": 4}
+
+
+def gather_nullable(col: torch.Tensor, perm: torch.Tensor, validity=None):
+    """gather with NIL (0xFFFFFFFF) permutation entries producing NULL
+    rows (outer-join build side). Returns (values, validity bitmap)."""
+    n = perm.numel()
+    out = torch.empty(n, dtype=col.dtype, device=col.device)
+    bits = torch.empty(_bitmap_bytes(n), dtype=torch.uint8, device=col.device)
+    _check(lib().gpuq_gather_nullable(_stream(), n, _col(col, validity),
+                                      perm.data_ptr(),
+                                      out.data_ptr(), bits.data_ptr()))
+    return out, bits
 
 
 def filter_cmp(col: torch.Tensor, op: str, literal, workspace=None, validity=None):

@@ -90,7 +90,12 @@ def compile_plan(plan, case, dct, agg_mode):
         right = compile_plan(plan["right"], case, dct, agg_mode)
         lk = resolve(left.output, plan["lkey"])
         rk = resolve(right.output, plan["rkey"])
-        return gx.ShuffledHashJoinExec(lk, rk, "right", left, right)
+        kind = plan.get("kind", "inner")
+        jt = {"inner": "inner", "left": "left_outer", "right": "right_outer",
+              "semi": "left_semi", "anti": "left_anti"}[kind]
+        build = "left" if kind == "right" else "right"
+        return gx.ShuffledHashJoinExec(lk, rk, build, left, right,
+                                       join_type=jt)
     if op == "agg":
         child = compile_plan(plan["child"], case, dct, agg_mode)
         keys = tuple(resolve(child.output, k) for k in plan["keys"])
