@@ -838,11 +838,11 @@ class GpuShuffledHashJoinExec(SparkPlan):
 
     def execute_columnar(self):
         from . import gpuq
-        left_b = list(self.children[0].execute_columnar())
-        right_b = list(self.children[1].execute_columnar())
-        assert len(left_b) == 1 and len(right_b) == 1, "one batch per partition"
-        lb, rb = left_b[0], right_b[0]
-        build, probe = (lb, rb) if self.build_side == "left" else (rb, lb)
+        bi = 0 if self.build_side == "left" else 1
+        build_b = list(self.children[bi].execute_columnar())
+        assert len(build_b) == 1, "one build batch per partition " \
+            "(buildHashedRelation materializes the build side)"
+        build = build_b[0]
         bkey = self.left_key if self.build_side == "left" else self.right_key
         pkey = self.right_key if self.build_side == "left" else self.left_key
         jt = {"inner": gpuq.JOIN_INNER, "left_outer": gpuq.JOIN_OUTER,
@@ -852,38 +852,43 @@ class GpuShuffledHashJoinExec(SparkPlan):
         bn = bk.numel()
         cap = 1 << max(4, int(bn * 2 - 1).bit_length() if bn else 4)
         ws = gpuq.join_build(bk, cap, key_validity=build.validity(bkey))
-        pk = probe.column(pkey)
-        out_cap = max(int(probe.num_rows() * 2) + 64, 64)
-        while True:
-            op, ob, nm = gpuq.join_probe(pk, ws, cap, bn, out_cap,
-                                         key_validity=probe.validity(pkey),
-                                         join_type=jt)
-            if op is not None:
-                break
-            out_cap = nm + 64
-        cols, validity = {}, {}
         semi = self.join_type in ("left_semi", "left_anti")
         outer = self.join_type in ("left_outer", "right_outer")
-        if not semi:
-            for name, t in build.columns().items():
-                if outer:
-                    # unmatched probe rows pair with NIL -> NULL build cols
-                    cols[name], validity[name] = gpuq.gather_nullable(
-                        t, ob, validity=build.validity(name))
-                else:
-                    cols[name] = gpuq.gather(t, ob)
-                    v = build.validity(name)
-                    if v is not None:
-                        validity[name] = gpuq.gather_bits(v, ob)
-        for name, t in probe.columns().items():
-            v = probe.validity(name)
-            if name in cols:
-                name = f"{name}#probe"
-            cols[name] = gpuq.gather(t, op)
-            if v is not None:
-                validity[name] = gpuq.gather_bits(v, op)
-        lb.close(), rb.close()
-        yield ColumnarBatch(cols, validity=validity or None)
+        # the probe (streamed) side is consumed batch-at-a-time
+        # (ShuffledHashJoinExec.doExecute streams streamedIter)
+        for probe in self.children[1 - bi].execute_columnar():
+            pk = probe.column(pkey)
+            out_cap = max(int(probe.num_rows() * 2) + 64, 64)
+            while True:
+                op, ob, nm = gpuq.join_probe(pk, ws, cap, bn, out_cap,
+                                             key_validity=probe.validity(pkey),
+                                             join_type=jt)
+                if op is not None:
+                    break
+                out_cap = nm + 64
+            cols, validity = {}, {}
+            if not semi:
+                for name, t in build.columns().items():
+                    if outer:
+                        # unmatched probe rows pair with NIL -> NULL build
+                        cols[name], validity[name] = gpuq.gather_nullable(
+                            t, ob, validity=build.validity(name))
+                    else:
+                        cols[name] = gpuq.gather(t, ob)
+                        v = build.validity(name)
+                        if v is not None:
+                            validity[name] = gpuq.gather_bits(v, ob)
+            pcols = probe.columns()
+            for name, t in pcols.items():
+                v = probe.validity(name)
+                if name in cols:
+                    name = f"{name}#probe"
+                cols[name] = gpuq.gather(t, op)
+                if v is not None:
+                    validity[name] = gpuq.gather_bits(v, op)
+            probe.close()
+            yield ColumnarBatch(cols, validity=validity or None)
+        build.close()
 
 
 class GpuRangeExec(SparkPlan):

@@ -228,3 +228,24 @@ def test_aqe_get_shuffle_partitions_coalesce():
         assert (got_p[got_valid] == pay[sel][got_valid]).all()
     # total coverage: the three ranges partition the full map output
     assert sum(b.num_rows() for b in batches) == n
+
+
+def test_join_streams_probe_batches():
+    """the probe side is consumed batch-at-a-time (ShuffledHashJoinExec
+    .doExecute streams streamedIter; the build side materializes once)."""
+    bkeys = oracle.gen_i64(seed=500, n=5_000, range_=4_000)
+    p1 = oracle.gen_i64(seed=501, n=8_000, range_=4_000)
+    p2 = oracle.gen_i64(seed=502, n=6_000, range_=4_000)
+    left = gx.InputBatches([dev_batch(lk=p1), dev_batch(lk=p2)])
+    right = gx.InputBatches([dev_batch(rk=bkeys)])
+    node = gx.ShuffledHashJoinExec("lk", "rk", "right", left, right)
+    outs = list(gx.GpuColumnarRule().pre_columnar_transitions(node)
+                .execute_columnar())
+    assert len(outs) == 2
+    for probe_keys, out in zip((p1, p2), outs):
+        op, ob = oracle.join_inner(bkeys, probe_keys)
+        assert out.num_rows() == len(op)
+        g = np.lexsort((out.column("rk").cpu().numpy(),
+                        out.column("lk").cpu().numpy()))
+        o = np.lexsort((bkeys[ob], probe_keys[op]))
+        assert (out.column("lk").cpu().numpy()[g] == probe_keys[op][o]).all()
