@@ -1,0 +1,23 @@
+"""Small probe for PMC counter runs (counter collection serializes kernel
+launches — keep the workload tiny)."""
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
+from spark_amd import gpuq as gq
+
+n = 100_000_000
+keys = gq.gen_i64(seed=42, n=n)
+pay1 = gq.gen_i64(seed=43, n=n)
+pay2 = gq.gen_f64_unit(seed=44, n=n)
+ws = gq.sort_workspace(n)
+perm, _ = gq.sort_perm(keys, workspace=ws, out_keys=False)
+out1 = torch.empty(n, dtype=torch.int64, device="cuda")
+out2 = torch.empty(n, dtype=torch.float64, device="cuda")
+pairs = torch.empty(n * 2, dtype=torch.int64, device="cuda")
+gq._check(gq.lib().gpuq_gather2_i64_fast(gq._stream(), n, pay1.data_ptr(),
+          pay2.data_ptr(), perm.data_ptr(), out1.data_ptr(), out2.data_ptr(),
+          pairs.data_ptr()))
+gq._check(gq.lib().gpuq_gather2_i64(gq._stream(), n, pay1.data_ptr(),
+          pay2.data_ptr(), perm.data_ptr(), out1.data_ptr(), out2.data_ptr()))
+torch.cuda.synchronize()
+print("pmc probe done")
