@@ -393,3 +393,21 @@ def test_typed_join_parity(gq, jt_name):
                for i in range(len(got_lk))]
     got = sorted(got, key=lambda r: tuple((x is None, x or 0) for x in r))
     assert got == expected
+
+
+def test_filter_fractional_literal_on_int64(gq):
+    """Spark compares int columns against fractional literals via
+    cast-to-double; the engine rewrites to equivalent integer predicates
+    (k < 0.5 must keep k == 0 — the truncation bug class)."""
+    n = 100_000
+    vals = oracle.gen_i64(seed=900, n=n, range_=7) - 3   # -3..3
+    dv = to_dev(vals)
+    for op, lit in [("<", 0.5), ("<=", 0.5), (">", 0.5), (">=", 0.5),
+                    ("<", -1.5), (">", -1.5), ("==", 0.5), ("!=", 0.5),
+                    ("<", 1e300), (">", 1e300)]:
+        perm, cnt = gq.filter_cmp(dv, op, lit)
+        ref = {"<": vals < lit, "<=": vals <= lit, ">": vals > lit,
+               ">=": vals >= lit, "==": vals == lit, "!=": vals != lit}[op]
+        exp = np.flatnonzero(ref)
+        assert cnt == len(exp), (op, lit, cnt, len(exp))
+        assert (perm.cpu().numpy().astype(np.int64) == exp).all(), (op, lit)
