@@ -367,11 +367,18 @@ def filter_cmp(col: torch.Tensor, op: str, literal, workspace=None, validity=Non
     (k < 0.5 must keep k == 0)."""
     n = col.numel()
     dev = col.device
+    import math
+    I64_MIN, I64_MAX = -(1 << 63), (1 << 63) - 1
     if col.dtype == torch.int64 and isinstance(literal, float) \
-            and literal != int(literal):
-        import math
+            and math.isinf(literal):
+        # +/-inf literal: constant predicate
+        all_ops = ("<", "<=") if literal > 0 else (">", ">=")
+        literal, op = I64_MIN, (">=" if op in all_ops or op == "!=" else "<")
+    if col.dtype == torch.int64 and isinstance(literal, float) \
+            and math.isfinite(literal) \
+            and (literal != int(literal)
+                 or not (I64_MIN <= int(literal) <= I64_MAX)):
         f = math.floor(literal)
-        I64_MIN, I64_MAX = -(1 << 63), (1 << 63) - 1
         if op in ("==",):
             op, literal = "<", I64_MIN          # never true
         elif op in ("!=",):
