@@ -307,6 +307,32 @@ __global__ void k_gather2_pairs(int64_t n, const ulonglong2* pairs,
   }
 }
 
+/* 4-way batched variant: each thread keeps 4 independent b128 loads in
+ * flight (A/B vs the scalar grid-stride loop; selected by env). */
+__global__ void k_gather2_pairs4(int64_t n, const ulonglong2* pairs,
+                                 const uint32_t* perm, uint64_t* oa,
+                                 uint64_t* ob) {
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t base = t * 4; base < n; base += stride) {
+    uint32_t p[4];
+    ulonglong2 v[4];
+    int top = (int)((n - base) < 4 ? (n - base) : 4);
+    #pragma unroll
+    for (int q = 0; q < 4; q++)
+      if (q < top) p[q] = perm[base + q];
+    #pragma unroll
+    for (int q = 0; q < 4; q++)
+      if (q < top) v[q] = pairs[p[q]];
+    #pragma unroll
+    for (int q = 0; q < 4; q++)
+      if (q < top) {
+        oa[base + q] = v[q].x;
+        ob[base + q] = v[q].y;
+      }
+  }
+}
+
 /* scratch: n * 16 bytes, device. */
 extern "C" int gpuq_gather2_i64_fast(void* stream, int64_t n, const void* a,
                                      const void* b, const uint32_t* perm,
@@ -319,10 +345,16 @@ extern "C" int gpuq_gather2_i64_fast(void* stream, int64_t n, const void* a,
                                           (ulonglong2*)scratch);
   prof_end("interleave2", s, _pe); }
   HIP_TRY(hipGetLastError());
+  static int mlp = -1;
+  if (mlp < 0) mlp = getenv("GPUQ_GATHER_MLP") ? atoi(getenv("GPUQ_GATHER_MLP")) : 0;
   { hipEvent_t _pe = prof_begin(s);
-  k_gather2_pairs<<<grid1d(n), 256, 0, s>>>(n, (const ulonglong2*)scratch,
-                                            perm, (uint64_t*)oa,
-                                            (uint64_t*)ob);
+  if (mlp)
+    k_gather2_pairs4<<<grid1d((n + 3) / 4), 256, 0, s>>>(
+        n, (const ulonglong2*)scratch, perm, (uint64_t*)oa, (uint64_t*)ob);
+  else
+    k_gather2_pairs<<<grid1d(n), 256, 0, s>>>(n, (const ulonglong2*)scratch,
+                                              perm, (uint64_t*)oa,
+                                              (uint64_t*)ob);
   prof_end("gather2_pairs", s, _pe); }
   HIP_TRY(hipGetLastError());
   return GPUQ_OK;
