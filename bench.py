@@ -364,6 +364,95 @@ class Q1Workload:
         self.cols.clear()
 
 
+def build_q3_plan(gx, lineitem, orders, customer, cut_date: int,
+                  segment_id: int):
+    """The TPC-H Q3 plan shape (resources/tpch/q3.sql) over device batches:
+    customer filtered on mktsegment feeds a BROADCAST dimension join with
+    date-filtered orders (BroadcastHashJoinExec, the §8(f).3 slot); the
+    result is the build side of the fact join with shipdate-filtered
+    lineitem; revenue = extendedprice*(100-discount) as exact scale-4
+    int64; 3-key composite GROUP BY (l_orderkey, o_orderdate,
+    o_shippriority); ORDER BY revenue DESC, o_orderdate. LIMIT is the
+    host's slice of the sorted output (a CollectLimit concern, not the
+    engine's)."""
+    cust = gx.FilterExec("c_mktsegment", "==", segment_id,
+                         gx.InputBatches([customer]))
+    ord_f = gx.FilterExec("o_orderdate", "<", cut_date,
+                          gx.InputBatches([orders]))
+    # dimension join: broadcast the filtered customers, stream orders
+    j1 = gx.BroadcastHashJoinExec("o_custkey", "c_custkey", "right",
+                                  ord_f, gx.BroadcastExchangeExec(cust))
+    j1p = gx.ProjectExec(["o_orderkey", "o_orderdate", "o_shippriority"], j1)
+    li = gx.FilterExec("l_shipdate", ">", cut_date,
+                       gx.InputBatches([lineitem]))
+    j2 = gx.ShuffledHashJoinExec("l_orderkey", "o_orderkey", "right",
+                                 li, j1p)
+    rev = gx.ProjectExec(
+        ["l_orderkey", "o_orderdate", "o_shippriority",
+         ("__omd", "l_discount", "rsub", None, 100)],
+        j2)
+    rev2 = gx.ProjectExec(
+        ["l_orderkey", "o_orderdate", "o_shippriority",
+         ("revenue", "l_extendedprice", "*", "__omd", None)],
+        rev)
+    agg = gx.HashAggregateExec(
+        ("l_orderkey", "o_orderdate", "o_shippriority"),
+        [("sum", "revenue")], "complete", rev2)
+    return gx.SortExec([gx.SortOrder("sum(revenue)", descending=True),
+                        gx.SortOrder("o_orderdate")], False, agg)
+
+
+class Q3Workload:
+    """Config-5 Q3 probe (single-GPU slice): two joins (broadcast dimension
+    + shuffled fact), date filters, exact decimal revenue, 3-key composite
+    grouping, multi-key ORDER BY with DESC over the aggregate output."""
+
+    def __init__(self, gq, rows, rank):
+        from spark_amd import exec as gx
+        self.gx = gx
+        orders_n = max(rows // 4, 1024)
+        cust_n = max(rows // 20, 1024)
+        off = rank * 11
+        # persistent column tensors; fresh ColumnarBatch wrappers per step
+        # (consumers close their input batches — the batch lifetime
+        # contract)
+        self.li = dict(
+            l_orderkey=gq.gen_i64(seed=81 + off, n=rows, range_=orders_n),
+            l_extendedprice=gq.gen_i64(seed=82 + off, n=rows,
+                                       range_=10_000_000),
+            l_discount=gq.gen_i64(seed=83 + off, n=rows, range_=11),
+            l_shipdate=gq.gen_i64(seed=84 + off, n=rows, range_=2556),
+        )
+        self.od = dict(
+            o_orderkey=gq.range_i64(orders_n),
+            o_custkey=gq.gen_i64(seed=85 + off, n=orders_n, range_=cust_n),
+            o_orderdate=gq.gen_i64(seed=86 + off, n=orders_n, range_=2556),
+            o_shippriority=gq.gen_i64(seed=87 + off, n=orders_n, range_=2),
+        )
+        self.cu = dict(
+            c_custkey=gq.range_i64(cust_n),
+            c_mktsegment=gq.gen_i64(seed=88 + off, n=cust_n, range_=5),
+        )
+        self.rows = rows
+        self.ngroups = 0
+
+    def step(self):
+        gx = self.gx
+        plan = build_q3_plan(gx, gx.ColumnarBatch(dict(self.li)),
+                             gx.ColumnarBatch(dict(self.od)),
+                             gx.ColumnarBatch(dict(self.cu)),
+                             cut_date=1169,   # ~1995-03-15 in day offsets
+                             segment_id=1)
+        plan = gx.GpuColumnarRule().pre_columnar_transitions(plan)
+        out = next(plan.execute_columnar())
+        self.ngroups = out.num_rows()
+        # top-10 = the already-sorted head (CollectLimit host slice)
+        out.close()
+
+    def free(self):
+        self.li.clear(), self.od.clear(), self.cu.clear()
+
+
 def time_workload(w, steps, warmup, world):
     for _ in range(warmup):
         w.step()
@@ -416,7 +505,7 @@ def main():
     p.add_argument("--quick", action="store_true", help="small sizes (CI/sanity)")
     p.add_argument("--no-cpu-baseline", action="store_true")
     p.add_argument("--cpu-sample-rows", type=int, default=40_000_000)
-    p.add_argument("--workloads", default="sort,agg,join,q1,gsort")
+    p.add_argument("--workloads", default="sort,agg,join,q1,q3,gsort")
     args = p.parse_args()
     if args.quick:
         args.rows, args.agg_rows, args.agg_groups, args.join_rows = \
@@ -528,6 +617,23 @@ def main():
 
     if "q1" in wl:
         guarded("q1", _run_q1)
+
+    def _run_q3():
+            w = Q3Workload(gq, args.join_rows, rank)
+            sec = time_workload(w, args.steps, args.warmup, world)
+            ng = w.ngroups
+            w.free()
+            torch.cuda.empty_cache()
+            results["q3"] = {"sec_per_step": sec,
+                             "rows_per_sec": args.join_rows * world / sec,
+                             "plan": "Q3: cust-filter->broadcast dim join->"
+                                     "fact join->revenue->3-key groupby->"
+                                     "orderby(revenue desc, date)",
+                             "ngroups": ng}
+
+
+    if "q3" in wl:
+        guarded("q3", _run_q3)
     if rank != 0:
         return
 

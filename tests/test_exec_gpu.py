@@ -249,3 +249,58 @@ def test_join_streams_probe_batches():
                         out.column("lk").cpu().numpy()))
         o = np.lexsort((bkeys[ob], probe_keys[op]))
         assert (out.column("lk").cpu().numpy()[g] == probe_keys[op][o]).all()
+
+
+def test_q3_plan_parity():
+    """the TPC-H Q3 plan shape (bench.build_q3_plan): broadcast dimension
+    join + shuffled fact join + exact scaled-decimal revenue + 3-key
+    composite GROUP BY + ORDER BY revenue DESC, date — vs a numpy
+    reference."""
+    import bench
+    rng = np.random.default_rng(17)
+    nl, no, nc = 200_000, 50_000, 10_000
+    li = dict(l_orderkey=rng.integers(0, no, nl),
+              l_extendedprice=rng.integers(1, 10_000_00, nl),
+              l_discount=rng.integers(0, 11, nl),
+              l_shipdate=rng.integers(0, 2556, nl))
+    od = dict(o_orderkey=np.arange(no, dtype=np.int64),
+              o_custkey=rng.integers(0, nc, no),
+              o_orderdate=rng.integers(0, 2556, no),
+              o_shippriority=rng.integers(0, 2, no))
+    cu = dict(c_custkey=np.arange(nc, dtype=np.int64),
+              c_mktsegment=rng.integers(0, 5, nc))
+    cut, seg = 1169, 1
+    plan = bench.build_q3_plan(
+        gx, dev_batch(**{k: v.astype(np.int64) for k, v in li.items()}),
+        dev_batch(**{k: v.astype(np.int64) for k, v in od.items()}),
+        dev_batch(**{k: v.astype(np.int64) for k, v in cu.items()}),
+        cut_date=cut, segment_id=seg)
+    out = list(gx.GpuColumnarRule().pre_columnar_transitions(plan)
+               .execute_columnar())[0]
+
+    # numpy reference
+    seg_cust = set(np.flatnonzero(cu["c_mktsegment"] == seg).tolist())
+    ok_orders = {}
+    for j in range(no):
+        if od["o_orderdate"][j] < cut and int(od["o_custkey"][j]) in seg_cust:
+            ok_orders[j] = (int(od["o_orderdate"][j]),
+                            int(od["o_shippriority"][j]))
+    groups = {}
+    for i in range(nl):
+        okey = int(li["l_orderkey"][i])
+        if li["l_shipdate"][i] > cut and okey in ok_orders:
+            d, p = ok_orders[okey]
+            rev = int(li["l_extendedprice"][i]) * (100 - int(li["l_discount"][i]))
+            t = (okey, d, p)
+            groups[t] = groups.get(t, 0) + rev
+    gk = out.column("l_orderkey").cpu().numpy()
+    gd = out.column("o_orderdate").cpu().numpy()
+    gp = out.column("o_shippriority").cpu().numpy()
+    gr = out.column("sum(revenue)").cpu().numpy()
+    assert len(gk) == len(groups)
+    for i in range(len(gk)):
+        assert groups[(int(gk[i]), int(gd[i]), int(gp[i]))] == int(gr[i])
+    # ordering: revenue DESC, then o_orderdate ASC
+    for i in range(1, len(gk)):
+        assert (gr[i - 1] > gr[i]
+                or (gr[i - 1] == gr[i] and gd[i - 1] <= gd[i]))
