@@ -388,7 +388,7 @@ def build_q3_plan(gx, lineitem, orders, customer, cut_date: int,
     j2 = gx.ShuffledHashJoinExec("l_orderkey", "o_orderkey", "right",
                                  li, j1p)
     rev = gx.ProjectExec(
-        ["l_orderkey", "o_orderdate", "o_shippriority",
+        ["l_orderkey", "o_orderdate", "o_shippriority", "l_extendedprice",
          ("__omd", "l_discount", "rsub", None, 100)],
         j2)
     rev2 = gx.ProjectExec(
