@@ -158,9 +158,9 @@ def evaluate(plan: dict, tables: Dict[str, dict]) -> Frame:
             if hits:
                 for j in hits:
                     li.append(i); ri.append(j)
-            elif kind in ("left",):
+            elif kind in ("left", "full"):
                 li.append(i); ri.append(None)
-        if kind == "right":
+        if kind in ("right", "full"):
             matched_r = set(j for j in ri if j is not None)
             for j in range(rf.n):
                 if j not in matched_r and rk[j] is not NULL:

@@ -2112,6 +2112,7 @@ DEV uint64_t join_slot(int64_t key, int64_t cap_mask) {
 struct join_ws {
   unsigned long long* slots;   /* 2 u64 per slot: [key][head|pad] */
   unsigned int* next;
+  unsigned long long* matched; /* build-row matched bits (full outer) */
   unsigned int* brid;          /* partitioned row id map (bucketed path) */
   uint64_t* pk_a; uint32_t* pi_a;   /* partition scratch: pairs in */
   uint64_t* pk_b;                   /* partitioned keys out (ids go to brid) */
@@ -2132,6 +2133,7 @@ static void join_ws_layout(int64_t cap, int64_t brows, join_ws* w, char* base, i
   };
   w->slots = (unsigned long long*)take(cap * 16);
   w->next = (unsigned int*)take(brows * 4);
+  w->matched = (unsigned long long*)take(((brows + 63) / 64) * 8);
   w->brid = (unsigned int*)take(brows * 4);
   w->pk_a = (uint64_t*)take(brows * 8);
   w->pi_a = (uint32_t*)take(brows * 4);
@@ -2274,7 +2276,7 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
  * row emitted once iff matched), 3 = LeftAnti (emitted iff unmatched;
  * NULL probe keys never match, so they emit — the non-null-aware anti). */
 DEV uint32_t jt_emit_count(int jt, uint32_t m) {
-  if (jt == 1) return m ? m : 1u;
+  if (jt == 1 || jt == 4) return m ? m : 1u;  /* 4 = FullOuter probe half */
   if (jt == 2) return m ? 1u : 0u;
   if (jt == 3) return m ? 0u : 1u;
   return m;
@@ -2285,7 +2287,8 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
                              const unsigned long long* slots, const unsigned int* next,
                              const unsigned int* brid_map, const unsigned int* prid_map,
                              join_sp* sp, int64_t cap_mask,
-                             uint32_t* out_p, uint32_t* out_b, int64_t out_cap) {
+                             uint32_t* out_p, uint32_t* out_b, int64_t out_cap,
+                             unsigned long long* matched = nullptr) {
   constexpr int ROUNDS = JOIN_CHUNK / 256;
   __shared__ unsigned long long block_base;
   __shared__ uint32_t wtot[JOIN_CHUNK / 256 > 4 ? 16 : 16];
@@ -2324,10 +2327,12 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
     if (head != JOIN_NIL) {
       if (!(head & JOIN_MULTI)) {
         c0[r] = head; cnt[r] = 1;
+        if (JT == 4) atomicOr(&matched[head >> 6], 1ull << (head & 63));
       } else {
         for (unsigned int b = head & ~JOIN_MULTI; b != JOIN_NIL; b = next[b]) {
           if (cnt[r] == 0) c0[r] = b; else if (cnt[r] == 1) c1[r] = b;
           cnt[r]++;
+          if (JT == 4) atomicOr(&matched[b >> 6], 1ull << (b & 63));
         }
       }
     }
@@ -2376,6 +2381,37 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
   }
 }
 
+/* FullOuter second half: one (NIL, build_rid) pair per unmatched build
+ * row — every build row never touched by a probe (incl. NULL-key build
+ * rows, which are never inserted and so never matched). */
+__global__ void k_join_unmatched_build(int64_t brows,
+                                       const unsigned long long* matched,
+                                       const unsigned int* brid_map,
+                                       join_sp* sp,
+                                       uint32_t* out_p, uint32_t* out_b,
+                                       int64_t out_cap) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t gs = (int64_t)gridDim.x * blockDim.x;
+  const int lane = threadIdx.x & (WAVE - 1);
+  for (; i - lane < brows; i += gs) {
+    bool un = i < brows && !((matched[i >> 6] >> (i & 63)) & 1);
+    uint64_t m = __ballot(un);
+    if (!m) continue;
+    int cntw = __popcll(m);
+    unsigned long long base = 0;
+    int leader = __ffsll((unsigned long long)m) - 1;
+    if (lane == leader)
+      base = atomicAdd(&sp->cursor, (unsigned long long)cntw);
+    base = __shfl(base, leader);
+    if (!un) continue;
+    int64_t o = (int64_t)base + __popcll(m & ((1ULL << lane) - 1));
+    if (o < out_cap) {
+      out_p[o] = JOIN_NIL;
+      out_b[o] = brid_map ? brid_map[i] : (unsigned int)i;
+    }
+  }
+}
+
 extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
                                    void* workspace, int64_t cap) {
   hipStream_t s = (hipStream_t)stream;
@@ -2386,6 +2422,7 @@ extern "C" int gpuq_join_build_i64(void* stream, int64_t brows, gpuq_col bkey,
   join_ws w; int64_t need;
   join_ws_layout(cap, brows, &w, (char*)workspace, &need);
   HIP_TRY(hipMemsetAsync(w.slots, 0xFF, cap * 16, s));  /* keys=-1, heads=NIL */
+  HIP_TRY(hipMemsetAsync(w.matched, 0, ((brows + 63) / 64) * 8, s));
   HIP_TRY(hipMemsetAsync(w.sp, 0xFF, 4, s));            /* m1_head = NIL */
   HIP_TRY(hipMemsetAsync(&w.sp->bucketed, 0, 4, s));
   HIP_TRY(hipMemsetAsync(&w.sp->cursor, 0, 8, s));
@@ -2422,7 +2459,7 @@ extern "C" int gpuq_join_probe_i64_typed(void* stream, int64_t prows, gpuq_col p
                                    uint32_t* out_p, uint32_t* out_b,
                                    int64_t out_cap, int64_t* out_nmatches) {
   hipStream_t s = (hipStream_t)stream;
-  if (join_type < 0 || join_type > 3)
+  if (join_type < 0 || join_type > 4)
     FAIL(GPUQ_ERR_INVALID, "join: bad join_type %d", join_type);
   if (pkey.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "join: key must be int64");
   join_ws w; int64_t need;
@@ -2466,10 +2503,25 @@ extern "C" int gpuq_join_probe_i64_typed(void* stream, int64_t prows, gpuq_col p
     else if (join_type == 3)
       k_join_probe<3><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
                                          bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);
+    else if (join_type == 4)
+      k_join_probe<4><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
+                                         bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap,
+                                         w.matched);
     else
       k_join_probe<0><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
                                          bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);
     prof_end("join_probe", s, _pe); }
+    HIP_TRY(hipGetLastError());
+  }
+  if (join_type == 4 && brows > 0) {
+    /* FullOuter second half; probes with zero rows still emit every
+     * build row */
+    join_sp hsp0b;
+    HIP_TRY(hipMemcpyAsync(&hsp0b, w.sp, sizeof(hsp0b), hipMemcpyDeviceToHost, s));
+    HIP_TRY(hipStreamSynchronize(s));
+    k_join_unmatched_build<<<grid1d(brows), 256, 0, s>>>(
+        brows, w.matched, hsp0b.bucketed ? w.brid : nullptr, w.sp,
+        out_p, out_b, out_cap);
     HIP_TRY(hipGetLastError());
   }
   join_sp hsp;

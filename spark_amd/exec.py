@@ -812,9 +812,10 @@ class GpuShuffledHashJoinExec(SparkPlan):
         super().__init__(left, right)
         assert build_side in ("left", "right")
         assert join_type in ("inner", "left_outer", "right_outer",
-                             "left_semi", "left_anti")
+                             "left_semi", "left_anti", "full_outer")
         # the preserved/streamed side must be the PROBE side
-        # (HashJoin.scala buildSide constraints)
+        # (HashJoin.scala buildSide constraints; full outer preserves both
+        # — either side may build)
         if join_type in ("left_outer", "left_semi", "left_anti"):
             assert build_side == "right", f"{join_type} builds on the right"
         if join_type == "right_outer":
@@ -847,13 +848,15 @@ class GpuShuffledHashJoinExec(SparkPlan):
         pkey = self.right_key if self.build_side == "left" else self.left_key
         jt = {"inner": gpuq.JOIN_INNER, "left_outer": gpuq.JOIN_OUTER,
               "right_outer": gpuq.JOIN_OUTER, "left_semi": gpuq.JOIN_SEMI,
-              "left_anti": gpuq.JOIN_ANTI}[self.join_type]
+              "left_anti": gpuq.JOIN_ANTI,
+              "full_outer": gpuq.JOIN_FULL}[self.join_type]
         bk = build.column(bkey)
         bn = bk.numel()
         cap = 1 << max(4, int(bn * 2 - 1).bit_length() if bn else 4)
         ws = gpuq.join_build(bk, cap, key_validity=build.validity(bkey))
         semi = self.join_type in ("left_semi", "left_anti")
-        outer = self.join_type in ("left_outer", "right_outer")
+        outer = self.join_type in ("left_outer", "right_outer", "full_outer")
+        full = self.join_type == "full_outer"
         # the probe (streamed) side is consumed batch-at-a-time
         # (ShuffledHashJoinExec.doExecute streams streamedIter)
         for probe in self.children[1 - bi].execute_columnar():
@@ -883,9 +886,14 @@ class GpuShuffledHashJoinExec(SparkPlan):
                 v = probe.validity(name)
                 if name in cols:
                     name = f"{name}#probe"
-                cols[name] = gpuq.gather(t, op)
-                if v is not None:
-                    validity[name] = gpuq.gather_bits(v, op)
+                if full:
+                    # build-side unmatched rows carry NIL probe rids
+                    cols[name], validity[name] = gpuq.gather_nullable(
+                        t, op, validity=v)
+                else:
+                    cols[name] = gpuq.gather(t, op)
+                    if v is not None:
+                        validity[name] = gpuq.gather_bits(v, op)
             probe.close()
             yield ColumnarBatch(cols, validity=validity or None)
         build.close()

@@ -326,7 +326,7 @@ def test_partial_final_merge_exec(gq):
 
 # ---------- typed joins (outer / semi / anti over the probe side) ----------
 
-@pytest.mark.parametrize("jt_name", ["left_outer", "left_semi", "left_anti"])
+@pytest.mark.parametrize("jt_name", ["left_outer", "left_semi", "left_anti", "full_outer"])
 def test_typed_join_parity(gq, jt_name):
     """probe-side outer/semi/anti against a python reference, with NULL
     probe keys, duplicate build chains and unmatched rows
@@ -355,12 +355,15 @@ def test_typed_join_parity(gq, jt_name):
     exp = []   # (probe_rid, build_rid or None)
     for i in range(pn):
         hits = buckets.get(int(pkeys[i]), []) if pvalid[i] else []
-        if jt_name == "left_outer":
+        if jt_name in ("left_outer", "full_outer"):
             exp += [(i, j) for j in hits] if hits else [(i, None)]
         elif jt_name == "left_semi":
             exp += [(i, None)] if hits else []
         else:
             exp += [] if hits else [(i, None)]
+    if jt_name == "full_outer":
+        matched_b = {j for _, j in exp if j is not None}
+        exp += [(None, j) for j in range(bn) if j not in matched_b]
     got_lk = out.column("lk").cpu().numpy()
     got_lp = out.column("lp").cpu().numpy()
     lkv = out.validity("lk")
@@ -370,21 +373,28 @@ def test_typed_join_parity(gq, jt_name):
     assert len(got_lk) == len(exp)
     # compare as multisets of full rows
     def row(i, j):
+        if i is None:
+            return (None, None, int(bkeys[j]), int(bpay[j]))
         lk = int(pkeys[i]) if pvalid[i] else None
-        if jt_name == "left_outer":
+        if jt_name in ("left_outer", "full_outer"):
             rp = int(bpay[j]) if j is not None else None
             rk = int(bkeys[j]) if j is not None else None
             return (lk, int(ppay[i]), rk, rp)
         return (lk, int(ppay[i]))
     expected = sorted((row(i, j) for i, j in exp),
                       key=lambda r: tuple((x is None, x or 0) for x in r))
-    if jt_name == "left_outer":
+    if jt_name in ("left_outer", "full_outer"):
         got_rk = out.column("rk").cpu().numpy()
         got_rp = out.column("rp").cpu().numpy()
         rkv = out.validity("rk")
         got_rkv = np.unpackbits(rkv.cpu().numpy(), count=len(got_lk),
                                 bitorder="little").astype(bool)
-        got = [(int(got_lk[i]) if got_lkv[i] else None, int(got_lp[i]),
+        lpv = out.validity("lp")
+        got_lpv = (np.unpackbits(lpv.cpu().numpy(), count=len(got_lk),
+                                 bitorder="little").astype(bool)
+                   if lpv is not None else np.ones(len(got_lk), bool))
+        got = [(int(got_lk[i]) if got_lkv[i] else None,
+                int(got_lp[i]) if got_lpv[i] else None,
                 int(got_rk[i]) if got_rkv[i] else None,
                 int(got_rp[i]) if got_rkv[i] else None)
                for i in range(len(got_lk))]
