@@ -933,7 +933,10 @@ static scatter_geom get_sort_geom(void) {
     const char* e = getenv("GPUQ_SORT_GEOM");
     int b = 0, it = 0;
     if (e && sscanf(e, "%dx%d", &b, &it) == 2) { g.block = b; g.items = it; }
-    else { g.block = 512; g.items = 10; }
+    else { g.block = 256; g.items = 22; }  /* two-box A/B winner: ~5% over
+                                            * 512x10 (fewer waves per
+                                            * barrier at the same tile
+                                            * residency) */
     /* only geometries with a dispatch entry are legal: an unknown pair
      * would silently run a different tile than the block count assumed
      * (incomplete sort). Clamp to the default. */
@@ -944,9 +947,9 @@ static scatter_geom get_sort_geom(void) {
     bool ok = false;
     for (auto& k : known) ok = ok || (k[0] == g.block && k[1] == g.items);
     if (!ok) {
-      fprintf(stderr, "gpuq: unsupported GPUQ_SORT_GEOM %dx%d, using 512x10\n",
+      fprintf(stderr, "gpuq: unsupported GPUQ_SORT_GEOM %dx%d, using 256x22\n",
               g.block, g.items);
-      g.block = 512; g.items = 10;
+      g.block = 256; g.items = 22;
     }
   }
   return g;
@@ -1291,10 +1294,20 @@ static void launch_scatter4(hipStream_t s, scatter_geom g, int64_t nb,
     k_radix_scatter<BIN_MODE, 1024, 8, LOOKBACK, uint32_t, 4><<<grid, 1024, 0, s>>>(
         n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
         err_flag, epoch, decode_out, decode_mode);
-  else
+  else if (g.block == 256 && g.items == 22)
+    k_radix_scatter<BIN_MODE, 256, 22, LOOKBACK, uint32_t, 4><<<grid, 256, 0, s>>>(
+        n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+        err_flag, epoch, decode_out, decode_mode);
+  else if (g.block == 512 && g.items == 10)
     k_radix_scatter<BIN_MODE, 512, 10, LOOKBACK, uint32_t, 4><<<grid, 512, 0, s>>>(
         n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
         err_flag, epoch, decode_out, decode_mode);
+  else {
+    fprintf(stderr, "gpuq: 4-bit mode needs geom 1024x8/512x10/256x22; using 256x22 kernel\n");
+    k_radix_scatter<BIN_MODE, 256, 22, LOOKBACK, uint32_t, 4><<<grid, 256, 0, s>>>(
+        n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
+        err_flag, epoch, decode_out, decode_mode);
+  }
 }
 
 /* decode sorted keys back to the output dtype */
