@@ -1302,12 +1302,10 @@ static void launch_scatter4(hipStream_t s, scatter_geom g, int64_t nb,
     k_radix_scatter<BIN_MODE, 512, 10, LOOKBACK, uint32_t, 4><<<grid, 512, 0, s>>>(
         n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
         err_flag, epoch, decode_out, decode_mode);
-  else {
-    fprintf(stderr, "gpuq: 4-bit mode needs geom 1024x8/512x10/256x22; using 256x22 kernel\n");
-    k_radix_scatter<BIN_MODE, 256, 22, LOOKBACK, uint32_t, 4><<<grid, 256, 0, s>>>(
+  else
+    k_radix_scatter<BIN_MODE, 512, 10, LOOKBACK, uint32_t, 4><<<grid, 512, 0, s>>>(
         n, kin, iin, kout, iout, scanned, shift, (int)nb, nparts, state, gbase,
-        err_flag, epoch, decode_out, decode_mode);
-  }
+        err_flag, epoch, decode_out, decode_mode);  /* unreachable: gated above */
 }
 
 /* decode sorted keys back to the output dtype */
@@ -1462,6 +1460,12 @@ extern "C" int gpuq_sort_perm(void* stream, int64_t n, gpuq_col key,
 
   bool nibble_mode = onesweep && getenv("GPUQ_SORT_BITS") != nullptr &&
                      atoi(getenv("GPUQ_SORT_BITS")) == 4;
+  /* the 4-bit kernel templates exist for these geometries only; a tile
+   * mismatch would overlap blocks (wrong results) — fall back to 8-bit */
+  if (nibble_mode && !((geom.block == 1024 && geom.items == 8) ||
+                       (geom.block == 512 && geom.items == 10) ||
+                       (geom.block == 256 && geom.items == 22)))
+    nibble_mode = false;
   int retries = 0;
 retry:
   if (onesweep && !nibble_mode) {
