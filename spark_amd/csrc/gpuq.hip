@@ -2295,7 +2295,7 @@ __global__ void k_join_build(int64_t n, const int64_t* keys, const uint8_t* kval
 DEV uint32_t jt_emit_count(int jt, uint32_t m) {
   if (jt == 1 || jt == 4) return m ? m : 1u;  /* 4 = FullOuter probe half */
   if (jt == 2) return m ? 1u : 0u;
-  if (jt == 3) return m ? 0u : 1u;
+  if (jt == 3 || jt == 5) return m ? 0u : 1u; /* 5 = null-aware anti */
   return m;
 }
 
@@ -2353,6 +2353,10 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
         }
       }
     }
+    if (JT == 5 && i < n && !bit_valid(kvalid, i))
+      cnt[r] = 1;  /* null-aware anti: NULL NOT IN (non-empty) is unknown
+                    * -> the row is filtered (treated as matched);
+                    * BroadcastHashJoinExec.scala:137 NAAJ semantics */
     if (JT == 0) lane_total += cnt[r];
     else if (i < n) lane_total += jt_emit_count(JT, cnt[r]);
   }
@@ -2375,7 +2379,7 @@ __global__ void k_join_probe(int64_t n, const int64_t* keys, const uint8_t* kval
     if (JT != 0) {
       if (i >= n || !jt_emit_count(JT, cnt[r])) continue;
       uint32_t pr = prid_map ? prid_map[i] : (uint32_t)i;
-      if (JT == 2 || JT == 3 || cnt[r] == 0) {
+      if (JT == 2 || JT == 3 || JT == 5 || cnt[r] == 0) {
         /* semi/anti emit the probe row once; outer's unmatched row pairs
          * with NIL (NULL build columns downstream) */
         if (o < out_cap) { out_p[o] = pr; out_b[o] = JOIN_NIL; }
@@ -2476,7 +2480,7 @@ extern "C" int gpuq_join_probe_i64_typed(void* stream, int64_t prows, gpuq_col p
                                    uint32_t* out_p, uint32_t* out_b,
                                    int64_t out_cap, int64_t* out_nmatches) {
   hipStream_t s = (hipStream_t)stream;
-  if (join_type < 0 || join_type > 4)
+  if (join_type < 0 || join_type > 5)
     FAIL(GPUQ_ERR_INVALID, "join: bad join_type %d", join_type);
   if (pkey.dtype != GPUQ_INT64) FAIL(GPUQ_ERR_INVALID, "join: key must be int64");
   join_ws w; int64_t need;
@@ -2524,6 +2528,9 @@ extern "C" int gpuq_join_probe_i64_typed(void* stream, int64_t prows, gpuq_col p
       k_join_probe<4><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
                                          bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap,
                                          w.matched);
+    else if (join_type == 5)
+      k_join_probe<5><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
+                                         bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);
     else
       k_join_probe<0><<<jg, 256, 0, s>>>(prows, pkeys, pvalid, w.slots, w.next,
                                          bm, prid_map, w.sp, cap - 1, out_p, out_b, out_cap);

@@ -808,11 +808,17 @@ class GpuShuffledHashJoinExec(SparkPlan):
     exchanges, as EnsureRequirements would arrange."""
 
     def __init__(self, left_key: str, right_key: str, build_side: str,
-                 left, right, join_type: str = "inner"):
+                 left, right, join_type: str = "inner",
+                 null_aware_anti: bool = False):
         super().__init__(left, right)
         assert build_side in ("left", "right")
         assert join_type in ("inner", "left_outer", "right_outer",
                              "left_semi", "left_anti", "full_outer")
+        # NOT IN rewrite (BroadcastHashJoinExec.scala:48,137
+        # isNullAwareAntiJoin): empty build -> all probe rows; any NULL
+        # build key -> empty result; NULL probe keys are filtered
+        assert not null_aware_anti or join_type == "left_anti"
+        self.null_aware_anti = null_aware_anti
         # the preserved/streamed side must be the PROBE side
         # (HashJoin.scala buildSide constraints; full outer preserves both
         # — either side may build)
@@ -852,6 +858,23 @@ class GpuShuffledHashJoinExec(SparkPlan):
               "full_outer": gpuq.JOIN_FULL}[self.join_type]
         bk = build.column(bkey)
         bn = bk.numel()
+        if self.null_aware_anti:
+            if bn == 0:
+                # NOT IN (empty) is true for every row, NULL included
+                yield from self.children[1 - bi].execute_columnar()
+                build.close()
+                return
+            _, _, valid_cnt = gpuq.minmax_i64(bk,
+                                              validity=build.validity(bkey))
+            if valid_cnt < bn:
+                # any NULL in the build side: NOT IN is never true
+                for probe in self.children[1 - bi].execute_columnar():
+                    empty = {n_: t[:0] for n_, t in probe.columns().items()}
+                    probe.close()
+                    yield ColumnarBatch(empty)
+                build.close()
+                return
+            jt = gpuq.JOIN_ANTI_NULLAWARE
         cap = 1 << max(4, int(bn * 2 - 1).bit_length() if bn else 4)
         ws = gpuq.join_build(bk, cap, key_validity=build.validity(bkey))
         semi = self.join_type in ("left_semi", "left_anti")

@@ -316,7 +316,7 @@ def join_probe_workspace(probe_rows: int, device="cuda") -> torch.Tensor:
                        dtype=torch.uint8, device=device)
 
 
-JOIN_INNER, JOIN_OUTER, JOIN_SEMI, JOIN_ANTI, JOIN_FULL = 0, 1, 2, 3, 4
+JOIN_INNER, JOIN_OUTER, JOIN_SEMI, JOIN_ANTI, JOIN_FULL, JOIN_ANTI_NULLAWARE = 0, 1, 2, 3, 4, 5
 JOIN_NIL = 0xFFFFFFFF
 
 

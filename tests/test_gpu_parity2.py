@@ -421,3 +421,43 @@ def test_filter_fractional_literal_on_int64(gq):
         exp = np.flatnonzero(ref)
         assert cnt == len(exp), (op, lit, cnt, len(exp))
         assert (perm.cpu().numpy().astype(np.int64) == exp).all(), (op, lit)
+
+
+def test_null_aware_anti_join(gq):
+    """NOT IN rewrite (BroadcastHashJoinExec.scala isNullAwareAntiJoin):
+    empty build -> all probe rows; any NULL build key -> empty; otherwise
+    anti with NULL probe keys filtered (NULL NOT IN (...) is unknown)."""
+    from spark_amd import exec as gx
+    pn = 50_000
+    pkeys = oracle.gen_i64(seed=700, n=pn, range_=1000)
+    pvalid = oracle.gen_i64(seed=701, n=pn, range_=8) != 0
+
+    def probe_batches():
+        return gx.InputBatches([gx.ColumnarBatch(
+            {"lk": to_dev(pkeys)}, validity={"lk": pack_validity(pvalid)})])
+
+    def run(bkeys, bvalid=None):
+        right = gx.InputBatches([gx.ColumnarBatch(
+            {"rk": to_dev(bkeys)},
+            validity={"rk": pack_validity(bvalid)} if bvalid is not None
+            else None)])
+        node = gx.GpuShuffledHashJoinExec(
+            "lk", "rk", "right", probe_batches(), right,
+            join_type="left_anti", null_aware_anti=True)
+        return list(node.execute_columnar())[0]
+
+    # empty build: every probe row (NULLs included)
+    out = run(np.empty(0, dtype=np.int64))
+    assert out.num_rows() == pn
+    # build containing a NULL key: empty result
+    bk = oracle.gen_i64(seed=702, n=100, range_=500)
+    bv = np.ones(100, dtype=bool); bv[17] = False
+    out = run(bk, bv)
+    assert out.num_rows() == 0
+    # plain: probe keys not in the build set, NULL probe keys filtered
+    out = run(bk)
+    bset = set(bk.tolist())
+    exp = sorted(int(pkeys[i]) for i in range(pn)
+                 if pvalid[i] and int(pkeys[i]) not in bset)
+    got = sorted(out.column("lk").cpu().numpy().tolist())
+    assert got == exp
