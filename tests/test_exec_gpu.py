@@ -304,3 +304,46 @@ def test_q3_plan_parity():
     for i in range(1, len(gk)):
         assert (gr[i - 1] > gr[i]
                 or (gr[i - 1] == gr[i] and gd[i - 1] <= gd[i]))
+
+
+def test_empty_batch_pipeline():
+    """degenerate shapes: a filter that drops every row feeding sort, agg
+    (grouped AND global), and join must stay well-formed end-to-end."""
+    n = 10_000
+    keys = oracle.gen_i64(seed=950, n=n, range_=100)
+    vals = oracle.gen_f64_unit(seed=951, n=n)
+    scan = gx.InputBatches([dev_batch(k=keys, v=vals)])
+    dead = gx.FilterExec("k", "<", -1, scan)   # nothing passes
+    out = list(gx.GpuColumnarRule().pre_columnar_transitions(
+        gx.SortExec(gx.SortOrder("k"), False,
+                    gx.HashAggregateExec("k", [("sum", "v"), ("count*", None)],
+                                         "complete", dead)))
+        .execute_columnar())[0]
+    assert out.num_rows() == 0
+
+    # global aggregate over empty input: exactly one row, COUNT 0, SUM NULL
+    scan2 = gx.InputBatches([dev_batch(k=keys, v=vals)])
+    dead2 = gx.FilterExec("k", "<", -1, scan2)
+    out = list(gx.GpuColumnarRule().pre_columnar_transitions(
+        gx.HashAggregateExec(None, [("count*", None), ("sum", "v")],
+                             "complete", dead2)).execute_columnar())[0]
+    assert out.num_rows() == 1
+    assert int(out.column("count(1)").cpu()[0]) == 0
+    sv = out.validity("sum(v)")
+    assert sv is not None and int(sv.cpu()[0]) & 1 == 0  # SUM -> NULL
+
+    # join with an empty probe side / empty build side
+    empty = gx.FilterExec("k", "<", -1,
+                          gx.InputBatches([dev_batch(k=keys)]))
+    right = gx.InputBatches([dev_batch(rk=keys)])
+    out = list(gx.GpuColumnarRule().pre_columnar_transitions(
+        gx.ShuffledHashJoinExec("k", "rk", "right", empty, right))
+        .execute_columnar())[0]
+    assert out.num_rows() == 0
+    empty2 = gx.FilterExec("rk", "<", -1,
+                           gx.InputBatches([dev_batch(rk=keys)]))
+    left = gx.InputBatches([dev_batch(k=keys)])
+    out = list(gx.GpuColumnarRule().pre_columnar_transitions(
+        gx.ShuffledHashJoinExec("k", "rk", "right", left, empty2))
+        .execute_columnar())[0]
+    assert out.num_rows() == 0
