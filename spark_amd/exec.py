@@ -596,7 +596,8 @@ class GpuHashAggregateExec(SparkPlan):
                     except KeyError:
                         return torch.float64
                 names = (list(slot) if self.mode == "partial"
-                         else [f"{fn}({col})" for fn, col in self.aggs])
+                         else [self.agg_out_name(fn, col)
+                               for fn, col in self.aggs])
                 for name in names:
                     if name.startswith("count"):
                         cols[name] = torch.zeros(1, dtype=torch.int64, device=dev)
