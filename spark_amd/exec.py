@@ -502,8 +502,14 @@ class GpuHashAggregateExec(SparkPlan):
         return [Distribution("unspecified")]
 
     def execute_columnar(self):
-        for batch in self.children[0].execute_columnar():
-            yield self._agg_batch(batch)
+        # HashAggregateExec aggregates the WHOLE partition (one
+        # TungstenAggregationIterator per task), not one table per input
+        # batch: all child batches accumulate into a single hash table
+        # (the kernel's first_batch/finalize contract) and ONE result
+        # batch emits per partition.
+        batches = list(self.children[0].execute_columnar())
+        assert batches, "aggregate over a batchless child"
+        yield self._agg_batches(batches)
 
     # ---- spec construction ----
 
@@ -571,28 +577,32 @@ class GpuHashAggregateExec(SparkPlan):
             specs.append(("count*",))
         return specs, slot
 
-    def _agg_batch(self, batch):
+    def _agg_batches(self, batches):
         from . import gpuq
         keys = self.group_keys
-        n = batch.num_rows()
-        cap = self.capacity or (1 << max(10, int(n).bit_length()))
+        total = sum(b.num_rows() for b in batches)
+        cap = self.capacity or (1 << max(10, int(total).bit_length()))
         if self.mode == "final":
-            specs, slot = self._merge_specs(batch)
+            per_specs = [self._merge_specs(b) for b in batches]
         else:
-            specs, slot = self._input_specs(batch)
-        mg = min(n, cap) + 2
+            per_specs = [self._input_specs(b) for b in batches]
+        slot = per_specs[0][1]
+        mg = min(total, cap) + 2
+        nspecs = len(per_specs[0][0])
+        last_i = len(batches) - 1
         if len(keys) == 0:
-            # global aggregate (empty grouping, HashAggregateExec.scala with
-            # no grouping expressions): one output row even on empty input —
-            # COUNT 0, SUM/MIN/MAX/AVG NULL (AggUtils emptyInputAggBuffer)
-            if n == 0:
+            # global aggregate (empty grouping): one output row even on
+            # empty input — COUNT 0, SUM/MIN/MAX/AVG NULL (AggUtils
+            # emptyInputAggBuffer)
+            if total == 0:
                 cols, validity = {}, {}
                 dev = "cuda"
+                b0 = batches[0]
 
                 def src_dtype(name):
                     base = name.split("(", 1)[1][:-1]
                     try:
-                        return batch.column(base).dtype
+                        return b0.column(base).dtype
                     except KeyError:
                         return torch.float64
                 names = (list(slot) if self.mode == "partial"
@@ -611,53 +621,90 @@ class GpuHashAggregateExec(SparkPlan):
                                                  device=dev)
                         validity[name] = torch.zeros(1, dtype=torch.uint8,
                                                      device=dev)
-                batch.close()
+                for b in batches:
+                    b.close()
                 return ColumnarBatch(cols, validity=validity or None)
-            key0 = gpuq.range_i64(n, 0, 0)   # constant key: one group
-            ok, okv, accs = gpuq.hash_agg_multi(key0, specs, cap, max_groups=mg)
+            ws = gpuq.agg_multi_workspace(cap, nspecs)
+            res = None
+            for i, (b, (specs, _)) in enumerate(zip(batches, per_specs)):
+                key0 = gpuq.range_i64(b.num_rows(), 0, 0)  # one group
+                res = gpuq.hash_agg_multi(key0, specs, cap, workspace=ws,
+                                          first_batch=(i == 0),
+                                          finalize=(i == last_i),
+                                          max_groups=mg)
+            ok, okv, accs = res
             key_cols, key_valid = {}, {}
         elif len(keys) == 1:
             k = keys[0]
-            ok, okv, accs = gpuq.hash_agg_multi(
-                batch.column(k), specs, cap,
-                key_validity=batch.validity(k), max_groups=mg)
+            ws = gpuq.agg_multi_workspace(cap, nspecs)
+            res = None
+            any_kv = any(b.validity(k) is not None for b in batches)
+            for i, (b, (specs, _)) in enumerate(zip(batches, per_specs)):
+                res = gpuq.hash_agg_multi(
+                    b.column(k), specs, cap, workspace=ws,
+                    key_validity=b.validity(k),
+                    first_batch=(i == 0), finalize=(i == last_i),
+                    max_groups=mg)
+            ok, okv, accs = res
             key_cols = {k: ok}
-            # NULL-key group: okv==0 row. Only materialize a bitmap when the
-            # input key was nullable.
+            # NULL-key group: okv==0 row. Only materialize a bitmap when
+            # some input batch's key was nullable.
             key_valid = {}
-            if batch.validity(k) is not None:
+            if any_kv:
                 key_valid[k] = gpuq.u8_to_bits(okv)
         else:
-            kc = [batch.column(k) for k in keys]
-            kv = [batch.validity(k) for k in keys]
+            kcs = [[b.column(k) for k in keys] for b in batches]
+            kvs = [[b.validity(k) for k in keys] for b in batches]
             packed = None
-            if (len(keys) == 2 and kv[0] is None and kv[1] is None
-                    and kc[0].dtype == torch.int64
-                    and kc[1].dtype == torch.int64):
-                # narrow-tuple pack rule: when both key ranges fit, pack
-                # (k1,k2) into one int64 so the single-key path (incl. its
-                # per-block LDS tables at low cardinality — the TPC-H Q1
-                # (returnflag, linestatus) shape) runs instead of the
-                # verify-slot composite table
-                mn0, mx0, c0 = gpuq.minmax_i64(kc[0])
-                mn1, mx1, c1 = gpuq.minmax_i64(kc[1])
-                if c0 and c1:
+            if (len(keys) == 2
+                    and all(v[0] is None and v[1] is None for v in kvs)
+                    and kcs[0][0].dtype == torch.int64
+                    and kcs[0][1].dtype == torch.int64):
+                # narrow-tuple pack rule: when the GLOBAL key ranges fit,
+                # pack (k1,k2) into one int64 so the single-key path
+                # (incl. its per-block LDS tables at low cardinality —
+                # the TPC-H Q1 (returnflag, linestatus) shape) runs
+                # instead of the verify-slot composite table
+                mm = [(gpuq.minmax_i64(kc[0]), gpuq.minmax_i64(kc[1]))
+                      for kc in kcs]
+                if all(a[2] and b_[2] for a, b_ in mm):
+                    mn0 = min(a[0] for a, _ in mm)
+                    mx0 = max(a[1] for a, _ in mm)
+                    mn1 = min(b_[0] for _, b_ in mm)
+                    mx1 = max(b_[1] for _, b_ in mm)
                     shift = max(1, int(mx1 - mn1).bit_length())
                     if (mx0 - mn0) < (1 << (63 - shift)):
-                        packed = gpuq.pack2_i64(kc[0], kc[1], mn0, mn1, shift)
+                        packed = [gpuq.pack2_i64(kc[0], kc[1], mn0, mn1,
+                                                 shift) for kc in kcs]
             if packed is not None:
-                ok, okv, accs = gpuq.hash_agg_multi(packed, specs, cap,
-                                                    max_groups=mg)
+                ws = gpuq.agg_multi_workspace(cap, nspecs)
+                res = None
+                for i, (pk, (specs, _)) in enumerate(zip(packed, per_specs)):
+                    res = gpuq.hash_agg_multi(pk, specs, cap, workspace=ws,
+                                              first_batch=(i == 0),
+                                              finalize=(i == last_i),
+                                              max_groups=mg)
+                ok, okv, accs = res
                 k0, k1 = gpuq.unpack2_i64(ok, mn0, mn1, shift)
                 key_cols = {keys[0]: k0, keys[1]: k1}
                 key_valid = {}
             else:
-                okeys, kmask, accs = gpuq.hash_agg_keys(
-                    kc, specs, cap, key_validities=kv, max_groups=mg)
+                ws = torch.empty(
+                    gpuq.lib().gpuq_hash_agg_keys_workspace_bytes(
+                        cap, len(keys), nspecs),
+                    dtype=torch.uint8, device="cuda")
+                res = None
+                for i, (kc, kv, (specs, _)) in enumerate(
+                        zip(kcs, kvs, per_specs)):
+                    res = gpuq.hash_agg_keys(
+                        kc, specs, cap, key_validities=kv, workspace=ws,
+                        first_batch=(i == 0), finalize=(i == last_i),
+                        max_groups=mg)
+                okeys, kmask, accs = res
                 key_cols = dict(zip(keys, okeys))
                 key_valid = {}
                 for c, k in enumerate(keys):
-                    if kv[c] is not None:
+                    if any(v[c] is not None for v in kvs):
                         key_valid[k] = gpuq.maskbit_to_bits(kmask, c)
         cols, validity = dict(key_cols), dict(key_valid)
         if self.mode == "partial":
@@ -692,7 +739,8 @@ class GpuHashAggregateExec(SparkPlan):
                 # no companion => non-null complete-mode input, always valid
                 if cnt_slot is not None:
                     validity[out] = gpuq.nonzero_to_bits(accs[cnt_slot])
-        batch.close()
+        for b in batches:
+            b.close()
         return ColumnarBatch(cols, validity=validity or None)
 
 

@@ -467,32 +467,44 @@ def _build_agg_specs(specs, dev, mg, placeholder):
     return cols_arr, ops_arr, cid_arr, outs
 
 
+def agg_multi_workspace(capacity: int, nspecs: int, device="cuda"):
+    return torch.empty(lib().gpuq_hash_agg_multi_workspace_bytes(capacity, nspecs),
+                       dtype=torch.uint8, device=device)
+
+
 def hash_agg_multi(keys: torch.Tensor, specs, capacity: int, key_validity=None,
-                   max_groups=None):
+                   max_groups=None, workspace=None, first_batch=True,
+                   finalize=True):
     """Multi-accumulator GROUP BY (single int64 key). specs per
-    _build_agg_specs. Returns (keys, key_valid, [acc_j...]) sliced to
-    ngroups."""
+    _build_agg_specs. Multiple batches accumulate into the same workspace
+    (first_batch=True resets it; finalize=True compacts) — the
+    whole-partition aggregation contract. Returns (keys, key_valid,
+    [acc_j...]) sliced to ngroups, or None before finalize."""
     n = keys.numel()
     dev = keys.device
     nspecs = len(specs)
     mg = max_groups or capacity + 2
     cols_arr, ops_arr, cid_arr, outs = _build_agg_specs(specs, dev, mg, keys)
     outp_arr = (ctypes.c_void_p * nspecs)(*[t.data_ptr() for t in outs])
-    ws = torch.empty(lib().gpuq_hash_agg_multi_workspace_bytes(capacity, nspecs),
-                     dtype=torch.uint8, device=dev)
+    if workspace is None:
+        workspace = agg_multi_workspace(capacity, nspecs, dev)
     ok = torch.empty(mg, dtype=torch.int64, device=dev)
     okv = torch.empty(mg, dtype=torch.uint8, device=dev)
     ng = ctypes.c_int64(0)
     _check(lib().gpuq_hash_agg_multi(
         _stream(), n, _col(keys, key_validity), cols_arr, ops_arr, cid_arr,
-        nspecs, ws.data_ptr(), capacity, 1, 1, ok.data_ptr(), okv.data_ptr(),
+        nspecs, workspace.data_ptr(), capacity, int(first_batch),
+        int(finalize), ok.data_ptr(), okv.data_ptr(),
         outp_arr, ctypes.byref(ng)))
+    if not finalize:
+        return None
     gn = ng.value
     return ok[:gn], okv[:gn], [t[:gn] for t in outs]
 
 
 def hash_agg_keys(key_cols, specs, capacity: int, key_validities=None,
-                  max_groups=None):
+                  max_groups=None, workspace=None, first_batch=True,
+                  finalize=True):
     """Composite-key GROUP BY (k1..kK), K <= 4 int64 columns.
     key_validities: optional list (same length) of validity bitmaps.
     Returns (key_out_cols list, kmask u8 tensor, [acc_j...]) sliced to
@@ -509,13 +521,17 @@ def hash_agg_keys(key_cols, specs, capacity: int, key_validities=None,
     okeys = [torch.empty(mg, dtype=torch.int64, device=dev) for _ in range(nkeys)]
     okp_arr = (ctypes.c_void_p * nkeys)(*[t.data_ptr() for t in okeys])
     omask = torch.empty(mg, dtype=torch.uint8, device=dev)
-    ws = torch.empty(lib().gpuq_hash_agg_keys_workspace_bytes(capacity, nkeys, nspecs),
-                     dtype=torch.uint8, device=dev)
+    if workspace is None:
+        workspace = torch.empty(
+            lib().gpuq_hash_agg_keys_workspace_bytes(capacity, nkeys, nspecs),
+            dtype=torch.uint8, device=dev)
     ng = ctypes.c_int64(0)
     _check(lib().gpuq_hash_agg_keys(
         _stream(), n, keys_arr, nkeys, cols_arr, ops_arr, cid_arr, nspecs,
-        ws.data_ptr(), capacity, 1, 1, okp_arr, omask.data_ptr(),
-        outp_arr, ctypes.byref(ng)))
+        workspace.data_ptr(), capacity, int(first_batch), int(finalize),
+        okp_arr, omask.data_ptr(), outp_arr, ctypes.byref(ng)))
+    if not finalize:
+        return None
     gn = ng.value
     return [t[:gn] for t in okeys], omask[:gn], [t[:gn] for t in outs]
 

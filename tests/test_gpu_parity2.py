@@ -278,11 +278,16 @@ def test_partial_final_merge_exec(gq):
             validity={"v": pack_validity(valid[lo:hi])}))
     aggs = [("sum", "v"), ("count", "v"), ("avg", "v"),
             ("min", "v"), ("max", "v"), ("count*", None)]
-    partial_node = gx.HashAggregateExec("k", aggs, "partial",
-                                        gx.InputBatches(batches))
-    gpu_partial = gx.GpuColumnarRule().pre_columnar_transitions(partial_node)
-    partials = list(gpu_partial.execute_columnar())
-    assert len(partials) == 2
+    # two partial nodes = two RANKS, each aggregating its own partition
+    # batches (the whole-partition contract: one output per partition)
+    partials = []
+    for bat in batches:
+        node = gx.HashAggregateExec("k", aggs, "partial",
+                                    gx.InputBatches([bat]))
+        outs = list(gx.GpuColumnarRule().pre_columnar_transitions(node)
+                    .execute_columnar())
+        assert len(outs) == 1
+        partials.append(outs[0])
     # concat partial outputs into one batch (what the exchange would yield)
     cols, validity = {}, {}
     names = list(partials[0].columns().keys())
@@ -461,3 +466,49 @@ def test_null_aware_anti_join(gq):
                  if pvalid[i] and int(pkeys[i]) not in bset)
     got = sorted(out.column("lk").cpu().numpy().tolist())
     assert got == exp
+
+
+def test_agg_accumulates_across_batches(gq):
+    """one hash table per PARTITION: multiple child batches accumulate and
+    emit a single result (TungstenAggregationIterator per-task contract)."""
+    from spark_amd import exec as gx
+    n = 90_000
+    keys = oracle.gen_i64(seed=970, n=n, range_=500)
+    vals = oracle.gen_f64_unit(seed=971, n=n)
+    thirds = [slice(0, n // 3), slice(n // 3, 2 * n // 3), slice(2 * n // 3, n)]
+    batches = [gx.ColumnarBatch({"k": to_dev(keys[s]), "v": to_dev(vals[s])})
+               for s in thirds]
+    node = gx.HashAggregateExec("k", [("sum", "v"), ("count*", None)],
+                                "complete", gx.InputBatches(batches))
+    outs = list(gx.GpuColumnarRule().pre_columnar_transitions(node)
+                .execute_columnar())
+    assert len(outs) == 1
+    out = outs[0]
+    ok, _, osum, _, _ = oracle.hash_agg(keys, vals)
+    assert out.num_rows() == len(ok)
+    gk = out.column("k").cpu().numpy()
+    gs = out.column("sum(v)").cpu().numpy()
+    gc = out.column("count(1)").cpu().numpy()
+    g, o = np.argsort(gk), np.argsort(ok)
+    assert (gk[g] == ok[o]).all()
+    np.testing.assert_allclose(gs[g], osum[o], rtol=1e-6)
+    ref_cnt = np.bincount(keys, minlength=500)
+    assert (gc[g] == ref_cnt[ok[o]]).all()
+
+    # composite keys across batches too
+    k2 = oracle.gen_i64(seed=972, n=n, range_=7)
+    b2 = [gx.ColumnarBatch({"a": to_dev(keys[s]), "b": to_dev(k2[s]),
+                            "v": to_dev(vals[s])}) for s in thirds]
+    node2 = gx.HashAggregateExec(("a", "b"), [("count*", None)], "complete",
+                                 gx.InputBatches(b2))
+    out2 = list(gx.GpuColumnarRule().pre_columnar_transitions(node2)
+                .execute_columnar())[0]
+    ref = {}
+    for i in range(n):
+        ref[(int(keys[i]), int(k2[i]))] = ref.get((int(keys[i]), int(k2[i])), 0) + 1
+    assert out2.num_rows() == len(ref)
+    ga = out2.column("a").cpu().numpy()
+    gb = out2.column("b").cpu().numpy()
+    gc2 = out2.column("count(1)").cpu().numpy()
+    for i in range(len(ga)):
+        assert ref[(int(ga[i]), int(gb[i]))] == int(gc2[i])
