@@ -193,8 +193,14 @@ class SortMergeJoinExec(_CpuNode):
 
     @property
     def output_ordering(self):
-        # SMJ's output is ordered by the streamed-side join keys
-        # (SortMergeJoinExec.scala:83 outputOrdering via getKeyOrdering)
+        # SMJ's output ordering follows the join type
+        # (SortMergeJoinExec.scala outputOrdering via getKeyOrdering:
+        # inner/left-ish -> left keys, RightOuter -> right keys,
+        # FullOuter -> none)
+        if self.join_type == "full_outer":
+            return []
+        if self.join_type == "right_outer":
+            return [SortOrder(self.right_key)]
         return [SortOrder(self.left_key)]
 
 
@@ -1144,8 +1150,9 @@ class GpuColumnarRule:
             j = GpuShuffledHashJoinExec(plan.left_key, plan.right_key,
                                         build, *children,
                                         join_type=plan.join_type)
-            if self.preserve_smj_ordering:
-                return GpuSortExec(SortOrder(plan.left_key), False, j)
+            declared = plan.output_ordering
+            if self.preserve_smj_ordering and declared:
+                return GpuSortExec(declared, False, j)
             return j
         if isinstance(plan, ShuffleExchangeExec):
             return GpuShuffleExchangeExec(plan.keys, *children)
