@@ -91,16 +91,29 @@ class SortWorkload:
 
     def step(self):
         gq = self.gq
+        # the payload interleave depends only on the input columns, not on
+        # the sort — run it on a SIDE stream concurrently with the radix
+        # passes (it is still inside the timed step; it just overlaps)
+        if not hasattr(self, "_side"):
+            self._side = torch.cuda.Stream()
+        ev = torch.cuda.Event()
+        # the scratch is read by the previous step's gather on the main
+        # stream — order the side-stream rewrite after it
+        self._side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._side):
+            gq._check(gq.lib().gpuq_interleave2_i64(
+                torch.cuda.current_stream().cuda_stream, self.rows,
+                self.pay1.data_ptr(), self.pay2.data_ptr(),
+                self.pair_scratch.data_ptr()))
+            ev.record(self._side)
         perm, skeys = gq.sort_perm(self.keys, workspace=self.ws,
                                    out_perm=self.perm, out_keys=self.skeys)
+        torch.cuda.current_stream().wait_event(ev)
         # interleaved-pair payload gather: one random b128 load per row
-        # serves both payload columns (the interleave pass runs inside the
-        # step — it is part of the operator's work)
-        gq._check(gq.lib().gpuq_gather2_i64_fast(
-            gq._stream(), self.rows,
-            self.pay1.data_ptr(), self.pay2.data_ptr(),
-            perm.data_ptr(), self.out1.data_ptr(), self.out2.data_ptr(),
-            self.pair_scratch.data_ptr()))
+        # serves both payload columns
+        gq._check(gq.lib().gpuq_gather2_pairs(
+            gq._stream(), self.rows, self.pair_scratch.data_ptr(),
+            perm.data_ptr(), self.out1.data_ptr(), self.out2.data_ptr()))
 
     def free(self):
         del self.keys, self.pay1, self.pay2, self.ws, self.out1, self.out2, \

@@ -123,6 +123,14 @@ int gpuq_gather2_i64_fast(void* stream, int64_t nrows, const void* a,
                           const void* b, const uint32_t* perm,
                           void* out_a, void* out_b, void* scratch);
 
+/* the two halves separately: the interleave depends only on the payload
+ * columns, so a host can run it on a SIDE stream concurrently with the
+ * sort passes and hide it entirely (event-sync before the pairs gather) */
+int gpuq_interleave2_i64(void* stream, int64_t nrows, const void* a,
+                         const void* b, void* pairs);
+int gpuq_gather2_pairs(void* stream, int64_t nrows, const void* pairs,
+                       const uint32_t* perm, void* out_a, void* out_b);
+
 /* ---------------------------------------------------------------- */
 /* HASH AGGREGATE — replaces HashAggregateExec                       */
 /* (execution/aggregate/HashAggregateExec.scala:99-151) for          */

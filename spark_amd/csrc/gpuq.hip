@@ -333,6 +333,38 @@ __global__ void k_gather2_pairs4(int64_t n, const ulonglong2* pairs,
   }
 }
 
+extern "C" int gpuq_interleave2_i64(void* stream, int64_t n, const void* a,
+                                    const void* b, void* pairs) {
+  hipStream_t s = (hipStream_t)stream;
+  if (n == 0) return GPUQ_OK;
+  { hipEvent_t _pe = prof_begin(s);
+  k_interleave2<<<grid1d(n), 256, 0, s>>>(n, (const uint64_t*)a,
+                                          (const uint64_t*)b,
+                                          (ulonglong2*)pairs);
+  prof_end("interleave2", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
+extern "C" int gpuq_gather2_pairs(void* stream, int64_t n, const void* pairs,
+                                  const uint32_t* perm, void* oa, void* ob) {
+  hipStream_t s = (hipStream_t)stream;
+  if (n == 0) return GPUQ_OK;
+  static int mlp0 = -1;
+  if (mlp0 < 0) mlp0 = getenv("GPUQ_GATHER_MLP") ? atoi(getenv("GPUQ_GATHER_MLP")) : 0;
+  { hipEvent_t _pe = prof_begin(s);
+  if (mlp0)
+    k_gather2_pairs4<<<grid1d((n + 3) / 4), 256, 0, s>>>(
+        n, (const ulonglong2*)pairs, perm, (uint64_t*)oa, (uint64_t*)ob);
+  else
+    k_gather2_pairs<<<grid1d(n), 256, 0, s>>>(n, (const ulonglong2*)pairs,
+                                              perm, (uint64_t*)oa,
+                                              (uint64_t*)ob);
+  prof_end("gather2_pairs", s, _pe); }
+  HIP_TRY(hipGetLastError());
+  return GPUQ_OK;
+}
+
 /* scratch: n * 16 bytes, device. */
 extern "C" int gpuq_gather2_i64_fast(void* stream, int64_t n, const void* a,
                                      const void* b, const uint32_t* perm,

@@ -65,6 +65,10 @@ def lib() -> ctypes.CDLL:
         L.gpuq_gather2_i64.argtypes = [vp, i64, vp, vp, vp, vp, vp]
         L.gpuq_gather2_i64_fast.restype = i32
         L.gpuq_gather2_i64_fast.argtypes = [vp, i64, vp, vp, vp, vp, vp, vp]
+        L.gpuq_interleave2_i64.restype = i32
+        L.gpuq_interleave2_i64.argtypes = [vp, i64, vp, vp, vp]
+        L.gpuq_gather2_pairs.restype = i32
+        L.gpuq_gather2_pairs.argtypes = [vp, i64, vp, vp, vp, vp]
         L.gpuq_hash_agg_workspace_bytes.restype = i64
         L.gpuq_hash_agg_workspace_bytes.argtypes = [i64]
         L.gpuq_hash_agg_i64_f64.restype = i32
