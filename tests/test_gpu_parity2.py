@@ -512,3 +512,25 @@ def test_agg_accumulates_across_batches(gq):
     gc2 = out2.column("count(1)").cpu().numpy()
     for i in range(len(ga)):
         assert ref[(int(ga[i]), int(gb[i]))] == int(gc2[i])
+
+
+def test_hash_agg_mid_cardinality_routing(gq):
+    """the auto-route band (cap 2^16..2^22, n >= 2^24) must take the
+    partitioned kernel and stay parity-green; outside the band the direct
+    table runs. Both against the oracle."""
+    n = 1 << 24
+    groups = 60_000
+    keys = oracle.gen_i64(seed=990, n=n, range_=groups)
+    vals = oracle.gen_f64_unit(seed=991, n=n)
+    cap = 1 << 17   # in band
+    gq.kernel_stats_reset()
+    gq.profiling(True)
+    gk, gkv, gs, gsv, gc = (t.cpu().numpy() for t in
+                            gq.hash_agg(to_dev(keys), to_dev(vals), cap))
+    _, scatter_calls = gq.kernel_stats("pagg_scatter")
+    assert scatter_calls > 0, "band input did not route to partitioned"
+    ok, okv, osum, osv, ocnt = oracle.hash_agg(keys, vals)
+    g, o = np.argsort(gk), np.argsort(ok)
+    assert (gk[g] == ok[o]).all() and (gc[g] == ocnt[o]).all()
+    np.testing.assert_allclose(gs[g], osum[o], rtol=1e-6)
+    gq.profiling(False)
