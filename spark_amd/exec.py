@@ -344,6 +344,33 @@ class InputBatches(SparkPlan):
 
 # ---------------- GPU exec nodes ----------------
 
+def concat_batches(batches):
+    """Concatenate a partition's batches into one (whole-partition
+    operators: sort, join build). Validity bitmaps concatenate through
+    the u8 wire form (row counts are not byte-aligned)."""
+    from . import gpuq
+    if len(batches) == 1:
+        return batches[0]
+    names = list(batches[0].columns().keys())
+    cols = {n: torch.cat([b.column(n) for b in batches]) for n in names}
+    validity = {}
+    for n in names:
+        if any(b.validity(n) is not None for b in batches):
+            parts = []
+            for b in batches:
+                v = b.validity(n)
+                nn = b.num_rows()
+                if v is None:
+                    parts.append(torch.ones(nn, dtype=torch.uint8,
+                                            device="cuda"))
+                else:
+                    parts.append(gpuq.bits_to_u8(v, nn))
+            validity[n] = gpuq.u8_to_bits(torch.cat(parts))
+    for b in batches:
+        b.close()
+    return ColumnarBatch(cols, validity=validity or None)
+
+
 class GpuSortExec(SparkPlan):
     """Replaces SortExec (SortExec.scala:75-126): radix-eligible
     int64/float64 keys (canUseRadixSort analog). Multi-key ORDER BY
@@ -404,7 +431,11 @@ class GpuSortExec(SparkPlan):
         import torch.distributed as dist
         from . import gpuq  # noqa: F401 (engine presence check)
         orders = self.sort_orders
-        for batch in self.children[0].execute_columnar():
+        # SortExec sorts the WHOLE partition (one sorter per task): a
+        # multi-batch child (e.g. Parquet row groups) concatenates first
+        batches = list(self.children[0].execute_columnar())
+        assert batches, "sort over a batchless child"
+        for batch in [concat_batches(batches)]:
             if (self.global_sort and dist.is_initialized()
                     and dist.get_world_size() > 1):
                 # global ORDER BY: range exchange on the PRIMARY key first
@@ -902,9 +933,10 @@ class GpuShuffledHashJoinExec(SparkPlan):
         from . import gpuq
         bi = 0 if self.build_side == "left" else 1
         build_b = list(self.children[bi].execute_columnar())
-        assert len(build_b) == 1, "one build batch per partition " \
-            "(buildHashedRelation materializes the build side)"
-        build = build_b[0]
+        assert build_b, "join build over a batchless child"
+        # buildHashedRelation materializes the whole build side per
+        # partition: multi-batch children concatenate
+        build = concat_batches(build_b)
         bkey = self.left_key if self.build_side == "left" else self.right_key
         pkey = self.right_key if self.build_side == "left" else self.left_key
         jt = {"inner": gpuq.JOIN_INNER, "left_outer": gpuq.JOIN_OUTER,

@@ -347,3 +347,40 @@ def test_empty_batch_pipeline():
         gx.ShuffledHashJoinExec("k", "rk", "right", left, empty2))
         .execute_columnar())[0]
     assert out.num_rows() == 0
+
+
+def test_whole_partition_sort_and_join_build_across_batches():
+    """SortExec sorts the whole partition and the join build side
+    materializes whole — multi-batch children (e.g. Parquet row groups)
+    concatenate, never per-batch results."""
+    n = 60_000
+    keys = oracle.gen_i64(seed=980, n=n, range_=10_000)
+    pay = oracle.gen_i64(seed=981, n=n)
+    valid = oracle.gen_i64(seed=982, n=n, range_=6) != 0
+    import numpy as _np
+    halves = [slice(0, n // 2), slice(n // 2, n)]
+    batches = [gx.ColumnarBatch(
+        {"k": torch.from_numpy(keys[s]).cuda(),
+         "p": torch.from_numpy(pay[s]).cuda()},
+        validity={"k": torch.from_numpy(
+            _np.packbits(valid[s], bitorder="little")).cuda()})
+        for s in halves]
+    out = list(gx.GpuColumnarRule().pre_columnar_transitions(
+        gx.SortExec(gx.SortOrder("k"), False, gx.InputBatches(batches)))
+        .execute_columnar())
+    assert len(out) == 1 and out[0].num_rows() == n
+    exp = oracle.sort_perm(keys, validity=_np.packbits(valid,
+                                                       bitorder="little"))
+    got_p = out[0].column("p").cpu().numpy()
+    assert (got_p == pay[exp]).all()
+
+    # join build side split across two batches
+    bk = oracle.gen_i64(seed=983, n=20_000, range_=9_000)
+    right = gx.InputBatches([
+        dev_batch(rk=bk[:12_000]), dev_batch(rk=bk[12_000:])])
+    left = gx.InputBatches([dev_batch(k=keys)])
+    j = list(gx.GpuColumnarRule().pre_columnar_transitions(
+        gx.ShuffledHashJoinExec("k", "rk", "right", left, right))
+        .execute_columnar())[0]
+    op, ob = oracle.join_inner(bk, keys)
+    assert j.num_rows() == len(op)
