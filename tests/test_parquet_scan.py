@@ -87,17 +87,15 @@ def test_q1_over_parquet_scan(tmp_path):
     ref = make_lineitem(path)
     cutoff = 10_470   # ~1998-09-02 (date '1998-12-01' - 90 days)
 
-    # scan the file to device (one batch per row group), concat into the
-    # task's single partition (the mirror is one-partition-per-rank)
+    # the scan feeds the plan directly: one batch per row group streams
+    # through the filter; the aggregate accumulates the whole partition
     scan_node = gx.GpuParquetScanExec(path)
-    batches = list(scan_node.execute_columnar())
-    assert len(batches) == 2   # two row groups
-    cols = {n_: torch.cat([b.column(n_) for b in batches])
-            for n_ in batches[0].columns()}
-    merged = gx.ColumnarBatch(cols)
+    probe = list(gx.GpuParquetScanExec(path).execute_columnar())
+    assert len(probe) == 2   # two row groups
+    for b in probe:
+        b.close()
 
-    filt = gx.FilterExec("l_shipdate", "<=", cutoff,
-                         gx.InputBatches([merged]))
+    filt = gx.FilterExec("l_shipdate", "<=", cutoff, scan_node)
     proj = gx.ProjectExec(
         ["l_returnflag", "l_linestatus", "l_quantity", "l_extendedprice",
          "l_discount",
