@@ -827,18 +827,20 @@ class GpuShuffleExchangeExec(SparkPlan):
         from this rank's partition-contiguous map output, no re-partition.
         specs: list of (start_partition, end_partition) half-open ranges.
         Returns one ColumnarBatch per spec (this rank's contribution)."""
-        assert hasattr(self, "_map_output"), "execute_columnar first"
-        cols, validity, offsets = self._map_output
+        assert hasattr(self, "_map_outputs"), "execute_columnar first"
+        from . import gpuq
         out = []
         for start, end in specs:
-            lo, hi = offsets[start], offsets[end]
-            sl = {n_: t[lo:hi] for n_, t in cols.items()}
-            va = {n_: None for n_ in sl}
-            for n_, u8 in validity.items():
-                from . import gpuq
-                va[n_] = gpuq.u8_to_bits(u8[lo:hi]) if hi > lo else None
-            out.append(ColumnarBatch(
-                sl, validity={k: v for k, v in va.items() if v is not None}))
+            parts = []
+            for cols, validity, offsets in self._map_outputs:
+                lo, hi = offsets[start], offsets[end]
+                sl = {n_: t[lo:hi] for n_, t in cols.items()}
+                va = {}
+                for n_, u8 in validity.items():
+                    if hi > lo:
+                        va[n_] = gpuq.u8_to_bits(u8[lo:hi])
+                parts.append(ColumnarBatch(sl, validity=va or None))
+            out.append(concat_batches(parts) if len(parts) > 1 else parts[0])
         return out
 
     def execute_columnar(self):
@@ -874,7 +876,9 @@ class GpuShuffleExchangeExec(SparkPlan):
             offsets = [0]
             for c in in_splits:
                 offsets.append(offsets[-1] + c)
-            self._map_output = (cols, vcols, offsets)
+            if not hasattr(self, "_map_outputs"):
+                self._map_outputs = []
+            self._map_outputs.append((cols, vcols, offsets))
             wire = dict(cols)
             wire.update({f"__valid__{n}": u8 for n, u8 in vcols.items()})
             out, _ = exchange_columns(wire, in_splits)

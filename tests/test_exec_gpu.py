@@ -209,7 +209,7 @@ def test_aqe_get_shuffle_partitions_coalesce():
         offsets.append(offsets[-1] + c)
 
     node = gx.GpuShuffleExchangeExec(("k",), gx.InputBatches([]))
-    node._map_output = (cols, vcols, offsets)
+    node._map_outputs = [(cols, vcols, offsets)]
     specs = [(0, 3), (3, 4), (4, 8)]
     batches = node.get_shuffle_partitions(specs)
     pids = oracle.partition_ids(keys, nparts)
